@@ -1,0 +1,52 @@
+// Batched slab ops — the hot loops of the parameter store.
+//
+// These replace the reference's scalar host loops (mergeValue / pull-copy /
+// delta-extract / replica-refresh, coloc_kv_server_handle.h:404-415, 464,
+// 630-648, 789-809) with batched kernels over the HBM slab. Each op takes
+// explicit per-key slab offsets computed by the host metadata pass, so the
+// device needs no key metadata at all.
+//
+// Two backends with identical semantics:
+//  - HIP/gfx950 (ops_hip.hip) — one workgroup per key, float4-vectorized,
+//    atomicAdd for concurrent-push safety.
+//  - CPU (ops_cpu.cpp) — plain loops, used by the no-GPU test tier.
+#pragma once
+#include <cstdint>
+
+namespace adapm {
+
+struct OpsBatch {
+  const int64_t* src_off;  // per-key slab offset (floats), or -1 to skip
+  const int64_t* dst_off;  // per-key offset into the out/in buffer (floats)
+  const int32_t* lens;     // per-key value length (floats)
+  int n;                   // number of keys in the batch
+};
+
+// out[dst_off[i] : +len] = slab[src_off[i] : +len]
+void ops_gather_gpu(const float* slab, const OpsBatch& b, float* out, void* stream);
+void ops_gather_cpu(const float* slab, const OpsBatch& b, float* out);
+
+// slab[dst(src)_off[i]] += in[...]   (atomic on GPU)  — or assign when set=true
+void ops_scatter_gpu(float* slab, const OpsBatch& b, const float* in, bool set, void* stream);
+void ops_scatter_cpu(float* slab, const OpsBatch& b, const float* in, bool set);
+
+// replica delta extraction (src_off = val offsets, dst_off = out buffer
+// offsets, aux_off = sync_state offsets):
+//   v = val[e]; out[e] = v - sync[e]; sync[e] = v
+// single-read-per-element so a concurrent atomic push is never lost (it
+// stays in val and is extracted next round).
+void ops_extract_gpu(float* slab, const OpsBatch& b, const int64_t* sync_off, float* out, void* stream);
+void ops_extract_cpu(float* slab, const OpsBatch& b, const int64_t* sync_off, float* out);
+
+// replica refresh apply (delta form so concurrent pushes are preserved):
+//   s = state_in[e]; atomicAdd(&val[e], s - sync[e]); sync[e] = s
+void ops_refresh_gpu(float* slab, const OpsBatch& b, const int64_t* sync_off, const float* state_in, void* stream);
+void ops_refresh_cpu(float* slab, const OpsBatch& b, const int64_t* sync_off, const float* state_in);
+
+// slab[dst_off[i] : +len] = 0
+void ops_zero_gpu(float* slab, const OpsBatch& b, void* stream);
+void ops_zero_cpu(float* slab, const OpsBatch& b);
+
+bool hip_available();
+
+}  // namespace adapm

@@ -1,0 +1,152 @@
+// HIP/gfx950 slab ops — see ops.h for semantics.
+//
+// Design (CDNA4): one 256-thread workgroup per key, grid-stride over keys
+// so batches of any size fill the 256 CUs; float4 when the slot is
+// 16B-aligned (allocator pads every slot to a multiple of 4 floats, so the
+// vector path is the common case); scalar tail otherwise. All slab writes
+// that can race with concurrent worker pushes are atomicAdd or follow the
+// single-read discipline documented in ops.h.
+#include <hip/hip_runtime.h>
+#include "ops.h"
+
+namespace adapm {
+
+#define THREADS 256
+
+__device__ inline bool vec_ok(int64_t a, int64_t b, int32_t len) {
+  return ((a | b) & 3) == 0 && (len & 3) == 0;
+}
+
+__global__ void k_gather(const float* __restrict__ slab, const int64_t* __restrict__ src_off,
+                         const int64_t* __restrict__ dst_off, const int32_t* __restrict__ lens,
+                         int n, float* __restrict__ out) {
+  for (int i = blockIdx.x; i < n; i += gridDim.x) {
+    int64_t s = src_off[i];
+    if (s < 0) continue;
+    int64_t d = dst_off[i];
+    int32_t len = lens[i];
+    if (vec_ok(s, d, len)) {
+      const float4* sp = reinterpret_cast<const float4*>(slab + s);
+      float4* dp = reinterpret_cast<float4*>(out + d);
+      for (int e = threadIdx.x; e < (len >> 2); e += THREADS) dp[e] = sp[e];
+    } else {
+      for (int e = threadIdx.x; e < len; e += THREADS) out[d + e] = slab[s + e];
+    }
+  }
+}
+
+__global__ void k_scatter_add(float* __restrict__ slab, const int64_t* __restrict__ src_off,
+                              const int64_t* __restrict__ dst_off, const int32_t* __restrict__ lens,
+                              int n, const float* __restrict__ in) {
+  for (int i = blockIdx.x; i < n; i += gridDim.x) {
+    int64_t s = src_off[i];
+    if (s < 0) continue;
+    int64_t d = dst_off[i];
+    int32_t len = lens[i];
+    for (int e = threadIdx.x; e < len; e += THREADS) atomicAdd(&slab[s + e], in[d + e]);
+  }
+}
+
+__global__ void k_scatter_set(float* __restrict__ slab, const int64_t* __restrict__ src_off,
+                              const int64_t* __restrict__ dst_off, const int32_t* __restrict__ lens,
+                              int n, const float* __restrict__ in) {
+  for (int i = blockIdx.x; i < n; i += gridDim.x) {
+    int64_t s = src_off[i];
+    if (s < 0) continue;
+    int64_t d = dst_off[i];
+    int32_t len = lens[i];
+    if (vec_ok(s, d, len)) {
+      const float4* ip = reinterpret_cast<const float4*>(in + d);
+      float4* sp = reinterpret_cast<float4*>(slab + s);
+      for (int e = threadIdx.x; e < (len >> 2); e += THREADS) sp[e] = ip[e];
+    } else {
+      for (int e = threadIdx.x; e < len; e += THREADS) slab[s + e] = in[d + e];
+    }
+  }
+}
+
+__global__ void k_extract(float* __restrict__ slab, const int64_t* __restrict__ val_off,
+                          const int64_t* __restrict__ out_off, const int32_t* __restrict__ lens,
+                          int n, const int64_t* __restrict__ sync_off, float* __restrict__ out) {
+  for (int i = blockIdx.x; i < n; i += gridDim.x) {
+    int64_t v = val_off[i];
+    if (v < 0) continue;
+    int64_t o = out_off[i], sy = sync_off[i];
+    int32_t len = lens[i];
+    for (int e = threadIdx.x; e < len; e += THREADS) {
+      float cur = slab[v + e];        // single read: a concurrent atomicAdd
+      out[o + e] = cur - slab[sy + e];  // after this read stays in val and is
+      slab[sy + e] = cur;             // captured by the next round's extract
+    }
+  }
+}
+
+__global__ void k_refresh(float* __restrict__ slab, const int64_t* __restrict__ val_off,
+                          const int64_t* __restrict__ in_off, const int32_t* __restrict__ lens,
+                          int n, const int64_t* __restrict__ sync_off,
+                          const float* __restrict__ state_in) {
+  for (int i = blockIdx.x; i < n; i += gridDim.x) {
+    int64_t v = val_off[i];
+    if (v < 0) continue;
+    int64_t o = in_off[i], sy = sync_off[i];
+    int32_t len = lens[i];
+    for (int e = threadIdx.x; e < len; e += THREADS) {
+      float s = state_in[o + e];
+      atomicAdd(&slab[v + e], s - slab[sy + e]);  // delta form: preserves
+      slab[sy + e] = s;                           // concurrent pushes
+    }
+  }
+}
+
+__global__ void k_zero(float* __restrict__ slab, const int64_t* __restrict__ dst_off,
+                       const int32_t* __restrict__ lens, int n) {
+  for (int i = blockIdx.x; i < n; i += gridDim.x) {
+    int64_t d = dst_off[i];
+    if (d < 0) continue;
+    int32_t len = lens[i];
+    for (int e = threadIdx.x; e < len; e += THREADS) slab[d + e] = 0.f;
+  }
+}
+
+static inline int grid_for(int n) {
+  // >=2048 workgroups fills 256 CUs at 8 blocks/CU; grid-stride covers the rest
+  int g = n < 1 ? 1 : n;
+  return g > 16384 ? 16384 : g;
+}
+
+void ops_gather_gpu(const float* slab, const OpsBatch& b, float* out, void* stream) {
+  if (b.n == 0) return;
+  hipLaunchKernelGGL(k_gather, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
+                     slab, b.src_off, b.dst_off, b.lens, b.n, out);
+}
+void ops_scatter_gpu(float* slab, const OpsBatch& b, const float* in, bool set, void* stream) {
+  if (b.n == 0) return;
+  if (set)
+    hipLaunchKernelGGL(k_scatter_set, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
+                       slab, b.src_off, b.dst_off, b.lens, b.n, in);
+  else
+    hipLaunchKernelGGL(k_scatter_add, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
+                       slab, b.src_off, b.dst_off, b.lens, b.n, in);
+}
+void ops_extract_gpu(float* slab, const OpsBatch& b, const int64_t* sync_off, float* out, void* stream) {
+  if (b.n == 0) return;
+  hipLaunchKernelGGL(k_extract, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
+                     slab, b.src_off, b.dst_off, b.lens, b.n, sync_off, out);
+}
+void ops_refresh_gpu(float* slab, const OpsBatch& b, const int64_t* sync_off, const float* state_in, void* stream) {
+  if (b.n == 0) return;
+  hipLaunchKernelGGL(k_refresh, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
+                     slab, b.src_off, b.dst_off, b.lens, b.n, sync_off, state_in);
+}
+void ops_zero_gpu(float* slab, const OpsBatch& b, void* stream) {
+  if (b.n == 0) return;
+  hipLaunchKernelGGL(k_zero, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
+                     slab, b.dst_off, b.lens, b.n);
+}
+
+bool hip_available() {
+  int n = 0;
+  return hipGetDeviceCount(&n) == hipSuccess && n > 0;
+}
+
+}  // namespace adapm

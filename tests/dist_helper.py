@@ -1,0 +1,61 @@
+"""Multi-process test harness: simulates multi-node on one host, exactly
+like the reference's tracker/dmlc_local.py test setup (reference
+tests/run_tests.sh) but with torch.distributed rendezvous over loopback."""
+from __future__ import annotations
+
+import multiprocessing as mp
+import os
+import socket
+import sys
+import traceback
+
+
+def _free_port() -> int:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def _entry(rank: int, world: int, port: int, fn, args, q):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ.setdefault("OMP_NUM_THREADS", "2")
+    try:
+        fn(rank, world, *args)
+        q.put((rank, None))
+    except Exception:
+        q.put((rank, traceback.format_exc()))
+        sys.exit(1)
+
+
+def run_dist(world: int, fn, *args, timeout: float = 120.0):
+    """Run fn(rank, world, *args) in `world` spawned processes."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [
+        ctx.Process(target=_entry, args=(r, world, port, fn, args, q), daemon=True)
+        for r in range(world)
+    ]
+    for p in procs:
+        p.start()
+    errs = []
+    for _ in range(world):
+        try:
+            rank, err = q.get(timeout=timeout)
+        except Exception:
+            for p in procs:
+                p.terminate()
+            raise TimeoutError(f"distributed test timed out after {timeout}s")
+        if err:
+            errs.append(f"[rank {rank}]\n{err}")
+    for p in procs:
+        p.join(timeout=30)
+        if p.is_alive():
+            p.terminate()
+    if errs:
+        raise AssertionError("\n".join(errs))
