@@ -155,6 +155,8 @@ struct ChannelState {
   std::vector<RespRec> responses;               // built in process, sent in respond
   std::unordered_map<Key, uint32_t> reloc_ctr;  // per-key relocation counter, travels
                                                 // with ownership (sync thread only)
+  std::unordered_map<Key, int64_t> reloc_round;  // round a key relocated IN (sync thread
+                                                 // only): relocation cooldown, see below
   std::atomic<int64_t> rounds{0};
 };
 
@@ -860,7 +862,7 @@ class Server {
     OutRec nr = r;
     int64_t* hf = hops_field == 1 ? &nr.f1 : &nr.f2;
     int hops = (int)(*hf >> 32);
-    if (hops >= 16) {
+    if (hops >= 64) {
       // give up: complete tickets so callers don't hang; value lost is
       // impossible here (ownership exists somewhere; 16 hops means a
       // directory pathologie) — log via stats.
@@ -995,6 +997,14 @@ class Server {
     }
     bool relocate = false;
     if (techniques_ != TECH_REPLICATION_ONLY && !local_intent && other_holders == 0) relocate = true;
+    // Relocation cooldown: a key that just relocated in may not relocate
+    // out for 3 rounds. Forwarded requests chase a moving key at one hop
+    // per round, so without this a key ping-ponging every round makes the
+    // chase a livelock; with cooldown >= 2 the request always catches up.
+    if (relocate) {
+      auto it = C.reloc_round.find(k);
+      if (it != C.reloc_round.end() && C.rounds.load() - it->second < 3) relocate = false;
+    }
     if (techniques_ == TECH_RELOCATION_ONLY && !relocate) {
       return;  // cannot replicate: requester keeps its stub, ops stay remote
     }
@@ -1015,6 +1025,7 @@ class Server {
       if (use_loc_cache_) loc_cache_[k] = origin_rank;
       uint32_t ctr = ++C.reloc_ctr[k];  // travels with ownership (sync thread only)
       C.reloc_ctr.erase(k);
+      C.reloc_round.erase(k);
       {
         std::lock_guard<std::mutex> g(C.mu);
         C.holders.erase(k);
@@ -1119,6 +1130,19 @@ class Server {
     HostBatch refreshes, acquires;
     std::vector<int64_t> refresh_sync;
     std::vector<std::pair<int64_t, int32_t>> local_frees;
+    // Flag transitions (STUB clear / owner upgrade) are DEFERRED until the
+    // value kernels are enqueued: a worker pull between the flag change
+    // and the kernel would otherwise serve an unrefreshed (zero) value.
+    // On GPU, a gather launched after the deferred flag change is
+    // stream-ordered behind the refresh kernel, so it reads fresh data.
+    struct Post {
+      Key k;
+      int64_t new_ver;
+      bool relocate;
+      int64_t acquire_off;  // >=0: set loc_ to this (stub was gone)
+      uint32_t ctr;
+    };
+    std::vector<Post> posts;
     int64_t poff = 0;
 
     for (int64_t i = 0; i < n_rec; ++i) {
@@ -1140,40 +1164,16 @@ class Server {
               // delta-form apply: val += state - sync; sync = state
               refreshes.add(loc_[k], poff, l);
               refresh_sync.push_back(sync_loc_[k]);
-              version_[k] = (uint32_t)f0;
-              if (relocate) {
-                local_frees.push_back({sync_loc_[k], l});  // after the kernel
-                sync_loc_[k] = -1;
-                flags_[k] = F_PRESENT | F_OWNER;
-                stat_relocated_in_ += 1;
-              } else {
-                flags_[k] = (f & ~F_STUB);
-              }
+              posts.push_back({k, f0, relocate, -1, ctr});
               handled = true;
             }
-          }
-          if (handled && relocate) {
-            C.reloc_ctr[k] = ctr;
-            {
-              std::lock_guard<std::mutex> g(C.mu);
-              C.replicas.erase(k);
-            }
-            if (use_loc_cache_) loc_cache_[k] = -1;
           }
           if (!handled && relocate) {
             // we dropped our stub while the relocation was in flight:
             // ownership transfer is unconditional — accept the value.
             int64_t voff = slab_.alloc(l);
             acquires.add(voff, poff, l);
-            {
-              std::lock_guard<std::mutex> lk(stripe(k));
-              loc_[k] = voff;
-              sync_loc_[k] = -1;
-              flags_[k] = F_PRESENT | F_OWNER;
-              version_[k] = (uint32_t)f0;
-            }
-            C.reloc_ctr[k] = ctr;
-            stat_relocated_in_ += 1;
+            posts.push_back({k, f0, true, voff, ctr});
           }
           poff += l;
           break;
@@ -1200,6 +1200,35 @@ class Server {
     }
     run_refresh(refreshes, refresh_sync, payload);
     run_scatter(acquires, payload, /*set=*/true);
+    // deferred flag transitions (see comment above)
+    for (auto& p : posts) {
+      int32_t l = len_of(p.k);
+      {
+        std::lock_guard<std::mutex> lk(stripe(p.k));
+        version_[p.k] = (uint32_t)p.new_ver;
+        if (p.acquire_off >= 0) {
+          loc_[p.k] = p.acquire_off;
+          sync_loc_[p.k] = -1;
+          flags_[p.k] = F_PRESENT | F_OWNER;
+        } else if (p.relocate) {
+          local_frees.push_back({sync_loc_[p.k], l});
+          sync_loc_[p.k] = -1;
+          flags_[p.k] = F_PRESENT | F_OWNER;
+        } else {
+          flags_[p.k] = flags_[p.k] & ~F_STUB;  // preserves a concurrent UPDATED
+        }
+      }
+      if (p.relocate) {
+        C.reloc_ctr[p.k] = p.ctr;
+        C.reloc_round[p.k] = C.rounds.load();
+        {
+          std::lock_guard<std::mutex> g(C.mu);
+          C.replicas.erase(p.k);
+        }
+        if (use_loc_cache_) loc_cache_[p.k] = -1;
+        stat_relocated_in_ += 1;
+      }
+    }
     for (auto& [off, len] : local_frees) slab_.free_(off, len);
   }
 
@@ -1208,9 +1237,11 @@ class Server {
     auto it = tickets_.find(ts);
     if (it == tickets_.end()) return;
     Ticket& t = *it->second;
+    if (t.out_off[out_index] < 0) return;  // duplicate response: ignore
     t.out.view({-1})
         .narrow(0, t.out_off[out_index], t.out_len[out_index])
         .copy_(data.view({-1}), /*non_blocking=*/true);
+    t.out_off[out_index] = -1 - t.out_off[out_index];  // mark answered
     t.received++;
     if (t.received >= t.expected) {
       if (t.caller_out.defined()) t.caller_out.view({-1}).copy_(t.out.view({-1}));

@@ -25,11 +25,24 @@ coordinated through a flag column in the size matrix.
 from __future__ import annotations
 
 import math
+import os
 import threading
 import time
 
 import torch
 import torch.distributed as dist
+
+_TRACE_KEY = int(os.environ.get("ADAPM_TRACE_KEY", "-1"))
+
+
+def _trace(rank, ch, direction, peer, meta):
+    if _TRACE_KEY == -1:
+        return
+    m = meta.reshape(-1, 5)
+    for r in m.tolist():
+        if _TRACE_KEY == -2 or r[1] == _TRACE_KEY:
+            print(f"[trace r{rank} ch{ch} {direction} peer{peer}] code={r[0]} key={r[1]} "
+                  f"f0={r[2]} f1={r[3]} f2={r[4]}", flush=True)
 
 
 class ActionTimer:
@@ -129,6 +142,7 @@ class SyncManager:
         sizes = torch.zeros(world, 3, dtype=torch.int64)
         msgs = {}
         for dest, meta, payload in outgoing:
+            _trace(rank, ch, "out", dest, meta)
             if dest == rank:
                 handler(ch, rank, meta, payload)
                 continue
@@ -176,6 +190,7 @@ class SyncManager:
         for peer in sorted(recv_bufs):
             rm, rp = recv_bufs[peer]
             meta = rm.cpu().reshape(-1, 5)
+            _trace(rank, ch, "in ", peer, meta)
             payload = rp if rp.device == store_dev else rp.to(store_dev)
             handler(ch, peer, meta, payload)
         return all_stopped

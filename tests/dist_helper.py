@@ -44,18 +44,31 @@ def run_dist(world: int, fn, *args, timeout: float = 120.0):
     for p in procs:
         p.start()
     errs = []
-    for _ in range(world):
+    done = 0
+    import time as _time
+
+    deadline = _time.monotonic() + timeout
+    while done < world and not errs:
         try:
-            rank, err = q.get(timeout=timeout)
+            rank, err = q.get(timeout=min(5.0, max(0.1, deadline - _time.monotonic())))
         except Exception:
-            for p in procs:
-                p.terminate()
-            raise TimeoutError(f"distributed test timed out after {timeout}s")
+            if _time.monotonic() >= deadline:
+                for p in procs:
+                    p.terminate()
+                raise TimeoutError(f"distributed test timed out after {timeout}s")
+            # a child may have died without reporting
+            if any(p.exitcode not in (None, 0) for p in procs):
+                errs.append("a child process died without reporting an error")
+            continue
+        done += 1
         if err:
             errs.append(f"[rank {rank}]\n{err}")
     for p in procs:
-        p.join(timeout=30)
-        if p.is_alive():
+        if errs:
             p.terminate()
+        else:
+            p.join(timeout=30)
+            if p.is_alive():
+                p.terminate()
     if errs:
         raise AssertionError("\n".join(errs))

@@ -1,0 +1,128 @@
+"""Intent-driven relocation/replication truth table (mirrors reference
+tests/test_locality_api.cc:50-133): after a sole intent the key relocates
+to the intender; under conflicting intent it is replicated; replicas are
+dropped when the intent window passes."""
+import time
+
+import torch
+
+from dist_helper import run_dist
+
+
+def _setup(rank, world, **kw):
+    import adapm_amd
+
+    adapm_amd.setup(num_keys=32, num_threads=1, device="cpu",
+                    max_sync_per_sec=2000.0, **kw)
+    s = adapm_amd.Server(4)
+    w = adapm_amd.Worker(0, s)
+    return adapm_amd, s, w
+
+
+def _wait_until(pred, timeout=20.0):
+    t0 = time.monotonic()
+    while time.monotonic() - t0 < timeout:
+        if pred():
+            return True
+        time.sleep(0.01)
+    return False
+
+
+def _relocation(rank, world):
+    _, s, w = _setup(rank, world)
+    w.barrier()
+    key = 1  # owned by rank 1 initially (manager = key % world)
+    assert w.is_local(key) == (rank == world - 1 if False else rank == key % world)
+    if rank == 0:
+        # rank 0 announces sole intent on key 1 -> relocation to rank 0
+        w.intent(torch.tensor([key]), 1, 1000)
+        assert _wait_until(lambda: w.is_local(key)), "key never relocated to intender"
+        # and pulls are now served locally (fast path, ts == -1)
+        out = torch.zeros(1, 4)
+        ts = w.pull(torch.tensor([key]), out, async_=True)
+        assert ts == -1
+    w.barrier()
+    if rank == key % world:
+        assert _wait_until(lambda: not w.is_local(key)), "old owner kept the key"
+    w.barrier()
+    w.finalize()
+    s.shutdown()
+
+
+def test_relocation_ws2():
+    run_dist(2, _relocation, timeout=180)
+
+
+def test_relocation_ws3():
+    run_dist(3, _relocation, timeout=180)
+
+
+def _replication(rank, world):
+    _, s, w = _setup(rank, world)
+    w.barrier()
+    key = 0  # owned by rank 0
+    # ALL ranks announce intent -> conflicting intent -> replication
+    w.intent(torch.tensor([key]), 1, 1_000_000)
+    assert _wait_until(lambda: w.is_local(key)), f"rank {rank} never got a copy"
+    w.barrier()
+    # owner still has it AND non-owners have replicas
+    assert w.is_local(key)
+    # pushes on replicas propagate to everyone (eventual consistency)
+    w.push(torch.tensor([key]), torch.ones(1, 4))
+    w.barrier()
+    w.wait_sync()
+    w.wait_sync()
+    out = torch.zeros(1, 4)
+    assert _wait_until(lambda: (w.pull(torch.tensor([key]), out) or True)
+                       and torch.equal(out, torch.full((1, 4), float(world)))), \
+        f"rank {rank} sees {out}"
+    w.barrier()
+    w.finalize()
+    s.shutdown()
+
+
+def test_replication_ws3():
+    run_dist(3, _replication, timeout=180)
+
+
+def _replica_drop(rank, world):
+    adapm, s, w = _setup(rank, world)
+    w.barrier()
+    key = 0
+    if rank == 0:
+        # owner keeps a long-lived local intent -> conflicting intent ->
+        # other ranks get replicas, not the relocation
+        w.intent(torch.tensor([key]), 1, 1_000_000)
+    w.barrier()
+    if rank != 0:
+        # intent window [1, 3): expires once clock reaches 3
+        w.intent(torch.tensor([key]), 1, 3)
+        assert _wait_until(lambda: w.is_local(key))
+        w.advance_clock()  # 1
+        w.advance_clock()  # 2
+        w.advance_clock()  # 3 -> intent expired
+        assert _wait_until(lambda: not w.is_local(key)), "replica never dropped"
+    w.barrier()
+    w.finalize()
+    s.shutdown()
+
+
+def test_replica_drop_ws2():
+    run_dist(2, _replica_drop, timeout=180)
+
+
+def _pull_if_local_semantics(rank, world):
+    _, s, w = _setup(rank, world)
+    w.barrier()
+    mine = torch.tensor([rank])
+    other = torch.tensor([(rank + 1) % world])
+    out = torch.zeros(1, 4)
+    assert w.pull_if_local(mine, out)
+    assert not w.pull_if_local(other, out)
+    w.barrier()
+    w.finalize()
+    s.shutdown()
+
+
+def test_pull_if_local_ws2():
+    run_dist(2, _pull_if_local_semantics, timeout=180)
