@@ -129,15 +129,18 @@ class SamplingManager:
     def _draw(self, n) -> np.ndarray:
         if self.with_replacement:
             return self.dist.draw(n)
-        out = np.empty(0, dtype=np.int64)
+        out = []
         seen = set()
         while len(out) < n:
             cand = self.dist.draw(max(2 * (n - len(out)), 16))
-            fresh = [c for c in cand if c not in seen]
-            for c in fresh:
-                seen.add(c)
-            out = np.concatenate([out, np.array(fresh[: n - len(out)], dtype=np.int64)])
-        return out
+            for c in cand:
+                c = int(c)
+                if c not in seen:
+                    seen.add(c)
+                    out.append(c)
+                    if len(out) == n:
+                        break
+        return np.array(out, dtype=np.int64)
 
     def _from_pool(self, n) -> np.ndarray:
         with self.lock:
