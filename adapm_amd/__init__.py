@@ -319,16 +319,21 @@ class Worker:
 
     # ---- collectives (replaces reference utils.h ps_allreduce)
 
-    def allreduce(self, value):
-        """Sum a scalar or tensor across ranks (loss/eval aggregation)."""
+    def allreduce(self, value, op: str = "sum"):
+        """Reduce a scalar or tensor across ranks (loss/eval aggregation;
+        replaces reference utils.h ps_allreduce). op: sum | max | min."""
         rt = self.server.rt
         scalar = not isinstance(value, torch.Tensor)
         t = torch.tensor([float(value)]) if scalar else value
         if rt.world > 1:
+            import torch.distributed as dist
+
+            ops = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX,
+                   "min": dist.ReduceOp.MIN}
             dev = rt.device if rt.backend == "nccl" else torch.device("cpu")
             td = t.to(dev)
             with rt.worker_group_lock:
-                torch.distributed.all_reduce(td, group=rt.worker_group)
+                dist.all_reduce(td, op=ops[op], group=rt.worker_group)
             t = td.to(t.device)
         return float(t.item()) if scalar else t
 

@@ -45,6 +45,7 @@
 #include <vector>
 
 #include "common.h"
+#include "kernels.h"
 #include "ops.h"
 
 #ifdef ADAPM_WITH_HIP
@@ -369,23 +370,41 @@ class Server {
     int64_t cum = 0;
     {
       InflightGuard g(this);
-      for (int64_t i = 0; i < n; ++i) {
-        Key k = kp[i];
-        int32_t l = len_of(k);
-        out_off[i] = cum;
-        out_len[i] = l;
-        {
-          std::lock_guard<std::mutex> lk(stripe(k));
-          uint8_t f = flags_[k];
-          if ((f & F_PRESENT) && !(f & F_STUB)) {
-            local.add(loc_[k], cum, l);
-            if (!(f & F_OWNER)) stat_pull_replica_ += 1;
-            stat_pull_local_ += 1;
+      if (layout_identity_.load(std::memory_order_acquire) && uniform_len_ >= 0) {
+        // fast path: offsets are pure arithmetic (see layout_identity_)
+        const int64_t plen = Slab::padded(uniform_len_);
+        const int32_t l = uniform_len_;
+        for (int64_t i = 0; i < n; ++i) {
+          Key k = kp[i];
+          out_off[i] = cum;
+          out_len[i] = l;
+          if (k % world_ == rank_) {
+            local.add((k / world_) * plen, cum, l);
           } else {
             remote.push_back({k, i});
           }
+          cum += l;
         }
-        cum += l;
+        stat_pull_local_ += n - (int64_t)remote.size();
+      } else {
+        for (int64_t i = 0; i < n; ++i) {
+          Key k = kp[i];
+          int32_t l = len_of(k);
+          out_off[i] = cum;
+          out_len[i] = l;
+          {
+            std::lock_guard<std::mutex> lk(stripe(k));
+            uint8_t f = flags_[k];
+            if ((f & F_PRESENT) && !(f & F_STUB)) {
+              local.add(loc_[k], cum, l);
+              if (!(f & F_OWNER)) stat_pull_replica_ += 1;
+              stat_pull_local_ += 1;
+            } else {
+              remote.push_back({k, i});
+            }
+          }
+          cum += l;
+        }
       }
       stat_pull_keys_ += n;
       run_gather(local, vals_dev);
@@ -438,26 +457,45 @@ class Server {
     int64_t cum = 0;
     {
       InflightGuard g(this);
-      for (int64_t i = 0; i < n; ++i) {
-        Key k = kp[i];
-        int32_t l = len_of(k);
-        {
-          std::lock_guard<std::mutex> lk(stripe(k));
-          uint8_t f = flags_[k];
-          if ((f & F_PRESENT) && (f & F_OWNER)) {
-            (set_mode ? assign : merge).add(loc_[k], cum, l);
-            version_[k]++;
-            stat_push_local_ += 1;
-          } else if ((f & F_PRESENT) && !set_mode) {
-            merge.add(loc_[k], cum, l);  // replica/stub: merge, flush at next sync
-            flags_[k] = f | F_UPDATED;
-            stat_push_local_ += 1;
-            stat_push_replica_ += 1;
+      if (layout_identity_.load(std::memory_order_acquire) && uniform_len_ >= 0) {
+        const int64_t plen = Slab::padded(uniform_len_);
+        const int32_t l = uniform_len_;
+        for (int64_t i = 0; i < n; ++i) {
+          Key k = kp[i];
+          if (k % world_ == rank_) {
+            (set_mode ? assign : merge).add((k / world_) * plen, cum, l);
           } else {
             remote.push_back({k, cum, l});
           }
+          cum += l;
         }
-        cum += l;
+        stat_push_local_ += n - (int64_t)remote.size();
+        // version bumps are skipped on the fast path: handle_owner_delta
+        // clears layout_identity_ when the first replica of one of our
+        // keys is granted, so while the flag holds no replica of our keys
+        // exists anywhere and versions are unobserved.
+      } else {
+        for (int64_t i = 0; i < n; ++i) {
+          Key k = kp[i];
+          int32_t l = len_of(k);
+          {
+            std::lock_guard<std::mutex> lk(stripe(k));
+            uint8_t f = flags_[k];
+            if ((f & F_PRESENT) && (f & F_OWNER)) {
+              (set_mode ? assign : merge).add(loc_[k], cum, l);
+              version_[k]++;
+              stat_push_local_ += 1;
+            } else if ((f & F_PRESENT) && !set_mode) {
+              merge.add(loc_[k], cum, l);  // replica/stub: merge, flush at next sync
+              flags_[k] = f | F_UPDATED;
+              stat_push_local_ += 1;
+              stat_push_replica_ += 1;
+            } else {
+              remote.push_back({k, cum, l});
+            }
+          }
+          cum += l;
+        }
       }
       stat_push_keys_ += n;
       run_scatter(merge, flat, false);
@@ -623,6 +661,7 @@ class Server {
           if (!(f & F_PRESENT)) {
             // replica stub: zeroed val+sync, absorbs pushes until first
             // refresh (reference handle registerNewIntentsForKeyUnsafe)
+            layout_identity_.store(false, std::memory_order_release);
             v_off = slab_.alloc(l);
             s_off = slab_.alloc(l);
             loc_[k] = v_off;
@@ -695,6 +734,7 @@ class Server {
                                     sync_loc_[k], len_of(k)});
           if (updated) flags_[k] = f & ~F_UPDATED;
           if (drop) {
+            layout_identity_.store(false, std::memory_order_release);
             frees.push_back({loc_[k], len_of(k)});
             frees.push_back({sync_loc_[k], len_of(k)});
             flags_[k] = 0;
@@ -1010,6 +1050,7 @@ class Server {
     }
 
     if (relocate) {
+      layout_identity_.store(false, std::memory_order_release);
       int64_t voff;
       int32_t l = len_of(k);
       int64_t new_ver;
@@ -1044,6 +1085,9 @@ class Server {
       stat_relocations_ += 1;
     } else {
       bool is_new = dflags & D_NEW;
+      // a replica of one of our keys now exists: versions become
+      // observable, so the fast path (which skips version bumps) ends.
+      layout_identity_.store(false, std::memory_order_release);
       {
         std::lock_guard<std::mutex> g(C.mu);
         C.holders[k] |= 1ULL << origin_rank;
@@ -1171,6 +1215,7 @@ class Server {
           if (!handled && relocate) {
             // we dropped our stub while the relocation was in flight:
             // ownership transfer is unconditional — accept the value.
+            layout_identity_.store(false, std::memory_order_release);
             int64_t voff = slab_.alloc(l);
             acquires.add(voff, poff, l);
             posts.push_back({k, f0, true, voff, ctr});
@@ -1369,6 +1414,15 @@ class Server {
   std::vector<std::atomic<Clock>> clocks_;
   Clock intent_ahead_ = 1LL << 40;  // default: act on intents immediately
 
+  // layout-identity fast path: true until the first structural change
+  // (stub / relocation / drop). While true, every key owned here sits at
+  // offset (k / world) * padded(uniform_len): the worker metadata pass
+  // becomes pure arithmetic (no locks, no random loc_ reads) — the "-1
+  // fast path" at full speed. The sync thread clears the flag BEFORE any
+  // structural change and then quiesces, so in-flight fast-path ops
+  // still see valid offsets (slot reuse is stream-ordered).
+  std::atomic<bool> layout_identity_{true};
+
   std::atomic<int> inflight_{0};
   std::atomic<int64_t> next_ts_{1};
   std::mutex tickets_mu_;
@@ -1384,11 +1438,85 @@ class Server {
       stat_dropped_records_{0}, stat_bytes_sent_{0}, stat_bytes_recv_{0}, stat_sampling_checks_{0};
 };
 
+// ------------------------------------------------------- app kernel wrappers
+
+static float* fp(torch::Tensor& t) { return t.data_ptr<float>(); }
+static const float* cfp(const torch::Tensor& t) { return t.data_ptr<float>(); }
+
+static void check_f32(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_contiguous() && t.scalar_type() == torch::kFloat32, name,
+              " must be contiguous float32");
+}
+
+// ComplEx train step; returns nothing, writes deltas + loss in place.
+void kge_complex_step(torch::Tensor s, torch::Tensor r, torch::Tensor o, torch::Tensor neg,
+                      torch::Tensor ds, torch::Tensor dr, torch::Tensor do_, torch::Tensor dneg,
+                      torch::Tensor loss, int64_t N, int64_t D, double lr, double eps) {
+  for (auto* t : {&s, &r, &o, &neg, &ds, &dr, &do_, &dneg, &loss}) check_f32(*t, "kge tensor");
+  int B = (int)loss.numel();
+  if (s.is_cuda()) {
+    TORCH_CHECK(hip_available(), "kge_complex_step: CUDA tensor but no HIP device");
+    kge_complex_step_gpu(cfp(s), cfp(r), cfp(o), cfp(neg), fp(ds), fp(dr), fp(do_), fp(dneg),
+                         fp(loss), B, (int)N, (int)D, (float)lr, (float)eps,
+                         current_stream(s.device()));
+  } else {
+    kge_complex_step_cpu(cfp(s), cfp(r), cfp(o), cfp(neg), fp(ds), fp(dr), fp(do_), fp(dneg),
+                         fp(loss), B, (int)N, (int)D, (float)lr, (float)eps);
+  }
+}
+
+void kge_complex_score(torch::Tensor s, torch::Tensor r, torch::Tensor cand,
+                       torch::Tensor scores, int64_t D) {
+  for (auto* t : {&s, &r, &cand, &scores}) check_f32(*t, "kge tensor");
+  int B = (int)scores.size(0), E = (int)scores.size(1);
+  if (s.is_cuda()) {
+    TORCH_CHECK(hip_available(), "kge_complex_score: CUDA tensor but no HIP device");
+    kge_complex_score_gpu(cfp(s), cfp(r), cfp(cand), fp(scores), B, E, (int)D,
+                          current_stream(s.device()));
+  } else {
+    kge_complex_score_cpu(cfp(s), cfp(r), cfp(cand), fp(scores), B, E, (int)D);
+  }
+}
+
+void w2v_sgns_step(torch::Tensor ctr, torch::Tensor ctx, torch::Tensor neg, torch::Tensor dctr,
+                   torch::Tensor dctx, torch::Tensor dneg, torch::Tensor loss, int64_t N,
+                   int64_t D, double lr, double eps) {
+  for (auto* t : {&ctr, &ctx, &neg, &dctr, &dctx, &dneg, &loss}) check_f32(*t, "w2v tensor");
+  int B = (int)loss.numel();
+  if (ctr.is_cuda()) {
+    TORCH_CHECK(hip_available(), "w2v_sgns_step: CUDA tensor but no HIP device");
+    w2v_sgns_step_gpu(cfp(ctr), cfp(ctx), cfp(neg), fp(dctr), fp(dctx), fp(dneg), fp(loss), B,
+                      (int)N, (int)D, (float)lr, (float)eps, current_stream(ctr.device()));
+  } else {
+    w2v_sgns_step_cpu(cfp(ctr), cfp(ctx), cfp(neg), fp(dctr), fp(dctx), fp(dneg), fp(loss), B,
+                      (int)N, (int)D, (float)lr, (float)eps);
+  }
+}
+
+void mf_update_step(torch::Tensor w, torch::Tensor h, torch::Tensor x, torch::Tensor dw,
+                    torch::Tensor dh, torch::Tensor loss, int64_t R, double lr, double lambda,
+                    double eps) {
+  for (auto* t : {&w, &h, &x, &dw, &dh, &loss}) check_f32(*t, "mf tensor");
+  int B = (int)loss.numel();
+  if (w.is_cuda()) {
+    TORCH_CHECK(hip_available(), "mf_update_step: CUDA tensor but no HIP device");
+    mf_update_step_gpu(cfp(w), cfp(h), cfp(x), fp(dw), fp(dh), fp(loss), B, (int)R, (float)lr,
+                       (float)lambda, (float)eps, current_stream(w.device()));
+  } else {
+    mf_update_step_cpu(cfp(w), cfp(h), cfp(x), fp(dw), fp(dh), fp(loss), B, (int)R, (float)lr,
+                       (float)lambda, (float)eps);
+  }
+}
+
 }  // namespace adapm
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   using namespace adapm;
   m.def("hip_available", &hip_available);
+  m.def("kge_complex_step", &kge_complex_step, py::call_guard<py::gil_scoped_release>());
+  m.def("kge_complex_score", &kge_complex_score, py::call_guard<py::gil_scoped_release>());
+  m.def("w2v_sgns_step", &w2v_sgns_step, py::call_guard<py::gil_scoped_release>());
+  m.def("mf_update_step", &mf_update_step, py::call_guard<py::gil_scoped_release>());
   py::class_<Server>(m, "Server")
       .def(py::init<int64_t, torch::Tensor, int, int, int, int, std::string, double, int, bool>(),
            py::arg("num_keys"), py::arg("value_lengths"), py::arg("rank"), py::arg("world"),

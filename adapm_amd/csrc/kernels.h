@@ -1,0 +1,57 @@
+// App compute kernels (the reference's per-datapoint scalar loops become
+// batched device kernels — SURVEY.md §2.5 inventory):
+//   - KGE ComplEx score/grad + fused AdaGrad (reference
+//     knowledge_graph_embeddings.cc:832-858 score/grad, 415-435 AdaGrad)
+//   - word2vec SGNS step (reference word2vec.cc:679-745)
+//   - MF NZSL+L2 update (reference apps/mf/update.h:32-69)
+//
+// Value layout convention (same as the reference apps): every key's value
+// is [embedding(D) | adagrad_accum(D)], so AdaGrad state rides the same
+// Push/Pull path. Kernels consume pulled rows and produce PUSH-ready
+// additive deltas: [delta_emb(D) | grad^2(D)].
+#pragma once
+#include <cstdint>
+
+namespace adapm {
+
+// ComplEx training step over B positives with N o-side negatives each.
+//  s, r, o:   [B][2D]   pulled rows (emb | accum)
+//  neg:       [B*N][2D] pulled negative-entity rows
+//  ds,dr,do_,dneg: same shapes, outputs (push deltas)
+//  loss:      [B] logistic loss of each positive (+ its negatives)
+// D = embedding dim (even: first half real, second half imaginary).
+void kge_complex_step_gpu(const float* s, const float* r, const float* o, const float* neg,
+                          float* ds, float* dr, float* do_, float* dneg, float* loss,
+                          int B, int N, int D, float lr, float eps, void* stream);
+void kge_complex_step_cpu(const float* s, const float* r, const float* o, const float* neg,
+                          float* ds, float* dr, float* do_, float* dneg, float* loss,
+                          int B, int N, int D, float lr, float eps);
+
+// ComplEx scoring only (evaluation): score[b][e] = psi(s_b, r_b, cand_e)
+//  cand: [E][2D] candidate entity rows; scores: [B][E]
+void kge_complex_score_gpu(const float* s, const float* r, const float* cand, float* scores,
+                           int B, int E, int D, void* stream);
+void kge_complex_score_cpu(const float* s, const float* r, const float* cand, float* scores,
+                           int B, int E, int D);
+
+// word2vec SGNS step: per center/context pair with N negatives.
+//  ctr:  [B][2D] center (syn0) rows;  ctx: [B][2D] context (syn1) rows
+//  neg:  [B*N][2D] negative (syn1) rows
+//  outputs: push deltas, same shapes; loss [B]
+void w2v_sgns_step_gpu(const float* ctr, const float* ctx, const float* neg, float* dctr,
+                       float* dctx, float* dneg, float* loss, int B, int N, int D, float lr,
+                       float eps, void* stream);
+void w2v_sgns_step_cpu(const float* ctr, const float* ctx, const float* neg, float* dctr,
+                       float* dctx, float* dneg, float* loss, int B, int N, int D, float lr,
+                       float eps);
+
+// Matrix-factorization NZSL step: per nonzero (i, j, x):
+//  w: [B][2R] row-factor rows, h: [B][2R] col-factor rows (R = rank)
+//  outputs dw, dh (push deltas), loss [B] squared error
+void mf_update_step_gpu(const float* w, const float* h, const float* x, float* dw, float* dh,
+                        float* loss, int B, int R, float lr, float lambda, float eps,
+                        void* stream);
+void mf_update_step_cpu(const float* w, const float* h, const float* x, float* dw, float* dh,
+                        float* loss, int B, int R, float lr, float lambda, float eps);
+
+}  // namespace adapm

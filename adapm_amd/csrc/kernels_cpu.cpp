@@ -1,1 +1,142 @@
-// app kernels (KGE/w2v/MF) — filled in as models land
+// CPU backend of the app kernels (kernels.h) — identical math to
+// kernels_hip.hip, used by the no-GPU test tier and as the reference for
+// GPU numerics tests (alongside plain-torch fp32 references in tests/).
+#include <cmath>
+#include <vector>
+
+#include "kernels.h"
+
+namespace adapm {
+
+static inline float sigmoidf_(float x) { return 1.f / (1.f + std::exp(-x)); }
+static inline float softplusf_(float x) { return x > 20.f ? x : std::log1p(std::exp(x)); }
+
+void kge_complex_step_cpu(const float* s, const float* r, const float* o, const float* neg,
+                          float* ds, float* dr, float* do_, float* dneg, float* loss, int B,
+                          int N, int D, float lr, float eps) {
+  const int dc = D >> 1;
+  const int row = D << 1;
+  std::vector<float> a_sre(dc), a_sim(dc), a_rre(dc), a_rim(dc), u_re(dc), u_im(dc);
+  for (int b = 0; b < B; ++b) {
+    const float* sb = s + (int64_t)b * row;
+    const float* rb = r + (int64_t)b * row;
+    for (int k = 0; k < dc; ++k) {
+      u_re[k] = sb[k] * rb[k] - sb[dc + k] * rb[dc + k];
+      u_im[k] = sb[dc + k] * rb[k] + sb[k] * rb[dc + k];
+      a_sre[k] = a_sim[k] = a_rre[k] = a_rim[k] = 0.f;
+    }
+    float lsum = 0.f;
+    for (int j = 0; j <= N; ++j) {
+      const float* ob = (j == 0) ? o + (int64_t)b * row : neg + ((int64_t)b * N + (j - 1)) * row;
+      float* dob = (j == 0) ? do_ + (int64_t)b * row : dneg + ((int64_t)b * N + (j - 1)) * row;
+      float y = (j == 0) ? 1.f : -1.f;
+      double psi = 0.0;
+      for (int k = 0; k < dc; ++k) psi += u_re[k] * ob[k] + u_im[k] * ob[dc + k];
+      float c = -y * sigmoidf_(-y * (float)psi);
+      lsum += softplusf_(-y * (float)psi);
+      for (int k = 0; k < dc; ++k) {
+        float o_re = ob[k], o_im = ob[dc + k];
+        a_sre[k] += c * (rb[k] * o_re + rb[dc + k] * o_im);
+        a_sim[k] += c * (rb[k] * o_im - rb[dc + k] * o_re);
+        a_rre[k] += c * (sb[k] * o_re + sb[dc + k] * o_im);
+        a_rim[k] += c * (sb[k] * o_im - sb[dc + k] * o_re);
+        float g_re = c * u_re[k], g_im = c * u_im[k];
+        dob[k] = -lr * g_re / std::sqrt(ob[D + k] + g_re * g_re + eps);
+        dob[dc + k] = -lr * g_im / std::sqrt(ob[D + dc + k] + g_im * g_im + eps);
+        dob[D + k] = g_re * g_re;
+        dob[D + dc + k] = g_im * g_im;
+      }
+    }
+    float* dsb = ds + (int64_t)b * row;
+    float* drb = dr + (int64_t)b * row;
+    for (int k = 0; k < dc; ++k) {
+      dsb[k] = -lr * a_sre[k] / std::sqrt(sb[D + k] + a_sre[k] * a_sre[k] + eps);
+      dsb[dc + k] = -lr * a_sim[k] / std::sqrt(sb[D + dc + k] + a_sim[k] * a_sim[k] + eps);
+      dsb[D + k] = a_sre[k] * a_sre[k];
+      dsb[D + dc + k] = a_sim[k] * a_sim[k];
+      drb[k] = -lr * a_rre[k] / std::sqrt(rb[D + k] + a_rre[k] * a_rre[k] + eps);
+      drb[dc + k] = -lr * a_rim[k] / std::sqrt(rb[D + dc + k] + a_rim[k] * a_rim[k] + eps);
+      drb[D + k] = a_rre[k] * a_rre[k];
+      drb[D + dc + k] = a_rim[k] * a_rim[k];
+    }
+    loss[b] = lsum;
+  }
+}
+
+void kge_complex_score_cpu(const float* s, const float* r, const float* cand, float* scores,
+                           int B, int E, int D) {
+  const int dc = D >> 1;
+  const int row = D << 1;
+  for (int b = 0; b < B; ++b) {
+    const float* sb = s + (int64_t)b * row;
+    const float* rb = r + (int64_t)b * row;
+    for (int e = 0; e < E; ++e) {
+      const float* ob = cand + (int64_t)e * row;
+      double psi = 0.0;
+      for (int k = 0; k < dc; ++k) {
+        float ure = sb[k] * rb[k] - sb[dc + k] * rb[dc + k];
+        float uim = sb[dc + k] * rb[k] + sb[k] * rb[dc + k];
+        psi += ure * ob[k] + uim * ob[dc + k];
+      }
+      scores[(int64_t)b * E + e] = (float)psi;
+    }
+  }
+}
+
+void w2v_sgns_step_cpu(const float* ctr, const float* ctx, const float* neg, float* dctr,
+                       float* dctx, float* dneg, float* loss, int B, int N, int D, float lr,
+                       float eps) {
+  const int row = D << 1;
+  std::vector<float> a_c(D);
+  for (int b = 0; b < B; ++b) {
+    const float* cb = ctr + (int64_t)b * row;
+    std::fill(a_c.begin(), a_c.end(), 0.f);
+    float lsum = 0.f;
+    for (int j = 0; j <= N; ++j) {
+      const float* xb = (j == 0) ? ctx + (int64_t)b * row : neg + ((int64_t)b * N + (j - 1)) * row;
+      float* dxb = (j == 0) ? dctx + (int64_t)b * row : dneg + ((int64_t)b * N + (j - 1)) * row;
+      float y = (j == 0) ? 1.f : -1.f;
+      double dot = 0.0;
+      for (int k = 0; k < D; ++k) dot += cb[k] * xb[k];
+      float g = -y * sigmoidf_(-y * (float)dot);
+      lsum += softplusf_(-y * (float)dot);
+      for (int k = 0; k < D; ++k) {
+        a_c[k] += g * xb[k];
+        float gx = g * cb[k];
+        dxb[k] = -lr * gx / std::sqrt(xb[D + k] + gx * gx + eps);
+        dxb[D + k] = gx * gx;
+      }
+    }
+    float* dcb = dctr + (int64_t)b * row;
+    for (int k = 0; k < D; ++k) {
+      dcb[k] = -lr * a_c[k] / std::sqrt(cb[D + k] + a_c[k] * a_c[k] + eps);
+      dcb[D + k] = a_c[k] * a_c[k];
+    }
+    loss[b] = lsum;
+  }
+}
+
+void mf_update_step_cpu(const float* w, const float* h, const float* x, float* dw, float* dh,
+                        float* loss, int B, int R, float lr, float lambda, float eps) {
+  const int row = R << 1;
+  for (int b = 0; b < B; ++b) {
+    const float* wb = w + (int64_t)b * row;
+    const float* hb = h + (int64_t)b * row;
+    double pred = 0.0;
+    for (int k = 0; k < R; ++k) pred += wb[k] * hb[k];
+    float e = x[b] - (float)pred;
+    loss[b] = e * e;
+    float* dwb = dw + (int64_t)b * row;
+    float* dhb = dh + (int64_t)b * row;
+    for (int k = 0; k < R; ++k) {
+      float gw = -2.f * e * hb[k] + 2.f * lambda * wb[k];
+      float gh = -2.f * e * wb[k] + 2.f * lambda * hb[k];
+      dwb[k] = -lr * gw / std::sqrt(wb[R + k] + gw * gw + eps);
+      dwb[R + k] = gw * gw;
+      dhb[k] = -lr * gh / std::sqrt(hb[R + k] + gh * gh + eps);
+      dhb[R + k] = gh * gh;
+    }
+  }
+}
+
+}  // namespace adapm

@@ -1,1 +1,304 @@
-// app kernels (KGE/w2v/MF) — filled in as models land
+// HIP/gfx950 app kernels — see kernels.h. One workgroup (256 threads =
+// 4 waves) per training sample; rows are 2D floats = [emb(D) | accum(D)].
+// These kernels are HBM-bound (each pulled row is read once, each delta
+// row written once); the arithmetic per element is small, so the design
+// goal is coalesced float4 row traffic and cheap block reductions
+// (wave __shfl_xor + one LDS hop across the 4 waves).
+#include <hip/hip_runtime.h>
+
+#include "kernels.h"
+
+namespace adapm {
+
+#define KT 256  // threads per workgroup
+
+__device__ inline float block_reduce_sum(float v, float* lds) {
+  // wave-level reduce (64 lanes)
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  int wave = threadIdx.x >> 6;
+  int lane = threadIdx.x & 63;
+  if (lane == 0) lds[wave] = v;
+  __syncthreads();
+  float r = (threadIdx.x < KT / 64) ? lds[threadIdx.x] : 0.f;
+  if (threadIdx.x < 64) {
+    for (int off = 2; off > 0; off >>= 1) r += __shfl_xor(r, off, 64);
+  }
+  if (threadIdx.x == 0) lds[0] = r;
+  __syncthreads();
+  float out = lds[0];
+  __syncthreads();
+  return out;
+}
+
+__device__ inline float sigmoidf(float x) { return 1.f / (1.f + __expf(-x)); }
+__device__ inline float softplusf(float x) {
+  // log(1+exp(x)), stable
+  return x > 20.f ? x : __logf(1.f + __expf(x));
+}
+
+// --------------------------------------------------------------- ComplEx
+
+// Per workgroup: one positive triple + its N negatives (o-side corruption).
+// Each thread owns complex positions k = tid + i*KT, i < KPT (compile-time
+// KPT so the per-thread arrays stay in registers — runtime-indexed arrays
+// spill to scratch on gfx950). D = 512 -> dc = 256 -> KPT = 1.
+template <int KPT>
+__global__ void k_kge_step(const float* __restrict__ s, const float* __restrict__ r,
+                           const float* __restrict__ o, const float* __restrict__ neg,
+                           float* __restrict__ ds, float* __restrict__ dr,
+                           float* __restrict__ do_, float* __restrict__ dneg,
+                           float* __restrict__ loss, int B, int N, int D, float lr, float eps) {
+  __shared__ float lds[KT / 64];
+  const int dc = D >> 1;
+  const int row = D << 1;  // floats per pulled row
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    const float* sb = s + (int64_t)b * row;
+    const float* rb = r + (int64_t)b * row;
+
+    float s_re[KPT], s_im[KPT], r_re[KPT], r_im[KPT];
+    float a_sre[KPT], a_sim[KPT], a_rre[KPT], a_rim[KPT];
+    float u_re[KPT], u_im[KPT];
+#pragma unroll
+    for (int i = 0; i < KPT; ++i) {
+      int k = threadIdx.x + i * KT;
+      a_sre[i] = a_sim[i] = a_rre[i] = a_rim[i] = 0.f;
+      if (k < dc) {
+        s_re[i] = sb[k];
+        s_im[i] = sb[dc + k];
+        r_re[i] = rb[k];
+        r_im[i] = rb[dc + k];
+        // per-object o-gradient direction (independent of o)
+        u_re[i] = s_re[i] * r_re[i] - s_im[i] * r_im[i];
+        u_im[i] = s_im[i] * r_re[i] + s_re[i] * r_im[i];
+      } else {
+        s_re[i] = s_im[i] = r_re[i] = r_im[i] = u_re[i] = u_im[i] = 0.f;
+      }
+    }
+
+    float lsum = 0.f;
+    for (int j = 0; j <= N; ++j) {
+      const float* ob = (j == 0) ? o + (int64_t)b * row
+                                 : neg + ((int64_t)b * N + (j - 1)) * row;
+      float y = (j == 0) ? 1.f : -1.f;
+      float part = 0.f;
+      float o_re[KPT], o_im[KPT];
+#pragma unroll
+      for (int i = 0; i < KPT; ++i) {
+        int k = threadIdx.x + i * KT;
+        o_re[i] = k < dc ? ob[k] : 0.f;
+        o_im[i] = k < dc ? ob[dc + k] : 0.f;
+        part += u_re[i] * o_re[i] + u_im[i] * o_im[i];
+      }
+      float psi = block_reduce_sum(part, lds);
+      float c = -y * sigmoidf(-y * psi);  // dL/dpsi, L = softplus(-y*psi)
+      if (threadIdx.x == 0) lsum += softplusf(-y * psi);
+
+      float* dob = (j == 0) ? do_ + (int64_t)b * row
+                            : dneg + ((int64_t)b * N + (j - 1)) * row;
+#pragma unroll
+      for (int i = 0; i < KPT; ++i) {
+        int k = threadIdx.x + i * KT;
+        if (k >= dc) continue;
+        // accumulate s/r grads
+        a_sre[i] += c * (r_re[i] * o_re[i] + r_im[i] * o_im[i]);
+        a_sim[i] += c * (r_re[i] * o_im[i] - r_im[i] * o_re[i]);
+        a_rre[i] += c * (s_re[i] * o_re[i] + s_im[i] * o_im[i]);
+        a_rim[i] += c * (s_re[i] * o_im[i] - s_im[i] * o_re[i]);
+        // o grad + fused AdaGrad (G in the row's second half)
+        float g_re = c * u_re[i];
+        float g_im = c * u_im[i];
+        float G_re = ob[D + k] + g_re * g_re;
+        float G_im = ob[D + dc + k] + g_im * g_im;
+        dob[k] = -lr * g_re * __frsqrt_rn(G_re + eps);
+        dob[dc + k] = -lr * g_im * __frsqrt_rn(G_im + eps);
+        dob[D + k] = g_re * g_re;
+        dob[D + dc + k] = g_im * g_im;
+      }
+    }
+    // write s/r deltas with fused AdaGrad
+    float* dsb = ds + (int64_t)b * row;
+    float* drb = dr + (int64_t)b * row;
+#pragma unroll
+    for (int i = 0; i < KPT; ++i) {
+      int k = threadIdx.x + i * KT;
+      if (k >= dc) continue;
+      float Gsr = sb[D + k] + a_sre[i] * a_sre[i];
+      float Gsi = sb[D + dc + k] + a_sim[i] * a_sim[i];
+      dsb[k] = -lr * a_sre[i] * __frsqrt_rn(Gsr + eps);
+      dsb[dc + k] = -lr * a_sim[i] * __frsqrt_rn(Gsi + eps);
+      dsb[D + k] = a_sre[i] * a_sre[i];
+      dsb[D + dc + k] = a_sim[i] * a_sim[i];
+      float Grr = rb[D + k] + a_rre[i] * a_rre[i];
+      float Gri = rb[D + dc + k] + a_rim[i] * a_rim[i];
+      drb[k] = -lr * a_rre[i] * __frsqrt_rn(Grr + eps);
+      drb[dc + k] = -lr * a_rim[i] * __frsqrt_rn(Gri + eps);
+      drb[D + k] = a_rre[i] * a_rre[i];
+      drb[D + dc + k] = a_rim[i] * a_rim[i];
+    }
+    if (threadIdx.x == 0) loss[b] = lsum;
+  }
+}
+
+__global__ void k_kge_score(const float* __restrict__ s, const float* __restrict__ r,
+                            const float* __restrict__ cand, float* __restrict__ scores, int B,
+                            int E, int D) {
+  __shared__ float lds[KT / 64];
+  const int dc = D >> 1;
+  const int row = D << 1;
+  for (int be = blockIdx.x; be < B * E; be += gridDim.x) {
+    int b = be / E, e = be % E;
+    const float* sb = s + (int64_t)b * row;
+    const float* rb = r + (int64_t)b * row;
+    const float* ob = cand + (int64_t)e * row;
+    float part = 0.f;
+    for (int k = threadIdx.x; k < dc; k += KT) {
+      float ure = sb[k] * rb[k] - sb[dc + k] * rb[dc + k];
+      float uim = sb[dc + k] * rb[k] + sb[k] * rb[dc + k];
+      part += ure * ob[k] + uim * ob[dc + k];
+    }
+    float psi = block_reduce_sum(part, lds);
+    if (threadIdx.x == 0) scores[be] = psi;
+  }
+}
+
+// --------------------------------------------------------------- SGNS
+
+template <int KPT>
+__global__ void k_w2v_step(const float* __restrict__ ctr, const float* __restrict__ ctx,
+                           const float* __restrict__ neg, float* __restrict__ dctr,
+                           float* __restrict__ dctx, float* __restrict__ dneg,
+                           float* __restrict__ loss, int B, int N, int D, float lr, float eps) {
+  __shared__ float lds[KT / 64];
+  const int row = D << 1;
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    const float* cb = ctr + (int64_t)b * row;
+    float c_emb[KPT], a_c[KPT];
+#pragma unroll
+    for (int i = 0; i < KPT; ++i) {
+      int k = threadIdx.x + i * KT;
+      c_emb[i] = k < D ? cb[k] : 0.f;
+      a_c[i] = 0.f;
+    }
+
+    float lsum = 0.f;
+    for (int j = 0; j <= N; ++j) {
+      const float* xb = (j == 0) ? ctx + (int64_t)b * row
+                                 : neg + ((int64_t)b * N + (j - 1)) * row;
+      float y = (j == 0) ? 1.f : -1.f;
+      float part = 0.f;
+      float x_emb[KPT];
+#pragma unroll
+      for (int i = 0; i < KPT; ++i) {
+        int k = threadIdx.x + i * KT;
+        x_emb[i] = k < D ? xb[k] : 0.f;
+        part += c_emb[i] * x_emb[i];
+      }
+      float dot = block_reduce_sum(part, lds);
+      float g = -y * sigmoidf(-y * dot);
+      if (threadIdx.x == 0) lsum += softplusf(-y * dot);
+      float* dxb = (j == 0) ? dctx + (int64_t)b * row
+                            : dneg + ((int64_t)b * N + (j - 1)) * row;
+#pragma unroll
+      for (int i = 0; i < KPT; ++i) {
+        int k = threadIdx.x + i * KT;
+        if (k >= D) continue;
+        a_c[i] += g * x_emb[i];
+        float gx = g * c_emb[i];
+        float G = xb[D + k] + gx * gx;
+        dxb[k] = -lr * gx * __frsqrt_rn(G + eps);
+        dxb[D + k] = gx * gx;
+      }
+    }
+    float* dcb = dctr + (int64_t)b * row;
+#pragma unroll
+    for (int i = 0; i < KPT; ++i) {
+      int k = threadIdx.x + i * KT;
+      if (k >= D) continue;
+      float G = cb[D + k] + a_c[i] * a_c[i];
+      dcb[k] = -lr * a_c[i] * __frsqrt_rn(G + eps);
+      dcb[D + k] = a_c[i] * a_c[i];
+    }
+    if (threadIdx.x == 0) loss[b] = lsum;
+  }
+}
+
+// --------------------------------------------------------------- MF
+
+__global__ void k_mf_step(const float* __restrict__ w, const float* __restrict__ h,
+                          const float* __restrict__ x, float* __restrict__ dw,
+                          float* __restrict__ dh, float* __restrict__ loss, int B, int R,
+                          float lr, float lambda, float eps) {
+  __shared__ float lds[KT / 64];
+  const int row = R << 1;
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    const float* wb = w + (int64_t)b * row;
+    const float* hb = h + (int64_t)b * row;
+    float part = 0.f;
+    for (int k = threadIdx.x; k < R; k += KT) part += wb[k] * hb[k];
+    float pred = block_reduce_sum(part, lds);
+    float e = x[b] - pred;
+    if (threadIdx.x == 0) loss[b] = e * e;
+    float* dwb = dw + (int64_t)b * row;
+    float* dhb = dh + (int64_t)b * row;
+    for (int k = threadIdx.x; k < R; k += KT) {
+      float gw = -2.f * e * hb[k] + 2.f * lambda * wb[k];
+      float gh = -2.f * e * wb[k] + 2.f * lambda * hb[k];
+      float Gw = wb[R + k] + gw * gw;
+      float Gh = hb[R + k] + gh * gh;
+      dwb[k] = -lr * gw * __frsqrt_rn(Gw + eps);
+      dwb[R + k] = gw * gw;
+      dhb[k] = -lr * gh * __frsqrt_rn(Gh + eps);
+      dhb[R + k] = gh * gh;
+    }
+  }
+}
+
+// --------------------------------------------------------------- launchers
+
+static inline int grid_for(int64_t n) {
+  int64_t g = n < 1 ? 1 : n;
+  return (int)(g > 16384 ? 16384 : g);
+}
+
+void kge_complex_step_gpu(const float* s, const float* r, const float* o, const float* neg,
+                          float* ds, float* dr, float* do_, float* dneg, float* loss, int B,
+                          int N, int D, float lr, float eps, void* stream) {
+  int dc = D >> 1;
+  dim3 g(grid_for(B)), t(KT);
+  auto st = (hipStream_t)stream;
+#define LAUNCH(KPT) \
+  hipLaunchKernelGGL(k_kge_step<KPT>, g, t, 0, st, s, r, o, neg, ds, dr, do_, dneg, loss, B, N, \
+                     D, lr, eps)
+  if (dc <= KT) LAUNCH(1);
+  else if (dc <= 2 * KT) LAUNCH(2);
+  else if (dc <= 4 * KT) LAUNCH(4);
+  else LAUNCH(8);
+#undef LAUNCH
+}
+void kge_complex_score_gpu(const float* s, const float* r, const float* cand, float* scores,
+                           int B, int E, int D, void* stream) {
+  hipLaunchKernelGGL(k_kge_score, dim3(grid_for((int64_t)B * E)), dim3(KT), 0,
+                     (hipStream_t)stream, s, r, cand, scores, B, E, D);
+}
+void w2v_sgns_step_gpu(const float* ctr, const float* ctx, const float* neg, float* dctr,
+                       float* dctx, float* dneg, float* loss, int B, int N, int D, float lr,
+                       float eps, void* stream) {
+  dim3 g(grid_for(B)), t(KT);
+  auto st = (hipStream_t)stream;
+#define LAUNCH(KPT) \
+  hipLaunchKernelGGL(k_w2v_step<KPT>, g, t, 0, st, ctr, ctx, neg, dctr, dctx, dneg, loss, B, N, \
+                     D, lr, eps)
+  if (D <= KT) LAUNCH(1);
+  else if (D <= 2 * KT) LAUNCH(2);
+  else if (D <= 4 * KT) LAUNCH(4);
+  else LAUNCH(8);
+#undef LAUNCH
+}
+void mf_update_step_gpu(const float* w, const float* h, const float* x, float* dw, float* dh,
+                        float* loss, int B, int R, float lr, float lambda, float eps,
+                        void* stream) {
+  hipLaunchKernelGGL(k_mf_step, dim3(grid_for(B)), dim3(KT), 0, (hipStream_t)stream, w, h, x, dw,
+                     dh, loss, B, R, lr, lambda, eps);
+}
+
+}  // namespace adapm

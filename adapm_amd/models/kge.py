@@ -1,0 +1,239 @@
+"""Knowledge-graph embeddings (ComplEx) on the adaptive parameter manager.
+
+Rebuild of the reference app (reference apps/knowledge_graph_embeddings.cc):
+  - value layout: [embedding(dim) | AdaGrad accumulators(dim)] per key
+    (reference entity_vector_length = 2*dim, :1243-1254)
+  - entity keys [0, E), relation keys [E, E+R)
+  - negatives via PrepareSample/PullSample (reference train() 437-531)
+  - intent look-ahead: announce the keys of future batches
+    `lookahead` batches ahead with per-batch clocks (reference
+    signal_intent_ahead, :1059-1123)
+  - score/grad/AdaGrad: one fused device kernel per batch
+    (kernels_hip.hip k_kge_step; reference :832-858, 415-435)
+  - eval: MRR / Hits@k by scoring candidate sets (reference :544-712)
+  - checkpoint: WaitSync -> Barrier -> rank-0 full pull -> save
+    (reference :327-401)
+"""
+from __future__ import annotations
+
+import dataclasses
+import math
+from typing import Optional
+
+import numpy as np
+import torch
+
+import adapm_amd
+from adapm_amd import _C
+
+
+@dataclasses.dataclass
+class ComplExConfig:
+    num_entities: int = 1_000_000
+    num_relations: int = 1_000
+    dim: int = 512            # embedding dim (complex dim = dim/2)
+    neg_samples: int = 16     # o-side negatives per positive
+    batch_size: int = 4096
+    lr: float = 0.1
+    eps: float = 1e-6
+    lookahead: int = 4        # batches of intent signaled ahead
+    init_scale: float = 0.1
+    seed: int = 42
+
+    @property
+    def num_keys(self) -> int:
+        return self.num_entities + self.num_relations
+
+    @property
+    def row(self) -> int:
+        return 2 * self.dim
+
+
+class ComplEx:
+    def __init__(self, cfg: ComplExConfig, server: "adapm_amd.Server",
+                 worker: "adapm_amd.Worker"):
+        self.cfg = cfg
+        self.server = server
+        self.worker = worker
+        self.dev = server.rt.device
+        self.rank = server.rt.rank
+        self.world = server.rt.world
+        self.rng = np.random.default_rng(cfg.seed + self.rank)
+        self._pending = []
+
+    # ------------------------------------------------------------ init
+
+    def init_embeddings(self):
+        """Each rank initializes the keys it manages (key % world == rank)
+        with set(); AdaGrad accumulators start at 0."""
+        cfg = self.cfg
+        chunk = max(1, 2 ** 22 // cfg.row)  # ~16MB of values per set
+        my_keys = np.arange(self.rank, cfg.num_keys, self.world, dtype=np.int64)
+        g = torch.Generator(device="cpu").manual_seed(cfg.seed)
+        for i in range(0, len(my_keys), chunk):
+            ks = my_keys[i:i + chunk]
+            vals = torch.zeros(len(ks), cfg.row, dtype=torch.float32)
+            vals[:, :cfg.dim] = torch.randn(len(ks), cfg.dim, generator=g) * cfg.init_scale
+            self.worker.set(ks, vals.to(self.dev))
+        self.worker.wait_sync()
+        self.worker.barrier()
+
+    # ------------------------------------------------------------ train
+
+    def keys_of(self, triples: np.ndarray):
+        s = triples[:, 0].astype(np.int64)
+        r = (self.cfg.num_entities + triples[:, 1]).astype(np.int64)
+        o = triples[:, 2].astype(np.int64)
+        return s, r, o
+
+    def signal_intent(self, triples: np.ndarray, start: int, end: int = 0):
+        s, r, o = self.keys_of(triples)
+        self.worker.intent(np.concatenate([s, r, o]), start, end)
+
+    def train_batch(self, triples: np.ndarray, async_push: bool = True) -> float:
+        """One training step over B positive triples. Returns mean loss
+        (local; host-synchronizing — pass loss_out=None style usage for
+        fully async pipelines)."""
+        cfg = self.cfg
+        w = self.worker
+        B = len(triples)
+        s_keys, r_keys, o_keys = self.keys_of(triples)
+
+        # negatives via the sampling manager (reference PrepareSample path)
+        if self.server.sampling is not None:
+            sid = w.prepare_sample(B * cfg.neg_samples, w.current_clock(),
+                                   w.current_clock() + 2)
+            neg_keys = self.server.sampling.pull(w, sid, B * cfg.neg_samples)
+            w.finish_sample(sid)
+        else:
+            neg_keys = self.rng.integers(0, cfg.num_entities, size=B * cfg.neg_samples,
+                                         dtype=np.int64)
+
+        dev = self.dev
+        row = cfg.row
+        opts = dict(dtype=torch.float32, device=dev)
+        s_v = torch.empty(B, row, **opts)
+        r_v = torch.empty(B, row, **opts)
+        o_v = torch.empty(B, row, **opts)
+        n_v = torch.empty(B * cfg.neg_samples, row, **opts)
+        ts = [w.pull(s_keys, s_v, async_=True), w.pull(r_keys, r_v, async_=True),
+              w.pull(o_keys, o_v, async_=True), w.pull(neg_keys, n_v, async_=True)]
+        for t in ts:
+            w.wait(t)
+
+        ds = torch.empty_like(s_v)
+        dr = torch.empty_like(r_v)
+        do = torch.empty_like(o_v)
+        dn = torch.empty_like(n_v)
+        loss = torch.empty(B, dtype=torch.float32, device=dev)
+        _C.kge_complex_step(s_v, r_v, o_v, n_v, ds, dr, do, dn, loss,
+                            cfg.neg_samples, cfg.dim, cfg.lr, cfg.eps)
+
+        pts = [w.push(s_keys, ds, async_=True), w.push(r_keys, dr, async_=True),
+               w.push(o_keys, do, async_=True), w.push(neg_keys, dn, async_=True)]
+        self._pending.extend(t for t in pts if t != -1)
+        if not async_push:
+            for t in pts:
+                w.wait(t)
+        # bounded async: cap outstanding pushes
+        while len(self._pending) > 64:
+            w.wait(self._pending.pop(0))
+        return float(loss.mean().item())
+
+    def drain(self):
+        for t in self._pending:
+            self.worker.wait(t)
+        self._pending.clear()
+
+    # ------------------------------------------------------------ eval
+
+    @torch.no_grad()
+    def evaluate(self, triples: np.ndarray, num_candidates: int = 1000,
+                 hits_at=(1, 3, 10)) -> dict:
+        """Rank the true object among `num_candidates` random candidates
+        (+ the true one). Aggregated over ranks via allreduce (reference
+        eval: knowledge_graph_embeddings.cc:544-712)."""
+        cfg = self.cfg
+        w = self.worker
+        B = len(triples)
+        s_keys, r_keys, o_keys = self.keys_of(triples)
+        cand = self.rng.integers(0, cfg.num_entities, size=num_candidates, dtype=np.int64)
+
+        opts = dict(dtype=torch.float32, device=self.dev)
+        s_v = torch.empty(B, cfg.row, **opts)
+        r_v = torch.empty(B, cfg.row, **opts)
+        o_v = torch.empty(B, cfg.row, **opts)
+        c_v = torch.empty(num_candidates, cfg.row, **opts)
+        w.pull(s_keys, s_v)
+        w.pull(r_keys, r_v)
+        w.pull(o_keys, o_v)
+        w.pull(cand, c_v)
+
+        cand_scores = torch.empty(B, num_candidates, **opts)
+        _C.kge_complex_score(s_v, r_v, c_v, cand_scores, cfg.dim)
+        true_scores = torch.empty(B, 1, **opts)
+        # score the true o: reuse the kernel with E=1 per row via batched diag
+        # (cheap path: einsum on device)
+        dc = cfg.dim // 2
+        sr_re = s_v[:, :dc] * r_v[:, :dc] - s_v[:, dc:cfg.dim] * r_v[:, dc:cfg.dim]
+        sr_im = s_v[:, dc:cfg.dim] * r_v[:, :dc] + s_v[:, :dc] * r_v[:, dc:cfg.dim]
+        true_scores[:, 0] = (sr_re * o_v[:, :dc] + sr_im * o_v[:, dc:cfg.dim]).sum(1)
+
+        rank = 1 + (cand_scores > true_scores).sum(1).float()
+        out = {
+            "mrr": float((1.0 / rank).mean().item()),
+            "mr": float(rank.mean().item()),
+            "n": float(B),
+        }
+        for h in hits_at:
+            out[f"hits@{h}"] = float((rank <= h).float().mean().item())
+        # aggregate across ranks (weighted by n)
+        if self.world > 1:
+            vec = torch.tensor([out["n"]] + [out[k] * out["n"] for k in sorted(out) if k != "n"])
+            vec = w.allreduce(vec)
+            names = [k for k in sorted(out) if k != "n"]
+            out = {k: float(vec[1 + i] / vec[0]) for i, k in enumerate(names)}
+            out["n"] = float(vec[0])
+        return out
+
+    # ------------------------------------------------------------ checkpoint
+
+    def save_checkpoint(self, path: str, chunk_keys: int = 65536):
+        """rank 0 pulls the full model (incl. AdaGrad state) and writes an
+        .npz (reference pull_full_model + write_checkpoint, :209-231,327-401)."""
+        self.drain()
+        self.worker.wait_sync()
+        self.worker.barrier()
+        if self.rank == 0:
+            cfg = self.cfg
+            out = np.empty((cfg.num_keys, cfg.row), dtype=np.float32)
+            for i in range(0, cfg.num_keys, chunk_keys):
+                ks = np.arange(i, min(i + chunk_keys, cfg.num_keys), dtype=np.int64)
+                buf = np.zeros((len(ks), cfg.row), dtype=np.float32)
+                self.worker.pull(ks, buf)
+                out[i:i + len(ks)] = buf
+            np.savez(path, values=out,
+                     num_entities=cfg.num_entities, num_relations=cfg.num_relations,
+                     dim=cfg.dim)
+        self.worker.barrier()
+
+    def load_checkpoint(self, path: str, chunk_keys: int = 65536):
+        """rank 0 reads and set()s the full model."""
+        if self.rank == 0:
+            data = np.load(path)
+            vals = data["values"]
+            for i in range(0, len(vals), chunk_keys):
+                ks = np.arange(i, min(i + chunk_keys, len(vals)), dtype=np.int64)
+                self.worker.set(ks, np.ascontiguousarray(vals[i:i + len(ks)]))
+        self.worker.wait_sync()
+        self.worker.barrier()
+
+
+def make_synthetic_triples(n: int, num_entities: int, num_relations: int,
+                           seed: int = 0) -> np.ndarray:
+    rng = np.random.default_rng(seed)
+    return np.stack([
+        rng.integers(0, num_entities, size=n),
+        rng.integers(0, num_relations, size=n),
+        rng.integers(0, num_entities, size=n),
+    ], axis=1).astype(np.int64)

@@ -1,0 +1,211 @@
+"""App-kernel numerics vs plain PyTorch fp32 references (autograd).
+
+CPU tests run everywhere; the @gpu twins run the HIP kernels on an MI355X
+and compare against the same torch references.
+"""
+import numpy as np
+import pytest
+import torch
+
+from adapm_amd import _C
+
+
+# ------------------------------------------------------------ references
+
+def torch_kge_ref(s, r, o, neg, N, D, lr, eps):
+    """ComplEx logistic loss with o-side negatives; returns expected push
+    deltas [delta_emb | grad^2] for s, r, o, neg rows."""
+    B = s.shape[0]
+    dc = D // 2
+
+    def leaf(x):
+        t = x[:, :D].clone().detach().requires_grad_(True)
+        return t
+
+    se, re_, oe, ne = leaf(s), leaf(r), leaf(o), leaf(neg)
+
+    def score(sv, rv, ov):
+        sre, sim = sv[:, :dc], sv[:, dc:]
+        rre, rim = rv[:, :dc], rv[:, dc:]
+        ore, oim = ov[:, :dc], ov[:, dc:]
+        ure = sre * rre - sim * rim
+        uim = sim * rre + sre * rim
+        return (ure * ore + uim * oim).sum(1)
+
+    pos = score(se, re_, oe)
+    loss = torch.nn.functional.softplus(-pos).sum()
+    nv = ne.view(B, N, D)
+    for j in range(N):
+        neg_score = score(se, re_, nv[:, j, :])
+        loss = loss + torch.nn.functional.softplus(neg_score).sum()
+    loss.backward()
+
+    def delta(x_full, leaf_t):
+        g = leaf_t.grad
+        G = x_full[:, D:] + g * g
+        d = torch.cat([-lr * g / torch.sqrt(G + eps), g * g], dim=1)
+        return d
+
+    return delta(s, se), delta(r, re_), delta(o, oe), delta(neg, ne)
+
+
+def run_kge(device, B=4, N=3, D=16, tol=2e-4):
+    g = torch.Generator().manual_seed(0)
+    mk = lambda n: torch.cat([torch.randn(n, D, generator=g) * 0.3,
+                              torch.rand(n, D, generator=g) * 0.1], dim=1)
+    s, r, o, neg = mk(B), mk(B), mk(B), mk(B * N)
+    lr, eps = 0.05, 1e-6
+
+    sd, rd, od, nd = [t.to(device) for t in (s, r, o, neg)]
+    ds, dr, do, dn = [torch.empty_like(t) for t in (sd, rd, od, nd)]
+    loss = torch.empty(B, dtype=torch.float32, device=device)
+    _C.kge_complex_step(sd, rd, od, nd, ds, dr, do, dn, loss, N, D, lr, eps)
+
+    eds, edr, edo, edn = torch_kge_ref(s, r, o, neg, N, D, lr, eps)
+    for got, exp, name in [(ds, eds, "ds"), (dr, edr, "dr"), (do, edo, "do"), (dn, edn, "dn")]:
+        got = got.cpu()
+        err = (got - exp).abs().max().item()
+        assert err < tol, f"{name} max err {err}"
+    assert torch.isfinite(loss).all()
+
+
+def test_kge_step_cpu():
+    run_kge("cpu")
+
+
+def test_kge_step_cpu_large_dim():
+    run_kge("cpu", B=2, N=2, D=512)
+
+
+@pytest.mark.gpu
+def test_kge_step_gpu():
+    run_kge("cuda:0")
+
+
+@pytest.mark.gpu
+def test_kge_step_gpu_dim512():
+    run_kge("cuda:0", B=8, N=8, D=512)
+
+
+@pytest.mark.gpu
+def test_kge_step_gpu_dim1024():
+    run_kge("cuda:0", B=4, N=4, D=1024)
+
+
+# ------------------------------------------------------------ scoring
+
+def run_kge_score(device, B=3, E=5, D=32):
+    g = torch.Generator().manual_seed(1)
+    mk = lambda n: torch.cat([torch.randn(n, D, generator=g) * 0.3,
+                              torch.zeros(n, D)], dim=1)
+    s, r, cand = mk(B), mk(B), mk(E)
+    dc = D // 2
+    sre, sim = s[:, :dc], s[:, dc:D]
+    rre, rim = r[:, :dc], r[:, dc:D]
+    ure = sre * rre - sim * rim
+    uim = sim * rre + sre * rim
+    exp = ure @ cand[:, :dc].T + uim @ cand[:, dc:D].T
+
+    sd, rd, cd = s.to(device), r.to(device), cand.to(device)
+    scores = torch.empty(B, E, dtype=torch.float32, device=device)
+    _C.kge_complex_score(sd, rd, cd, scores, D)
+    assert (scores.cpu() - exp).abs().max().item() < 1e-4
+
+
+def test_kge_score_cpu():
+    run_kge_score("cpu")
+
+
+@pytest.mark.gpu
+def test_kge_score_gpu():
+    run_kge_score("cuda:0", B=8, E=64, D=512)
+
+
+# ------------------------------------------------------------ word2vec
+
+def torch_w2v_ref(ctr, ctx, neg, N, D, lr, eps):
+    B = ctr.shape[0]
+    ce = ctr[:, :D].clone().detach().requires_grad_(True)
+    xe = ctx[:, :D].clone().detach().requires_grad_(True)
+    ne = neg[:, :D].clone().detach().requires_grad_(True)
+    loss = torch.nn.functional.softplus(-(ce * xe).sum(1)).sum()
+    nv = ne.view(B, N, D)
+    for j in range(N):
+        loss = loss + torch.nn.functional.softplus((ce * nv[:, j, :]).sum(1)).sum()
+    loss.backward()
+
+    def delta(full, leaf):
+        g = leaf.grad
+        G = full[:, D:] + g * g
+        return torch.cat([-lr * g / torch.sqrt(G + eps), g * g], dim=1)
+
+    return delta(ctr, ce), delta(ctx, xe), delta(neg, ne)
+
+
+def run_w2v(device, B=4, N=3, D=24, tol=2e-4):
+    g = torch.Generator().manual_seed(2)
+    mk = lambda n: torch.cat([torch.randn(n, D, generator=g) * 0.3,
+                              torch.rand(n, D, generator=g) * 0.1], dim=1)
+    ctr, ctx, neg = mk(B), mk(B), mk(B * N)
+    lr, eps = 0.025, 1e-6
+    cd, xd, nd = [t.to(device) for t in (ctr, ctx, neg)]
+    dc, dx, dn = [torch.empty_like(t) for t in (cd, xd, nd)]
+    loss = torch.empty(B, dtype=torch.float32, device=device)
+    _C.w2v_sgns_step(cd, xd, nd, dc, dx, dn, loss, N, D, lr, eps)
+    edc, edx, edn = torch_w2v_ref(ctr, ctx, neg, N, D, lr, eps)
+    for got, exp, name in [(dc, edc, "dctr"), (dx, edx, "dctx"), (dn, edn, "dneg")]:
+        err = (got.cpu() - exp).abs().max().item()
+        assert err < tol, f"{name} max err {err}"
+
+
+def test_w2v_step_cpu():
+    run_w2v("cpu")
+
+
+@pytest.mark.gpu
+def test_w2v_step_gpu():
+    run_w2v("cuda:0", B=8, N=5, D=300)
+
+
+# ------------------------------------------------------------ MF
+
+def torch_mf_ref(w, h, x, R, lr, lam, eps):
+    we = w[:, :R].clone().detach().requires_grad_(True)
+    he = h[:, :R].clone().detach().requires_grad_(True)
+    e = x - (we * he).sum(1)
+    loss = (e * e).sum() + lam * ((we * we).sum() + (he * he).sum())
+    loss.backward()
+
+    def delta(full, leaf):
+        g = leaf.grad
+        G = full[:, R:] + g * g
+        return torch.cat([-lr * g / torch.sqrt(G + eps), g * g], dim=1)
+
+    return delta(w, we), delta(h, he)
+
+
+def run_mf(device, B=6, R=32, tol=2e-4):
+    g = torch.Generator().manual_seed(3)
+    mk = lambda n: torch.cat([torch.randn(n, R, generator=g) * 0.3,
+                              torch.rand(n, R, generator=g) * 0.1], dim=1)
+    w, h = mk(B), mk(B)
+    x = torch.randn(B, generator=g)
+    lr, lam, eps = 0.01, 0.05, 1e-6
+    wd, hd, xd = w.to(device), h.to(device), x.to(device)
+    dw, dh = torch.empty_like(wd), torch.empty_like(hd)
+    loss = torch.empty(B, dtype=torch.float32, device=device)
+    _C.mf_update_step(wd, hd, xd, dw, dh, loss, R, lr, lam, eps)
+    edw, edh = torch_mf_ref(w, h, x, R, lr, lam, eps)
+    assert (dw.cpu() - edw).abs().max().item() < tol
+    assert (dh.cpu() - edh).abs().max().item() < tol
+    exp_loss = (x - (w[:, :R] * h[:, :R]).sum(1)) ** 2
+    assert (loss.cpu() - exp_loss).abs().max().item() < 1e-4
+
+
+def test_mf_step_cpu():
+    run_mf("cpu")
+
+
+@pytest.mark.gpu
+def test_mf_step_gpu():
+    run_mf("cuda:0", B=16, R=128)
