@@ -18,10 +18,15 @@ from __future__ import annotations
 
 import dataclasses
 import math
+import os
+import time
+from collections import defaultdict
 from typing import Optional
 
 import numpy as np
 import torch
+
+_PHASE_TIMING = os.environ.get("ADAPM_PHASE_TIMING", "0") == "1"
 
 import adapm_amd
 from adapm_amd import _C
@@ -60,6 +65,16 @@ class ComplEx:
         self.world = server.rt.world
         self.rng = np.random.default_rng(cfg.seed + self.rank)
         self._pending = []
+        self.phase_times = defaultdict(float)
+
+    def _ph(self, name, t0):
+        if _PHASE_TIMING:
+            if self.dev.type == "cuda":
+                torch.cuda.synchronize()
+            t = time.perf_counter()
+            self.phase_times[name] += t - t0
+            return t
+        return t0
 
     # ------------------------------------------------------------ init
 
@@ -67,14 +82,13 @@ class ComplEx:
         """Each rank initializes the keys it manages (key % world == rank)
         with set(); AdaGrad accumulators start at 0."""
         cfg = self.cfg
-        chunk = max(1, 2 ** 22 // cfg.row)  # ~16MB of values per set
+        chunk = max(1, 2 ** 25 // cfg.row)  # ~128MB of values per set
         my_keys = np.arange(self.rank, cfg.num_keys, self.world, dtype=np.int64)
-        g = torch.Generator(device="cpu").manual_seed(cfg.seed)
         for i in range(0, len(my_keys), chunk):
             ks = my_keys[i:i + chunk]
-            vals = torch.zeros(len(ks), cfg.row, dtype=torch.float32)
-            vals[:, :cfg.dim] = torch.randn(len(ks), cfg.dim, generator=g) * cfg.init_scale
-            self.worker.set(ks, vals.to(self.dev))
+            vals = torch.zeros(len(ks), cfg.row, dtype=torch.float32, device=self.dev)
+            vals[:, :cfg.dim].normal_(0.0, cfg.init_scale)
+            self.worker.set(ks, vals)
         self.worker.wait_sync()
         self.worker.barrier()
 
@@ -90,13 +104,14 @@ class ComplEx:
         s, r, o = self.keys_of(triples)
         self.worker.intent(np.concatenate([s, r, o]), start, end)
 
-    def train_batch(self, triples: np.ndarray, async_push: bool = True) -> float:
+    def train_batch(self, triples: np.ndarray, async_push: bool = True,
+                    sync_loss: bool = True):
         """One training step over B positive triples. Returns mean loss
-        (local; host-synchronizing — pass loss_out=None style usage for
-        fully async pipelines)."""
+        (float if sync_loss else a device tensor, no host sync)."""
         cfg = self.cfg
         w = self.worker
         B = len(triples)
+        t0 = time.perf_counter() if _PHASE_TIMING else 0.0
         s_keys, r_keys, o_keys = self.keys_of(triples)
 
         # negatives via the sampling manager (reference PrepareSample path)
@@ -108,37 +123,47 @@ class ComplEx:
         else:
             neg_keys = self.rng.integers(0, cfg.num_entities, size=B * cfg.neg_samples,
                                          dtype=np.int64)
+        t0 = self._ph("sample", t0)
 
+        # one fused pull of all rows: [s | r | o | neg]
         dev = self.dev
         row = cfg.row
-        opts = dict(dtype=torch.float32, device=dev)
-        s_v = torch.empty(B, row, **opts)
-        r_v = torch.empty(B, row, **opts)
-        o_v = torch.empty(B, row, **opts)
-        n_v = torch.empty(B * cfg.neg_samples, row, **opts)
-        ts = [w.pull(s_keys, s_v, async_=True), w.pull(r_keys, r_v, async_=True),
-              w.pull(o_keys, o_v, async_=True), w.pull(neg_keys, n_v, async_=True)]
-        for t in ts:
-            w.wait(t)
+        NN = B * cfg.neg_samples
+        all_keys = np.concatenate([s_keys, r_keys, o_keys, neg_keys])
+        all_v = torch.empty((3 * B + NN) * row, dtype=torch.float32, device=dev)
+        ts = w.pull(all_keys, all_v, async_=True)
+        w.wait(ts)
+        t0 = self._ph("pull", t0)
 
-        ds = torch.empty_like(s_v)
-        dr = torch.empty_like(r_v)
-        do = torch.empty_like(o_v)
-        dn = torch.empty_like(n_v)
+        s_v = all_v[: B * row].view(B, row)
+        r_v = all_v[B * row: 2 * B * row].view(B, row)
+        o_v = all_v[2 * B * row: 3 * B * row].view(B, row)
+        n_v = all_v[3 * B * row:].view(NN, row)
+
+        all_d = torch.empty_like(all_v)
+        ds = all_d[: B * row].view(B, row)
+        dr = all_d[B * row: 2 * B * row].view(B, row)
+        do = all_d[2 * B * row: 3 * B * row].view(B, row)
+        dn = all_d[3 * B * row:].view(NN, row)
         loss = torch.empty(B, dtype=torch.float32, device=dev)
         _C.kge_complex_step(s_v, r_v, o_v, n_v, ds, dr, do, dn, loss,
                             cfg.neg_samples, cfg.dim, cfg.lr, cfg.eps)
+        t0 = self._ph("kernel", t0)
 
-        pts = [w.push(s_keys, ds, async_=True), w.push(r_keys, dr, async_=True),
-               w.push(o_keys, do, async_=True), w.push(neg_keys, dn, async_=True)]
-        self._pending.extend(t for t in pts if t != -1)
+        pt = w.push(all_keys, all_d, async_=True)
+        if pt != -1:
+            self._pending.append(pt)
         if not async_push:
-            for t in pts:
-                w.wait(t)
+            w.wait(pt)
         # bounded async: cap outstanding pushes
         while len(self._pending) > 64:
             w.wait(self._pending.pop(0))
-        return float(loss.mean().item())
+        t0 = self._ph("push", t0)
+        if sync_loss:
+            out = float(loss.mean().item())
+            self._ph("loss_sync", t0)
+            return out
+        return loss
 
     def drain(self):
         for t in self._pending:

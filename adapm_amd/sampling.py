@@ -155,8 +155,11 @@ class SamplingManager:
         """Draw candidates, substitute each with the next locally-present
         key (reference sampling.h Local scheme, 366-525)."""
         cands = self.dist.draw(n)
-        keys, _checks = self.server.scan_local(torch.from_numpy(cands), self.lo, self.hi)
-        keys = keys.numpy()
+        if self.server.world() == 1:
+            keys = cands  # single rank: every key is local, no scan needed
+        else:
+            keys, _checks = self.server.scan_local(torch.from_numpy(cands), self.lo, self.hi)
+            keys = keys.numpy()
         if not self.with_replacement:
             out = []
             attempts = 0

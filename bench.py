@@ -80,11 +80,11 @@ def main():
     total_steps = args.warmup + args.steps
     batches = [make_batch(i) for i in range(total_steps + args.lookahead)]
 
-    def run_step(i):
+    def run_step(i, sync_loss=False):
         model.signal_intent(batches[i + args.lookahead],
                             worker.current_clock() + args.lookahead,
                             worker.current_clock() + args.lookahead + 2)
-        loss = model.train_batch(batches[i])
+        loss = model.train_batch(batches[i], sync_loss=sync_loss)
         worker.advance_clock()
         return loss
 
@@ -100,7 +100,7 @@ def main():
     t_start = time.perf_counter()
     last_loss = 0.0
     for i in range(args.warmup, total_steps):
-        last_loss = run_step(i)
+        last_loss = run_step(i, sync_loss=(i == total_steps - 1))
     model.drain()
     if is_cuda:
         torch.cuda.synchronize()
@@ -119,6 +119,9 @@ def main():
     ms_per_step = 1000.0 * t_elapsed / args.steps
 
     st = server.stats()
+    if rank == 0 and model.phase_times:
+        print("phase times (s over timed+warmup):",
+              dict(sorted(model.phase_times.items())), file=sys.stderr, flush=True)
     if rank == 0:
         out = {
             "metric": "pull_push_ops_per_s",
