@@ -40,6 +40,7 @@
 #include <memory>
 #include <mutex>
 #include <thread>
+#include <chrono>
 #include <unordered_map>
 #include <unordered_set>
 #include <vector>
@@ -368,6 +369,8 @@ class Server {
     std::vector<int64_t> out_off(n);
     std::vector<int32_t> out_len(n);
     int64_t cum = 0;
+    auto tick = [&]() { return cpp_timing_ ? std::chrono::steady_clock::now().time_since_epoch().count() : 0; };
+    int64_t tp0 = tick();
     {
       InflightGuard g(this);
       if (layout_identity_.load(std::memory_order_acquire) && uniform_len_ >= 0) {
@@ -407,7 +410,9 @@ class Server {
         }
       }
       stat_pull_keys_ += n;
+      if (cpp_timing_) { t_pass_ += tick() - tp0; tp0 = tick(); }
       run_gather(local, vals_dev);
+      if (cpp_timing_) { t_launch_ += tick() - tp0; t_calls_++; }
     }
     stat_pulls_ += 1;
 
@@ -1378,6 +1383,9 @@ class Server {
     int64_t rounds = 0;
     for (auto& c : channels_) rounds += c.rounds.load();
     d["sync_rounds"] = rounds;
+    d["t_pass_ms"] = t_pass_.load() / 1e6;
+    d["t_launch_ms"] = t_launch_.load() / 1e6;
+    d["t_calls"] = t_calls_.load();
     return d;
   }
 
@@ -1425,6 +1433,11 @@ class Server {
 
   std::atomic<int> inflight_{0};
   std::atomic<int64_t> next_ts_{1};
+ public:
+  // env ADAPM_CPP_TIMING=1: nanosecond accounting of the worker-op host path
+  std::atomic<int64_t> t_pass_{0}, t_todev_{0}, t_launch_{0}, t_calls_{0};
+  bool cpp_timing_ = getenv("ADAPM_CPP_TIMING") != nullptr;
+ private:
   std::mutex tickets_mu_;
   std::condition_variable tickets_cv_;
   std::unordered_map<int64_t, std::unique_ptr<Ticket>> tickets_;

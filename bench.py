@@ -88,10 +88,17 @@ def main():
         worker.advance_clock()
         return loss
 
+    prof = None
+    if os.environ.get("ADAPM_PROFILE") == "1":
+        import cProfile
+        prof = cProfile.Profile()
+
     # warmup
     for i in range(args.warmup):
         run_step(i)
     model.drain()
+    if prof is not None:
+        prof.enable()
 
     # timed region
     worker.barrier()
@@ -105,6 +112,10 @@ def main():
     if is_cuda:
         torch.cuda.synchronize()
     t_elapsed = time.perf_counter() - t_start
+    if prof is not None:
+        prof.disable()
+        import pstats
+        pstats.Stats(prof, stream=sys.stderr).sort_stats("cumulative").print_stats(25)
     worker.barrier()
 
     # MAX elapsed over ranks
