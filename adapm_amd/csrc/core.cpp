@@ -270,6 +270,7 @@ class Server {
   };
 
   DevBatch to_dev(const HostBatch& hb, const std::vector<int64_t>* aux = nullptr) {
+    int64_t t0 = cpp_timing_ ? std::chrono::steady_clock::now().time_since_epoch().count() : 0;
     DevBatch d;
     auto i64 = torch::TensorOptions().dtype(torch::kInt64);
     auto i32 = torch::TensorOptions().dtype(torch::kInt32);
@@ -287,6 +288,8 @@ class Server {
     d.b.dst_off = d.dst_t.data_ptr<int64_t>();
     d.b.lens = d.len_t.data_ptr<int32_t>();
     d.b.n = (int)hb.size();
+    if (cpp_timing_)
+      t_todev_ += std::chrono::steady_clock::now().time_since_epoch().count() - t0;
     return d;
   }
 
@@ -344,6 +347,37 @@ class Server {
     }
   }
 
+  // identity-layout direct ops: offsets derived from keys on the device
+  KeyBatch key_batch(const torch::Tensor& keys_dev) {
+    KeyBatch b;
+    b.keys = keys_dev.data_ptr<int64_t>();
+    b.n = (int)keys_dev.numel();
+    b.len = uniform_len_;
+    b.plen = Slab::padded(uniform_len_);
+    b.world = world_;
+    b.rank = rank_;
+    return b;
+  }
+  void run_gather_keys(const torch::Tensor& keys_cpu, torch::Tensor out) {
+    if (dev_.is_cuda()) {
+      auto kd = keys_cpu.to(dev_, /*non_blocking=*/true);
+      ops_gather_keys_gpu(slab_.data, key_batch(kd), out.data_ptr<float>(), current_stream(dev_));
+    } else {
+      std::lock_guard<std::mutex> g(cpu_val_mu_);
+      ops_gather_keys_cpu(slab_.data, key_batch(keys_cpu), out.data_ptr<float>());
+    }
+  }
+  void run_scatter_keys(const torch::Tensor& keys_cpu, const torch::Tensor& in, bool set) {
+    if (dev_.is_cuda()) {
+      auto kd = keys_cpu.to(dev_, /*non_blocking=*/true);
+      ops_scatter_keys_gpu(slab_.data, key_batch(kd), in.data_ptr<float>(), set,
+                           current_stream(dev_));
+    } else {
+      std::lock_guard<std::mutex> g(cpu_val_mu_);
+      ops_scatter_keys_cpu(slab_.data, key_batch(keys_cpu), in.data_ptr<float>(), set);
+    }
+  }
+
   // ------------------------------------------------ worker API
 
   // Pull: local fast path returns -1 with the gather already enqueued on
@@ -374,21 +408,26 @@ class Server {
     {
       InflightGuard g(this);
       if (layout_identity_.load(std::memory_order_acquire) && uniform_len_ >= 0) {
-        // fast path: offsets are pure arithmetic (see layout_identity_)
-        const int64_t plen = Slab::padded(uniform_len_);
+        // fast path: zero host per-key work — the device kernel derives
+        // every offset from the key (see layout_identity_). The host only
+        // scans for non-owned keys (pure arithmetic, nothing when world=1).
         const int32_t l = uniform_len_;
-        for (int64_t i = 0; i < n; ++i) {
-          Key k = kp[i];
-          out_off[i] = cum;
-          out_len[i] = l;
-          if (k % world_ == rank_) {
-            local.add((k / world_) * plen, cum, l);
-          } else {
-            remote.push_back({k, i});
+        if (world_ > 1) {
+          for (int64_t i = 0; i < n; ++i) {
+            if (kp[i] % world_ != rank_) remote.push_back({kp[i], i});
           }
-          cum += l;
         }
+        if (!remote.empty()) {
+          for (int64_t i = 0; i < n; ++i) {
+            out_off[i] = i * (int64_t)l;
+            out_len[i] = l;
+          }
+        }
+        if (cpp_timing_) { t_pass_ += tick() - tp0; tp0 = tick(); }
+        run_gather_keys(keys, vals_dev);
+        if (cpp_timing_) { t_launch_ += tick() - tp0; t_calls_++; }
         stat_pull_local_ += n - (int64_t)remote.size();
+        stat_pull_keys_ += n;
       } else {
         for (int64_t i = 0; i < n; ++i) {
           Key k = kp[i];
@@ -408,11 +447,11 @@ class Server {
           }
           cum += l;
         }
+        stat_pull_keys_ += n;
+        if (cpp_timing_) { t_pass_ += tick() - tp0; tp0 = tick(); }
+        run_gather(local, vals_dev);
+        if (cpp_timing_) { t_launch_ += tick() - tp0; t_calls_++; }
       }
-      stat_pull_keys_ += n;
-      if (cpp_timing_) { t_pass_ += tick() - tp0; tp0 = tick(); }
-      run_gather(local, vals_dev);
-      if (cpp_timing_) { t_launch_ += tick() - tp0; t_calls_++; }
     }
     stat_pulls_ += 1;
 
@@ -446,11 +485,14 @@ class Server {
     (void)wid;
     check_keys(keys);
     TORCH_CHECK(vals.scalar_type() == torch::kFloat32, "vals must be float32");
+    auto tick = [&]() { return cpp_timing_ ? std::chrono::steady_clock::now().time_since_epoch().count() : 0; };
+    int64_t tp0 = tick();
     torch::Tensor vals_dev = vals.device() == dev_ ? vals : vals.to(dev_);
     if (!vals_dev.is_contiguous()) vals_dev = vals_dev.contiguous();
     torch::Tensor flat = vals_dev.view({-1});
     int64_t n = keys.numel();
     const int64_t* kp = keys.data_ptr<int64_t>();
+    if (cpp_timing_) { t_misc_ += tick() - tp0; }
 
     HostBatch merge, assign;
     struct Remote {
@@ -463,22 +505,22 @@ class Server {
     {
       InflightGuard g(this);
       if (layout_identity_.load(std::memory_order_acquire) && uniform_len_ >= 0) {
-        const int64_t plen = Slab::padded(uniform_len_);
+        // fast path: device kernel derives offsets from keys; host only
+        // scans for non-owned keys (pure arithmetic, nothing at world=1).
+        // Version bumps are skipped here: handle_owner_delta clears
+        // layout_identity_ when the first replica of one of our keys is
+        // granted, so while the flag holds versions are unobserved.
         const int32_t l = uniform_len_;
-        for (int64_t i = 0; i < n; ++i) {
-          Key k = kp[i];
-          if (k % world_ == rank_) {
-            (set_mode ? assign : merge).add((k / world_) * plen, cum, l);
-          } else {
-            remote.push_back({k, cum, l});
+        if (world_ > 1) {
+          for (int64_t i = 0; i < n; ++i) {
+            if (kp[i] % world_ != rank_) remote.push_back({kp[i], i * (int64_t)l, l});
           }
-          cum += l;
         }
+        if (cpp_timing_) { int64_t t1 = tick(); t_pass_ += t1 - tp0; tp0 = t1; }
+        run_scatter_keys(keys, flat, set_mode);
+        if (cpp_timing_) { t_launch_ += tick() - tp0; t_calls_++; }
         stat_push_local_ += n - (int64_t)remote.size();
-        // version bumps are skipped on the fast path: handle_owner_delta
-        // clears layout_identity_ when the first replica of one of our
-        // keys is granted, so while the flag holds no replica of our keys
-        // exists anywhere and versions are unobserved.
+        stat_push_keys_ += n;
       } else {
         for (int64_t i = 0; i < n; ++i) {
           Key k = kp[i];
@@ -501,10 +543,12 @@ class Server {
           }
           cum += l;
         }
+        stat_push_keys_ += n;
+        if (cpp_timing_) { int64_t t1 = tick(); t_pass_ += t1 - tp0; tp0 = t1; }
+        run_scatter(merge, flat, false);
+        run_scatter(assign, flat, true);
+        if (cpp_timing_) { t_launch_ += tick() - tp0; t_calls_++; }
       }
-      stat_push_keys_ += n;
-      run_scatter(merge, flat, false);
-      run_scatter(assign, flat, true);
     }
     stat_pushes_ += 1;
 
@@ -1384,7 +1428,9 @@ class Server {
     for (auto& c : channels_) rounds += c.rounds.load();
     d["sync_rounds"] = rounds;
     d["t_pass_ms"] = t_pass_.load() / 1e6;
+    d["t_todev_ms"] = t_todev_.load() / 1e6;
     d["t_launch_ms"] = t_launch_.load() / 1e6;
+    d["t_misc_ms"] = t_misc_.load() / 1e6;
     d["t_calls"] = t_calls_.load();
     return d;
   }
@@ -1435,7 +1481,7 @@ class Server {
   std::atomic<int64_t> next_ts_{1};
  public:
   // env ADAPM_CPP_TIMING=1: nanosecond accounting of the worker-op host path
-  std::atomic<int64_t> t_pass_{0}, t_todev_{0}, t_launch_{0}, t_calls_{0};
+  std::atomic<int64_t> t_pass_{0}, t_todev_{0}, t_launch_{0}, t_calls_{0}, t_misc_{0};
   bool cpp_timing_ = getenv("ADAPM_CPP_TIMING") != nullptr;
  private:
   std::mutex tickets_mu_;

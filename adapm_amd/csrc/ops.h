@@ -47,6 +47,23 @@ void ops_refresh_cpu(float* slab, const OpsBatch& b, const int64_t* sync_off, co
 void ops_zero_gpu(float* slab, const OpsBatch& b, void* stream);
 void ops_zero_cpu(float* slab, const OpsBatch& b);
 
+// Identity-layout direct ops: key k (owned: k % world == rank) lives at
+// slab offset (k / world) * plen; out/in row i sits at i * len. Keys not
+// owned here are skipped (the host enqueues them as remote ops). This is
+// the zero-host-work fast path: keys go H2D and the kernel derives every
+// offset itself.
+struct KeyBatch {
+  const int64_t* keys;  // device (gpu) / host (cpu)
+  int n;
+  int32_t len;    // uniform value length (floats)
+  int32_t plen;   // padded slot length
+  int world, rank;
+};
+void ops_gather_keys_gpu(const float* slab, const KeyBatch& b, float* out, void* stream);
+void ops_gather_keys_cpu(const float* slab, const KeyBatch& b, float* out);
+void ops_scatter_keys_gpu(float* slab, const KeyBatch& b, const float* in, bool set, void* stream);
+void ops_scatter_keys_cpu(float* slab, const KeyBatch& b, const float* in, bool set);
+
 bool hip_available();
 
 }  // namespace adapm

@@ -57,6 +57,29 @@ void ops_refresh_cpu(float* slab, const OpsBatch& b, const int64_t* sync_off, co
   }
 }
 
+void ops_gather_keys_cpu(const float* slab, const KeyBatch& b, float* out) {
+  for (int i = 0; i < b.n; ++i) {
+    int64_t k = b.keys[i];
+    if ((int)(k % b.world) != b.rank) continue;
+    std::memcpy(out + (int64_t)i * b.len, slab + (k / b.world) * (int64_t)b.plen,
+                sizeof(float) * b.len);
+  }
+}
+
+void ops_scatter_keys_cpu(float* slab, const KeyBatch& b, const float* in, bool set) {
+  for (int i = 0; i < b.n; ++i) {
+    int64_t k = b.keys[i];
+    if ((int)(k % b.world) != b.rank) continue;
+    float* s = slab + (k / b.world) * (int64_t)b.plen;
+    const float* d = in + (int64_t)i * b.len;
+    if (set) {
+      std::memcpy(s, d, sizeof(float) * b.len);
+    } else {
+      for (int e = 0; e < b.len; ++e) s[e] += d[e];
+    }
+  }
+}
+
 void ops_zero_cpu(float* slab, const OpsBatch& b) {
   for (int i = 0; i < b.n; ++i) {
     int64_t d = b.dst_off[i];

@@ -108,6 +108,54 @@ __global__ void k_zero(float* __restrict__ slab, const int64_t* __restrict__ dst
   }
 }
 
+__global__ void k_gather_keys(const float* __restrict__ slab, const int64_t* __restrict__ keys,
+                              int n, int32_t len, int32_t plen, int world, int rank,
+                              float* __restrict__ out) {
+  for (int i = blockIdx.x; i < n; i += gridDim.x) {
+    int64_t k = keys[i];
+    if ((int)(k % world) != rank) continue;
+    const float4* sp = reinterpret_cast<const float4*>(slab + (k / world) * (int64_t)plen);
+    float4* dp = reinterpret_cast<float4*>(out + (int64_t)i * len);
+    if ((len & 3) == 0) {
+      for (int e = threadIdx.x; e < (len >> 2); e += THREADS) dp[e] = sp[e];
+    } else {
+      const float* s = reinterpret_cast<const float*>(sp);
+      float* d = reinterpret_cast<float*>(dp);
+      for (int e = threadIdx.x; e < len; e += THREADS) d[e] = s[e];
+    }
+  }
+}
+
+__global__ void k_scatter_add_keys(float* __restrict__ slab, const int64_t* __restrict__ keys,
+                                   int n, int32_t len, int32_t plen, int world, int rank,
+                                   const float* __restrict__ in) {
+  for (int i = blockIdx.x; i < n; i += gridDim.x) {
+    int64_t k = keys[i];
+    if ((int)(k % world) != rank) continue;
+    float* s = slab + (k / world) * (int64_t)plen;
+    const float* d = in + (int64_t)i * len;
+    for (int e = threadIdx.x; e < len; e += THREADS) atomicAdd(&s[e], d[e]);
+  }
+}
+
+__global__ void k_scatter_set_keys(float* __restrict__ slab, const int64_t* __restrict__ keys,
+                                   int n, int32_t len, int32_t plen, int world, int rank,
+                                   const float* __restrict__ in) {
+  for (int i = blockIdx.x; i < n; i += gridDim.x) {
+    int64_t k = keys[i];
+    if ((int)(k % world) != rank) continue;
+    float* s = slab + (k / world) * (int64_t)plen;
+    const float* d = in + (int64_t)i * len;
+    if ((len & 3) == 0) {
+      float4* sp = reinterpret_cast<float4*>(s);
+      const float4* dp = reinterpret_cast<const float4*>(d);
+      for (int e = threadIdx.x; e < (len >> 2); e += THREADS) sp[e] = dp[e];
+    } else {
+      for (int e = threadIdx.x; e < len; e += THREADS) s[e] = d[e];
+    }
+  }
+}
+
 static inline int grid_for(int n) {
   // >=2048 workgroups fills 256 CUs at 8 blocks/CU; grid-stride covers the rest
   int g = n < 1 ? 1 : n;
@@ -142,6 +190,21 @@ void ops_zero_gpu(float* slab, const OpsBatch& b, void* stream) {
   if (b.n == 0) return;
   hipLaunchKernelGGL(k_zero, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
                      slab, b.dst_off, b.lens, b.n);
+}
+
+void ops_gather_keys_gpu(const float* slab, const KeyBatch& b, float* out, void* stream) {
+  if (b.n == 0) return;
+  hipLaunchKernelGGL(k_gather_keys, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
+                     slab, b.keys, b.n, b.len, b.plen, b.world, b.rank, out);
+}
+void ops_scatter_keys_gpu(float* slab, const KeyBatch& b, const float* in, bool set, void* stream) {
+  if (b.n == 0) return;
+  if (set)
+    hipLaunchKernelGGL(k_scatter_set_keys, dim3(grid_for(b.n)), dim3(THREADS), 0,
+                       (hipStream_t)stream, slab, b.keys, b.n, b.len, b.plen, b.world, b.rank, in);
+  else
+    hipLaunchKernelGGL(k_scatter_add_keys, dim3(grid_for(b.n)), dim3(THREADS), 0,
+                       (hipStream_t)stream, slab, b.keys, b.n, b.len, b.plen, b.world, b.rank, in);
 }
 
 bool hip_available() {

@@ -130,6 +130,9 @@ def main():
     ms_per_step = 1000.0 * t_elapsed / args.steps
 
     st = server.stats()
+    if rank == 0 and os.environ.get("ADAPM_CPP_TIMING"):
+        print("cpp timing:", {k: v for k, v in st.items() if k.startswith("t_")},
+              file=sys.stderr, flush=True)
     if rank == 0 and model.phase_times:
         print("phase times (s over timed+warmup):",
               dict(sorted(model.phase_times.items())), file=sys.stderr, flush=True)
