@@ -45,7 +45,7 @@ def build(verbose: bool = True) -> Path:
     out = REPO / "adapm_amd" / soname
 
     cflags = [
-        "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC", "-DNDEBUG",
+        "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC", "-DNDEBUG", "-munsafe-fp-atomics",
         "-D_GLIBCXX_USE_CXX11_ABI=1", f"-DTORCH_EXTENSION_NAME={MODULE}",
         "-DUSE_ROCM", "-D__HIP_PLATFORM_AMD__", "-DADAPM_WITH_HIP",
         "-Wno-unused-result",
