@@ -31,6 +31,7 @@
 //  - the CPU backend serializes value ops with one mutex (cpu_val_mu_) —
 //    it is the test tier, not the perf path.
 #include <torch/extension.h>
+#include <ATen/Parallel.h>
 
 #include <algorithm>
 #include <atomic>
@@ -189,10 +190,15 @@ class Server {
       lens_.assign(lens.data_ptr<int32_t>(), lens.data_ptr<int32_t>() + num_keys_);
     }
 
-    flags_.assign(num_keys_, 0);
-    loc_.assign(num_keys_, -1);
+    flags_ = std::vector<std::atomic<uint8_t>>(num_keys_);
+    loc_ = std::vector<std::atomic<int64_t>>(num_keys_);
+    version_ = std::vector<std::atomic<uint32_t>>(num_keys_);
+    for (int64_t i = 0; i < num_keys_; ++i) {
+      flags_[i].store(0, std::memory_order_relaxed);
+      loc_[i].store(-1, std::memory_order_relaxed);
+      version_[i].store(0, std::memory_order_relaxed);
+    }
     sync_loc_.assign(num_keys_, -1);
-    version_.assign(num_keys_, 0);
     if (use_loc_cache_) loc_cache_.assign(num_keys_, -1);
     int64_t n_managed = (num_keys_ + world_ - 1) / world_;
     owner_of_.assign(n_managed, rank_);  // initially every key lives at its manager
@@ -429,24 +435,50 @@ class Server {
         stat_pull_local_ += n - (int64_t)remote.size();
         stat_pull_keys_ += n;
       } else {
-        for (int64_t i = 0; i < n; ++i) {
-          Key k = kp[i];
-          int32_t l = len_of(k);
-          out_off[i] = cum;
-          out_len[i] = l;
-          {
-            std::lock_guard<std::mutex> lk(stripe(k));
-            uint8_t f = flags_[k];
-            if ((f & F_PRESENT) && !(f & F_STUB)) {
-              local.add(loc_[k], cum, l);
-              if (!(f & F_OWNER)) stat_pull_replica_ += 1;
-              stat_pull_local_ += 1;
-            } else {
-              remote.push_back({k, i});
+        // lock-free parallel metadata pass (atomic reads; see meta rules)
+        if (uniform_len_ >= 0) {
+          const int32_t l = uniform_len_;
+          for (int64_t i = 0; i < n; ++i) { out_off[i] = i * (int64_t)l; out_len[i] = l; }
+        } else {
+          for (int64_t i = 0; i < n; ++i) {
+            int32_t l = len_of(kp[i]);
+            out_off[i] = cum;
+            out_len[i] = l;
+            cum += l;
+          }
+        }
+        constexpr int64_t G = 8192;
+        int64_t nchunks = (n + G - 1) / G;
+        struct Part {
+          HostBatch local;
+          std::vector<Remote> remote;
+          int64_t n_repl = 0;
+        };
+        std::vector<Part> parts(nchunks);
+        at::parallel_for(0, nchunks, 1, [&](int64_t c0, int64_t c1) {
+          for (int64_t c = c0; c < c1; ++c) {
+            Part& P = parts[c];
+            int64_t e = std::min(n, (c + 1) * G);
+            for (int64_t i = c * G; i < e; ++i) {
+              Key k = kp[i];
+              uint8_t f = flags_[k].load(std::memory_order_acquire);
+              if ((f & F_PRESENT) && !(f & F_STUB)) {
+                P.local.add(loc_[k].load(std::memory_order_acquire), out_off[i], out_len[i]);
+                if (!(f & F_OWNER)) P.n_repl++;
+              } else {
+                P.remote.push_back({k, i});
+              }
             }
           }
-          cum += l;
+        });
+        for (auto& P : parts) {
+          local.src.insert(local.src.end(), P.local.src.begin(), P.local.src.end());
+          local.dst.insert(local.dst.end(), P.local.dst.begin(), P.local.dst.end());
+          local.len.insert(local.len.end(), P.local.len.begin(), P.local.len.end());
+          remote.insert(remote.end(), P.remote.begin(), P.remote.end());
+          stat_pull_replica_ += P.n_repl;
         }
+        stat_pull_local_ += n - (int64_t)remote.size();
         stat_pull_keys_ += n;
         if (cpp_timing_) { t_pass_ += tick() - tp0; tp0 = tick(); }
         run_gather(local, vals_dev);
@@ -522,27 +554,58 @@ class Server {
         stat_push_local_ += n - (int64_t)remote.size();
         stat_push_keys_ += n;
       } else {
-        for (int64_t i = 0; i < n; ++i) {
-          Key k = kp[i];
-          int32_t l = len_of(k);
-          {
-            std::lock_guard<std::mutex> lk(stripe(k));
-            uint8_t f = flags_[k];
-            if ((f & F_PRESENT) && (f & F_OWNER)) {
-              (set_mode ? assign : merge).add(loc_[k], cum, l);
-              version_[k]++;
-              stat_push_local_ += 1;
-            } else if ((f & F_PRESENT) && !set_mode) {
-              merge.add(loc_[k], cum, l);  // replica/stub: merge, flush at next sync
-              flags_[k] = f | F_UPDATED;
-              stat_push_local_ += 1;
-              stat_push_replica_ += 1;
-            } else {
-              remote.push_back({k, cum, l});
+        // lock-free parallel metadata pass (atomic reads; see meta rules)
+        std::vector<int64_t> offs(n);
+        if (uniform_len_ >= 0) {
+          for (int64_t i = 0; i < n; ++i) offs[i] = i * (int64_t)uniform_len_;
+        } else {
+          for (int64_t i = 0; i < n; ++i) {
+            offs[i] = cum;
+            cum += len_of(kp[i]);
+          }
+        }
+        constexpr int64_t G = 8192;
+        int64_t nchunks = (n + G - 1) / G;
+        struct Part {
+          HostBatch merge, assign;
+          std::vector<Remote> remote;
+          int64_t n_repl = 0;
+        };
+        std::vector<Part> parts(nchunks);
+        at::parallel_for(0, nchunks, 1, [&](int64_t c0, int64_t c1) {
+          for (int64_t c = c0; c < c1; ++c) {
+            Part& P = parts[c];
+            int64_t e = std::min(n, (c + 1) * G);
+            for (int64_t i = c * G; i < e; ++i) {
+              Key k = kp[i];
+              int32_t l = len_of(k);
+              uint8_t f = flags_[k].load(std::memory_order_acquire);
+              if ((f & F_PRESENT) && (f & F_OWNER)) {
+                (set_mode ? P.assign : P.merge).add(loc_[k].load(std::memory_order_acquire),
+                                                    offs[i], l);
+                version_[k].fetch_add(1, std::memory_order_relaxed);
+              } else if ((f & F_PRESENT) && !set_mode) {
+                // replica/stub: merge locally, flush at next sync round
+                P.merge.add(loc_[k].load(std::memory_order_acquire), offs[i], l);
+                flags_[k].fetch_or(F_UPDATED);
+                P.n_repl++;
+              } else {
+                P.remote.push_back({k, offs[i], l});
+              }
             }
           }
-          cum += l;
+        });
+        for (auto& P : parts) {
+          std::pair<HostBatch*, HostBatch*> prs[2] = {{&merge, &P.merge}, {&assign, &P.assign}};
+          for (auto& pr : prs) {
+            pr.first->src.insert(pr.first->src.end(), pr.second->src.begin(), pr.second->src.end());
+            pr.first->dst.insert(pr.first->dst.end(), pr.second->dst.begin(), pr.second->dst.end());
+            pr.first->len.insert(pr.first->len.end(), pr.second->len.begin(), pr.second->len.end());
+          }
+          remote.insert(remote.end(), P.remote.begin(), P.remote.end());
+          stat_push_replica_ += P.n_repl;
         }
+        stat_push_local_ += n - (int64_t)remote.size();
         stat_push_keys_ += n;
         if (cpp_timing_) { int64_t t1 = tick(); t_pass_ += t1 - tp0; tp0 = t1; }
         run_scatter(merge, flat, false);
@@ -586,10 +649,9 @@ class Server {
         Key k = kp[i];
         int32_t l = len_of(k);
         {
-          std::lock_guard<std::mutex> lk(stripe(k));
-          uint8_t f = flags_[k];
+            uint8_t f = flags_[k].load(std::memory_order_acquire);
           if (!((f & F_PRESENT) && !(f & F_STUB))) return false;
-          local.add(loc_[k], cum, l);
+          local.add(loc_[k].load(std::memory_order_acquire), cum, l);
         }
         cum += l;
       }
@@ -600,8 +662,7 @@ class Server {
   }
 
   bool is_local(Key k) {
-    std::lock_guard<std::mutex> lk(stripe(k));
-    uint8_t f = flags_[k];
+    uint8_t f = flags_[k].load(std::memory_order_acquire);
     return (f & F_PRESENT) && !(f & F_STUB);
   }
 
@@ -777,17 +838,21 @@ class Server {
           bool updated = f & F_UPDATED;
           bool is_new = f & F_STUB;
           bool drop = !has_intent && !is_new;
-          int64_t fl = (updated ? D_HAS_PAYLOAD : 0) | (drop ? D_DROPPING : 0) |
+          // dropped replicas ALWAYS carry a payload: a concurrent local
+          // push between our flags read and the absent-store would
+          // otherwise be lost (its kernel lands before our extract by
+          // quiesce + stream order, so the extracted delta captures it)
+          bool payload = updated || drop;
+          int64_t fl = (payload ? D_HAS_PAYLOAD : 0) | (drop ? D_DROPPING : 0) |
                        (has_intent ? D_WANT_REFRESH : 0) | (is_new ? D_NEW : 0);
-          deltas.push_back(DeltaRec{k, (int64_t)version_[k], fl, updated ? loc_[k] : -1,
-                                    sync_loc_[k], len_of(k)});
-          if (updated) flags_[k] = f & ~F_UPDATED;
+          deltas.push_back(DeltaRec{k, (int64_t)version_[k].load(), fl,
+                                    payload ? loc_[k].load() : -1, sync_loc_[k], len_of(k)});
+          if (updated && !drop) flags_[k].fetch_and((uint8_t)~F_UPDATED);
           if (drop) {
             layout_identity_.store(false, std::memory_order_release);
             frees.push_back({loc_[k], len_of(k)});
             frees.push_back({sync_loc_[k], len_of(k)});
-            flags_[k] = 0;
-            loc_[k] = -1;
+            flags_[k] = 0;  // loc field left stale on purpose (see meta rules)
             sync_loc_[k] = -1;
             erase_from_replicas = true;
             stat_drops_ += 1;
@@ -799,8 +864,11 @@ class Server {
         C.replicas.erase(k);
       }
     }
-    // no worker op may still reference a dropped slot's old metadata
-    if (!frees.empty()) quiesce();
+    // no worker op may still be between its metadata read and kernel
+    // launch: (a) dropped slots must not be reused under it, (b) the
+    // delta-extract kernels must be enqueued AFTER any concurrent push
+    // kernel whose UPDATED bit we just consumed.
+    if (!frees.empty() || !deltas.empty()) quiesce();
 
     // 4. build per-destination messages
     struct Msg {
@@ -1107,9 +1175,8 @@ class Server {
         std::lock_guard<std::mutex> lk(stripe(k));
         if (!(flags_[k] & F_OWNER)) return;  // raced
         voff = loc_[k];
-        flags_[k] = 0;  // absent: new local ops route remotely
-        loc_[k] = -1;
-        new_ver = ++version_[k];
+        flags_[k] = 0;  // absent: new local ops route remotely (loc left stale)
+        new_ver = version_[k].fetch_add(1) + 1;
       }
       quiesce();  // no worker op may still hold the old offset
       if (use_loc_cache_) loc_cache_[k] = origin_rank;
@@ -1170,6 +1237,7 @@ class Server {
 
   std::vector<std::tuple<int, torch::Tensor, torch::Tensor>> sync_respond(int ch) {
     ChannelState& C = channels_[ch];
+    quiesce();  // refresh/pull-resp gathers must follow in-flight push kernels
     std::vector<RespRec> resp;
     {
       std::lock_guard<std::mutex> g(C.mu);
@@ -1309,7 +1377,7 @@ class Server {
           sync_loc_[p.k] = -1;
           flags_[p.k] = F_PRESENT | F_OWNER;
         } else {
-          flags_[p.k] = flags_[p.k] & ~F_STUB;  // preserves a concurrent UPDATED
+          flags_[p.k].fetch_and((uint8_t)~F_STUB);  // preserves a concurrent UPDATED
         }
       }
       if (p.relocate) {
@@ -1436,9 +1504,10 @@ class Server {
   }
 
   torch::Tensor debug_flags() {
-    return torch::from_blob(flags_.data(), {(int64_t)num_keys_},
-                            torch::TensorOptions().dtype(torch::kUInt8))
-        .clone();
+    auto t = torch::empty({(int64_t)num_keys_}, torch::TensorOptions().dtype(torch::kUInt8));
+    uint8_t* p = t.data_ptr<uint8_t>();
+    for (int64_t i = 0; i < num_keys_; ++i) p[i] = flags_[i].load();
+    return t;
   }
 
  private:
@@ -1455,9 +1524,19 @@ class Server {
   std::vector<int32_t> lens_;
 
   Slab slab_;
-  std::vector<uint8_t> flags_;
-  std::vector<int64_t> loc_, sync_loc_;
-  std::vector<uint32_t> version_;
+  // Per-key metadata. flags_/loc_/version_ are atomics so the worker
+  // metadata pass reads them LOCK-FREE (and in parallel); all WRITERS
+  // still serialize on the stripe mutexes. Consistency of the (flags,
+  // loc) pair without a common lock relies on three rules:
+  //  1. transitions to available write loc BEFORE flags,
+  //  2. transitions to absent write flags only — the stale loc field is
+  //     harmless because slab-slot reuse is stream-ordered and the sync
+  //     thread quiesces in-flight ops before structural changes,
+  //  3. bit updates racing with reads use fetch_or/fetch_and.
+  std::vector<std::atomic<uint8_t>> flags_;
+  std::vector<std::atomic<int64_t>> loc_;
+  std::vector<int64_t> sync_loc_;
+  std::vector<std::atomic<uint32_t>> version_;
   std::vector<int32_t> loc_cache_;
   std::vector<int32_t> owner_of_;
   std::vector<uint32_t> mgr_reloc_ctr_;
