@@ -1,1 +1,1 @@
-from . import kge  # noqa: F401
+from . import ctr, kge, mf, simple, word2vec  # noqa: F401
