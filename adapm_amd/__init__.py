@@ -46,7 +46,8 @@ _SETUP = {}
 def setup(num_keys: int, num_threads: int, use_techniques: str = "", num_channels: int = -1,
           device: Optional[str] = None, max_sync_per_sec: float = 1000.0,
           time_intent_actions: bool = True, capacity_factor: float = 2.0,
-          location_caches: bool = True):
+          location_caches: bool = True, locality_stats: bool = False,
+          trace_keys=None, stats_out: Optional[str] = None):
     """Global configuration (reference bindings.cc:18-31 `setup`)."""
     tech = TECH_ALL
     t = use_techniques.strip().lower()
@@ -63,7 +64,8 @@ def setup(num_keys: int, num_threads: int, use_techniques: str = "", num_channel
                        max_sync_per_sec=max_sync_per_sec,
                        time_intent_actions=time_intent_actions,
                        capacity_factor=capacity_factor,
-                       location_caches=location_caches))
+                       location_caches=location_caches, locality_stats=locality_stats,
+                       trace_keys=trace_keys, stats_out=stats_out))
 
 
 def scheduler(num_keys: int = 0, num_threads: int = 0):
@@ -107,6 +109,13 @@ class Server:
             device=str(rt.device), capacity_factor=cfg["capacity_factor"],
             techniques=cfg["techniques"], location_caches=cfg["location_caches"],
         )
+        if cfg.get("locality_stats"):
+            self._s.enable_locality_stats()
+        if cfg.get("trace_keys") is not None:
+            tk = cfg["trace_keys"]
+            t = torch.tensor([-1], dtype=torch.int64) if tk == "all" else _to_key_tensor(tk)
+            self._s.enable_key_trace(t)
+        self._stats_out = cfg.get("stats_out")
         self.sampling: Optional[SamplingManager] = None
         self._sync = SyncManager(self._s, rt, max_per_sec=cfg["max_sync_per_sec"],
                                  time_intent_actions=cfg["time_intent_actions"])
@@ -137,7 +146,32 @@ class Server:
         self._shut = True
         self._sync.request_stop()
         self._sync.join()
+        self._final_report()
         _rt.shutdown_runtime()
+
+    def _final_report(self):
+        """End-of-run locality summary (reference coloc_kv_server.h:147-157)
+        + optional TSV dumps (--sys.stats.out equivalent)."""
+        import os
+        import sys
+
+        st = self.stats()
+        if os.environ.get("ADAPM_VERBOSE", "0") != "0":
+            pl = st["pull_local"] / max(1, st["pull_keys"])
+            ph = st["push_local"] / max(1, st["push_keys"])
+            print(f"[adapm rank {self.rt.rank}] pulls: {st['pull_keys']} ({pl:.1%} local, "
+                  f"{st['pull_replica']} from replicas); pushes: {st['push_keys']} "
+                  f"({ph:.1%} local); relocations {st['relocations_out']}/"
+                  f"{st['relocations_in']} out/in; replications {st['replications']}; "
+                  f"drops {st['replica_drops']}; sync rounds {st['sync_rounds']}; "
+                  f"{st['bytes_sent']/1e6:.1f}/{st['bytes_recv']/1e6:.1f} MB sent/recv",
+                  file=sys.stderr, flush=True)
+        if self._stats_out:
+            os.makedirs(self._stats_out, exist_ok=True)
+            r = self.rt.rank
+            self._s.dump_locality_stats(os.path.join(self._stats_out,
+                                                     f"locality_stats.rank.{r}.tsv"))
+            self._s.dump_traces(os.path.join(self._stats_out, f"traces.{r}.tsv"))
 
     # ---- extras
 

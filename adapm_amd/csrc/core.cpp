@@ -44,6 +44,7 @@
 #include <chrono>
 #include <unordered_map>
 #include <unordered_set>
+#include <cstdio>
 #include <vector>
 
 #include "common.h"
@@ -465,8 +466,13 @@ class Server {
               if ((f & F_PRESENT) && !(f & F_STUB)) {
                 P.local.add(loc_[k].load(std::memory_order_acquire), out_off[i], out_len[i]);
                 if (!(f & F_OWNER)) P.n_repl++;
+                if (locality_stats_) {
+                  key_accesses_[k].fetch_add(1, std::memory_order_relaxed);
+                  key_local_[k].fetch_add(1, std::memory_order_relaxed);
+                }
               } else {
                 P.remote.push_back({k, i});
+                if (locality_stats_) key_accesses_[k].fetch_add(1, std::memory_order_relaxed);
               }
             }
           }
@@ -786,6 +792,7 @@ class Server {
           C.intents[k].push_back({req.wid, req.end});
           if (need_stub) C.replicas.insert(k);
         }
+        trace_event(k, need_stub ? "REPLICA_SETUP" : "INTENT_START");
         if (need_stub) {  // freelist reuse leaves stale data: zero it
           zero_batch.add(0, v_off, l);
           zero_batch.add(0, s_off, l);
@@ -856,6 +863,7 @@ class Server {
             sync_loc_[k] = -1;
             erase_from_replicas = true;
             stat_drops_ += 1;
+            trace_event(k, "REPLICA_DROP");
           }
         }
       }
@@ -1199,6 +1207,7 @@ class Server {
             RespRec{mgr, M_RESIDENCE, k, origin_rank, (int64_t)ctr, 0, -1, 0, false, {}});
       }
       stat_relocations_ += 1;
+      trace_event(k, "RELOC_OUT");
     } else {
       bool is_new = dflags & D_NEW;
       // a replica of one of our keys now exists: versions become
@@ -1389,6 +1398,7 @@ class Server {
         }
         if (use_loc_cache_) loc_cache_[p.k] = -1;
         stat_relocated_in_ += 1;
+        trace_event(p.k, "RELOC_IN");
       }
     }
     for (auto& [off, len] : local_frees) slab_.free_(off, len);
@@ -1468,6 +1478,56 @@ class Server {
   int world() const { return world_; }
   int num_channels() const { return nch_; }
   int owner_hint(Key k) { return directions(k); }
+
+  void enable_locality_stats() {
+    locality_stats_ = true;
+    key_accesses_ = std::vector<std::atomic<uint32_t>>(num_keys_);
+    key_local_ = std::vector<std::atomic<uint32_t>>(num_keys_);
+    for (int64_t i = 0; i < num_keys_; ++i) {
+      key_accesses_[i].store(0);
+      key_local_[i].store(0);
+    }
+  }
+
+  void enable_key_trace(torch::Tensor keys) {
+    std::lock_guard<std::mutex> g(trace_mu_);
+    if (keys.numel() == 1 && keys.data_ptr<int64_t>()[0] == -1) {
+      trace_all_ = true;
+      return;
+    }
+    auto kc = keys.contiguous();
+    for (int64_t i = 0; i < kc.numel(); ++i) traced_keys_.insert(kc.data_ptr<int64_t>()[i]);
+  }
+
+  void trace_event(Key k, const char* ev) {
+    if (!trace_all_ && traced_keys_.empty()) return;
+    std::lock_guard<std::mutex> g(trace_mu_);
+    if (!trace_all_ && !traced_keys_.count(k)) return;
+    double t = std::chrono::duration<double>(std::chrono::steady_clock::now() - t0_).count();
+    trace_.push_back({t, k, ev});
+  }
+
+  // TSV dumps (reference locality_stats.rank.<r>.tsv / traces.<r>.tsv)
+  void dump_locality_stats(std::string path) {
+    if (!locality_stats_) return;
+    FILE* f = fopen(path.c_str(), "w");
+    if (!f) throw std::runtime_error("cannot open " + path);
+    fprintf(f, "key\taccesses\tlocal\n");
+    for (int64_t k = 0; k < num_keys_; ++k) {
+      uint32_t a = key_accesses_[k].load();
+      if (a) fprintf(f, "%lld\t%u\t%u\n", (long long)k, a, key_local_[k].load());
+    }
+    fclose(f);
+  }
+
+  void dump_traces(std::string path) {
+    std::lock_guard<std::mutex> g(trace_mu_);
+    FILE* f = fopen(path.c_str(), "w");
+    if (!f) throw std::runtime_error("cannot open " + path);
+    fprintf(f, "time_s\tkey\tevent\n");
+    for (auto& e : trace_) fprintf(f, "%.6f\t%lld\t%s\n", e.t, (long long)e.k, e.ev);
+    fclose(f);
+  }
 
   py::dict stats() {
     py::dict d;
@@ -1568,6 +1628,17 @@ class Server {
   std::unordered_map<int64_t, std::unique_ptr<Ticket>> tickets_;
   std::mutex rounds_mu_;
   std::condition_variable rounds_cv_;
+
+  // opt-in observability (reference PS_LOCALITY_STATS / PS_TRACE_KEYS,
+  // coloc_kv_server_handle.h:86-118, 960-992)
+  bool locality_stats_ = false;
+  std::vector<std::atomic<uint32_t>> key_accesses_, key_local_;
+  std::mutex trace_mu_;
+  std::unordered_set<Key> traced_keys_;
+  bool trace_all_ = false;
+  struct TraceEv { double t; Key k; const char* ev; };
+  std::vector<TraceEv> trace_;
+  std::chrono::steady_clock::time_point t0_ = std::chrono::steady_clock::now();
 
   std::atomic<int64_t> stat_pulls_{0}, stat_pushes_{0}, stat_pull_keys_{0}, stat_push_keys_{0},
       stat_pull_local_{0}, stat_push_local_{0}, stat_pull_replica_{0}, stat_push_replica_{0},
@@ -1687,6 +1758,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("world", &Server::world)
       .def("num_channels", &Server::num_channels)
       .def("owner_hint", &Server::owner_hint)
+      .def("enable_locality_stats", &Server::enable_locality_stats)
+      .def("enable_key_trace", &Server::enable_key_trace)
+      .def("dump_locality_stats", &Server::dump_locality_stats)
+      .def("dump_traces", &Server::dump_traces)
       .def("stats", &Server::stats)
       .def("debug_flags", &Server::debug_flags);
 }

@@ -113,14 +113,24 @@ class SyncManager:
         if rt.is_cuda:
             torch.cuda.set_device(rt.device)
 
+        verbose = __import__("os").environ.get("ADAPM_VERBOSE", "0") != "0"
+        last_report = time.monotonic()
+        rounds_at_report = 0
+        n_rounds = 0
         while True:
             t0 = time.monotonic()
+            if verbose and ch == 0 and t0 - last_report > 10.0:
+                clocks = self.server.worker_clocks()
+                print(f"[adapm sync r{rank} ch{ch}] {(n_rounds - rounds_at_report) / (t0 - last_report):.0f} rounds/s, "
+                      f"worker clocks {clocks}", flush=True)
+                last_report, rounds_at_report = t0, n_rounds
             if ch == 0 and self.time_intent_actions:
                 self.server.set_intent_ahead(self.timer.update(self.server.worker_clocks()))
 
             stop = self.stop_requested.is_set()
             all_stopped = self._round(ch, group, world, rank, dev, stop)
             self.server.sync_finish(ch)
+            n_rounds += 1
             if all_stopped:
                 return
             if self.min_period > 0:
