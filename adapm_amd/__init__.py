@@ -47,7 +47,8 @@ def setup(num_keys: int, num_threads: int, use_techniques: str = "", num_channel
           device: Optional[str] = None, max_sync_per_sec: float = 1000.0,
           time_intent_actions: bool = True, capacity_factor: float = 2.0,
           location_caches: bool = True, locality_stats: bool = False,
-          trace_keys=None, stats_out: Optional[str] = None):
+          trace_keys=None, stats_out: Optional[str] = None,
+          device_cap_gb: float = 0.0, host_spill_gb: float = 0.0):
     """Global configuration (reference bindings.cc:18-31 `setup`)."""
     tech = TECH_ALL
     t = use_techniques.strip().lower()
@@ -65,7 +66,8 @@ def setup(num_keys: int, num_threads: int, use_techniques: str = "", num_channel
                        time_intent_actions=time_intent_actions,
                        capacity_factor=capacity_factor,
                        location_caches=location_caches, locality_stats=locality_stats,
-                       trace_keys=trace_keys, stats_out=stats_out))
+                       trace_keys=trace_keys, stats_out=stats_out,
+                       device_cap_gb=device_cap_gb, host_spill_gb=host_spill_gb))
 
 
 def scheduler(num_keys: int = 0, num_threads: int = 0):
@@ -108,6 +110,8 @@ class Server:
             num_channels=cfg["num_channels"], num_workers=cfg["num_threads"],
             device=str(rt.device), capacity_factor=cfg["capacity_factor"],
             techniques=cfg["techniques"], location_caches=cfg["location_caches"],
+            device_cap_floats=int(cfg.get("device_cap_gb", 0) * (1 << 30) / 4),
+            host_spill_floats=int(cfg.get("host_spill_gb", 0) * (1 << 30) / 4),
         )
         if cfg.get("locality_stats"):
             self._s.enable_locality_stats()

@@ -55,6 +55,12 @@
 #include <c10/hip/HIPStream.h>
 #endif
 
+extern "C" {
+int adapm_host_arena_alloc(long long floats, int want_device_visible, void** host_ptr,
+                           void** dev_ptr);
+void adapm_host_arena_free(void* host_ptr, int was_device_visible);
+}
+
 namespace adapm {
 
 namespace py = pybind11;
@@ -78,14 +84,39 @@ struct Slab {
   std::unordered_map<int32_t, std::vector<int64_t>> freelists;
   std::mutex mu;
   std::atomic<int64_t> in_use{0};
+  // host-spill arena: pinned host memory, device-visible (zero-copy over
+  // PCIe). Used when the device arena is exhausted — "HBM as a cache over
+  // pinned host memory" (BASELINE config 5). Offsets carry SPILL_BIT.
+  bool is_cuda = false;
+  void* host_raw = nullptr;          // host pointer (hipHostMalloc / malloc)
+  float* host_host = nullptr;        // host-side view
+  float* host_dev = nullptr;         // device-visible pointer
+  int64_t host_capacity = 0;
+  int64_t host_bump = 0;
+  std::unordered_map<int32_t, std::vector<int64_t>> host_freelists;
+  std::atomic<int64_t> host_in_use{0};
 
   static int32_t padded(int32_t len) { return (len + 3) & ~3; }
 
   void init(int64_t cap, const torch::Device& dev) {
     capacity = cap;
+    is_cuda = dev.is_cuda();
     buf = torch::zeros({cap}, torch::TensorOptions().dtype(torch::kFloat32).device(dev));
     data = buf.data_ptr<float>();
   }
+
+  void init_host(int64_t cap) {
+    void *hp = nullptr, *dp = nullptr;
+    int rc = adapm_host_arena_alloc(cap, is_cuda ? 1 : 0, &hp, &dp);
+    if (rc != 0) throw std::runtime_error("host-spill arena allocation failed");
+    host_raw = hp;
+    host_host = (float*)hp;
+    host_dev = (float*)dp;
+    host_capacity = cap;
+  }
+  ~Slab() { adapm_host_arena_free(host_raw, is_cuda ? 1 : 0); }
+
+  SlabBases bases() const { return SlabBases{data, is_cuda ? host_dev : host_host}; }
 
   int64_t alloc(int32_t len) {
     int32_t p = padded(len);
@@ -97,18 +128,41 @@ struct Slab {
       in_use += p;
       return off;
     }
-    if (bump + p > capacity) throw std::runtime_error("adapm slab out of capacity");
-    int64_t off = bump;
-    bump += p;
-    in_use += p;
-    return off;
+    if (bump + p <= capacity) {
+      int64_t off = bump;
+      bump += p;
+      in_use += p;
+      return off;
+    }
+    // device arena exhausted: spill to the host arena
+    if (host_capacity > 0) {
+      auto hit = host_freelists.find(p);
+      if (hit != host_freelists.end() && !hit->second.empty()) {
+        int64_t off = hit->second.back();
+        hit->second.pop_back();
+        host_in_use += p;
+        return off | SPILL_BIT;
+      }
+      if (host_bump + p <= host_capacity) {
+        int64_t off = host_bump;
+        host_bump += p;
+        host_in_use += p;
+        return off | SPILL_BIT;
+      }
+    }
+    throw std::runtime_error("adapm slab out of capacity (device and host-spill)");
   }
 
   void free_(int64_t off, int32_t len) {
     int32_t p = padded(len);
     std::lock_guard<std::mutex> g(mu);
-    freelists[p].push_back(off);
-    in_use -= p;
+    if (off & SPILL_BIT) {
+      host_freelists[p].push_back(off & ~SPILL_BIT);
+      host_in_use -= p;
+    } else {
+      freelists[p].push_back(off);
+      in_use -= p;
+    }
   }
 };
 
@@ -170,7 +224,7 @@ class Server {
  public:
   Server(int64_t num_keys, torch::Tensor lens, int rank, int world, int num_channels,
          int num_workers, std::string device, double capacity_factor, int techniques,
-         bool location_caches)
+         bool location_caches, int64_t device_cap_floats = 0, int64_t host_spill_floats = 0)
       : num_keys_(num_keys),
         rank_(rank),
         world_(world),
@@ -210,7 +264,12 @@ class Server {
     int64_t owned_floats = 0;
     for (Key k = rank_; k < num_keys_; k += world_) owned_floats += Slab::padded(len_of(k));
     int64_t cap = (int64_t)((double)owned_floats * capacity_factor) + (1 << 20);
+    if (device_cap_floats > 0) cap = device_cap_floats;  // explicit HBM budget
     slab_.init(cap, dev_);
+    if (host_spill_floats > 0) slab_.init_host(host_spill_floats);
+    // if the initial allocation will not fit the device arena, identity
+    // breaks immediately (spilled slots are not at identity offsets)
+    if (owned_floats > cap) layout_identity_.store(false);
     for (Key k = rank_; k < num_keys_; k += world_) {
       int32_t l = len_of(k);
       loc_[k] = slab_.alloc(l);
@@ -304,10 +363,10 @@ class Server {
     if (hb.size() == 0) return;
     auto d = to_dev(hb);
     if (dev_.is_cuda()) {
-      ops_gather_gpu(slab_.data, d.b, out.data_ptr<float>(), current_stream(dev_));
+      ops_gather_gpu(slab_.bases(), d.b, out.data_ptr<float>(), current_stream(dev_));
     } else {
       std::lock_guard<std::mutex> g(cpu_val_mu_);
-      ops_gather_cpu(slab_.data, d.b, out.data_ptr<float>());
+      ops_gather_cpu(slab_.bases(), d.b, out.data_ptr<float>());
     }
   }
   void run_scatter(const HostBatch& hb, torch::Tensor in, bool set) {
@@ -315,42 +374,42 @@ class Server {
     auto d = to_dev(hb);
     auto in_c = in.is_contiguous() ? in : in.contiguous();
     if (dev_.is_cuda()) {
-      ops_scatter_gpu(slab_.data, d.b, in_c.data_ptr<float>(), set, current_stream(dev_));
+      ops_scatter_gpu(slab_.bases(), d.b, in_c.data_ptr<float>(), set, current_stream(dev_));
     } else {
       std::lock_guard<std::mutex> g(cpu_val_mu_);
-      ops_scatter_cpu(slab_.data, d.b, in_c.data_ptr<float>(), set);
+      ops_scatter_cpu(slab_.bases(), d.b, in_c.data_ptr<float>(), set);
     }
   }
   void run_extract(const HostBatch& hb, const std::vector<int64_t>& sync_off, torch::Tensor out) {
     if (hb.size() == 0) return;
     auto d = to_dev(hb, &sync_off);
     if (dev_.is_cuda()) {
-      ops_extract_gpu(slab_.data, d.b, d.aux_t.data_ptr<int64_t>(), out.data_ptr<float>(),
+      ops_extract_gpu(slab_.bases(), d.b, d.aux_t.data_ptr<int64_t>(), out.data_ptr<float>(),
                       current_stream(dev_));
     } else {
       std::lock_guard<std::mutex> g(cpu_val_mu_);
-      ops_extract_cpu(slab_.data, d.b, sync_off.data(), out.data_ptr<float>());
+      ops_extract_cpu(slab_.bases(), d.b, sync_off.data(), out.data_ptr<float>());
     }
   }
   void run_refresh(const HostBatch& hb, const std::vector<int64_t>& sync_off, torch::Tensor in) {
     if (hb.size() == 0) return;
     auto d = to_dev(hb, &sync_off);
     if (dev_.is_cuda()) {
-      ops_refresh_gpu(slab_.data, d.b, d.aux_t.data_ptr<int64_t>(), in.data_ptr<float>(),
+      ops_refresh_gpu(slab_.bases(), d.b, d.aux_t.data_ptr<int64_t>(), in.data_ptr<float>(),
                       current_stream(dev_));
     } else {
       std::lock_guard<std::mutex> g(cpu_val_mu_);
-      ops_refresh_cpu(slab_.data, d.b, sync_off.data(), in.data_ptr<float>());
+      ops_refresh_cpu(slab_.bases(), d.b, sync_off.data(), in.data_ptr<float>());
     }
   }
   void run_zero(const HostBatch& hb) {
     if (hb.size() == 0) return;
     auto d = to_dev(hb);
     if (dev_.is_cuda()) {
-      ops_zero_gpu(slab_.data, d.b, current_stream(dev_));
+      ops_zero_gpu(slab_.bases(), d.b, current_stream(dev_));
     } else {
       std::lock_guard<std::mutex> g(cpu_val_mu_);
-      ops_zero_cpu(slab_.data, d.b);
+      ops_zero_cpu(slab_.bases(), d.b);
     }
   }
 
@@ -368,20 +427,20 @@ class Server {
   void run_gather_keys(const torch::Tensor& keys_cpu, torch::Tensor out) {
     if (dev_.is_cuda()) {
       auto kd = keys_cpu.to(dev_, /*non_blocking=*/true);
-      ops_gather_keys_gpu(slab_.data, key_batch(kd), out.data_ptr<float>(), current_stream(dev_));
+      ops_gather_keys_gpu(slab_.bases(), key_batch(kd), out.data_ptr<float>(), current_stream(dev_));
     } else {
       std::lock_guard<std::mutex> g(cpu_val_mu_);
-      ops_gather_keys_cpu(slab_.data, key_batch(keys_cpu), out.data_ptr<float>());
+      ops_gather_keys_cpu(slab_.bases(), key_batch(keys_cpu), out.data_ptr<float>());
     }
   }
   void run_scatter_keys(const torch::Tensor& keys_cpu, const torch::Tensor& in, bool set) {
     if (dev_.is_cuda()) {
       auto kd = keys_cpu.to(dev_, /*non_blocking=*/true);
-      ops_scatter_keys_gpu(slab_.data, key_batch(kd), in.data_ptr<float>(), set,
+      ops_scatter_keys_gpu(slab_.bases(), key_batch(kd), in.data_ptr<float>(), set,
                            current_stream(dev_));
     } else {
       std::lock_guard<std::mutex> g(cpu_val_mu_);
-      ops_scatter_keys_cpu(slab_.data, key_batch(keys_cpu), in.data_ptr<float>(), set);
+      ops_scatter_keys_cpu(slab_.bases(), key_batch(keys_cpu), in.data_ptr<float>(), set);
     }
   }
 
@@ -1552,6 +1611,8 @@ class Server {
     d["sampling_checks"] = stat_sampling_checks_.load();
     d["slab_in_use"] = slab_.in_use.load();
     d["slab_capacity"] = slab_.capacity;
+    d["host_spill_in_use"] = slab_.host_in_use.load();
+    d["host_spill_capacity"] = slab_.host_capacity;
     int64_t rounds = 0;
     for (auto& c : channels_) rounds += c.rounds.load();
     d["sync_rounds"] = rounds;
@@ -1727,11 +1788,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("w2v_sgns_step", &w2v_sgns_step, py::call_guard<py::gil_scoped_release>());
   m.def("mf_update_step", &mf_update_step, py::call_guard<py::gil_scoped_release>());
   py::class_<Server>(m, "Server")
-      .def(py::init<int64_t, torch::Tensor, int, int, int, int, std::string, double, int, bool>(),
+      .def(py::init<int64_t, torch::Tensor, int, int, int, int, std::string, double, int, bool,
+                    int64_t, int64_t>(),
            py::arg("num_keys"), py::arg("value_lengths"), py::arg("rank"), py::arg("world"),
            py::arg("num_channels"), py::arg("num_workers"), py::arg("device"),
            py::arg("capacity_factor") = 2.0, py::arg("techniques") = 0,
-           py::arg("location_caches") = true)
+           py::arg("location_caches") = true, py::arg("device_cap_floats") = 0,
+           py::arg("host_spill_floats") = 0)
       .def("pull", &Server::pull, py::call_guard<py::gil_scoped_release>())
       .def("push", &Server::push, py::call_guard<py::gil_scoped_release>())
       .def("pull_if_local", &Server::pull_if_local, py::call_guard<py::gil_scoped_release>())

@@ -15,6 +15,16 @@
 
 namespace adapm {
 
+// offsets with bit 62 set live in the host-spill arena (pinned host
+// memory, device-visible zero-copy): kernels decode the bit and pick the
+// base pointer. See Slab host spill in core.cpp.
+constexpr int64_t SPILL_BIT = 1LL << 62;
+
+struct SlabBases {
+  float* dev;   // HBM slab
+  float* host;  // device-visible pointer to the pinned host arena (may be null)
+};
+
 struct OpsBatch {
   const int64_t* src_off;  // per-key slab offset (floats), or -1 to skip
   const int64_t* dst_off;  // per-key offset into the out/in buffer (floats)
@@ -23,29 +33,29 @@ struct OpsBatch {
 };
 
 // out[dst_off[i] : +len] = slab[src_off[i] : +len]
-void ops_gather_gpu(const float* slab, const OpsBatch& b, float* out, void* stream);
-void ops_gather_cpu(const float* slab, const OpsBatch& b, float* out);
+void ops_gather_gpu(const SlabBases& slab, const OpsBatch& b, float* out, void* stream);
+void ops_gather_cpu(const SlabBases& slab, const OpsBatch& b, float* out);
 
 // slab[dst(src)_off[i]] += in[...]   (atomic on GPU)  — or assign when set=true
-void ops_scatter_gpu(float* slab, const OpsBatch& b, const float* in, bool set, void* stream);
-void ops_scatter_cpu(float* slab, const OpsBatch& b, const float* in, bool set);
+void ops_scatter_gpu(const SlabBases& slab, const OpsBatch& b, const float* in, bool set, void* stream);
+void ops_scatter_cpu(const SlabBases& slab, const OpsBatch& b, const float* in, bool set);
 
 // replica delta extraction (src_off = val offsets, dst_off = out buffer
 // offsets, aux_off = sync_state offsets):
 //   v = val[e]; out[e] = v - sync[e]; sync[e] = v
 // single-read-per-element so a concurrent atomic push is never lost (it
 // stays in val and is extracted next round).
-void ops_extract_gpu(float* slab, const OpsBatch& b, const int64_t* sync_off, float* out, void* stream);
-void ops_extract_cpu(float* slab, const OpsBatch& b, const int64_t* sync_off, float* out);
+void ops_extract_gpu(const SlabBases& slab, const OpsBatch& b, const int64_t* sync_off, float* out, void* stream);
+void ops_extract_cpu(const SlabBases& slab, const OpsBatch& b, const int64_t* sync_off, float* out);
 
 // replica refresh apply (delta form so concurrent pushes are preserved):
 //   s = state_in[e]; atomicAdd(&val[e], s - sync[e]); sync[e] = s
-void ops_refresh_gpu(float* slab, const OpsBatch& b, const int64_t* sync_off, const float* state_in, void* stream);
-void ops_refresh_cpu(float* slab, const OpsBatch& b, const int64_t* sync_off, const float* state_in);
+void ops_refresh_gpu(const SlabBases& slab, const OpsBatch& b, const int64_t* sync_off, const float* state_in, void* stream);
+void ops_refresh_cpu(const SlabBases& slab, const OpsBatch& b, const int64_t* sync_off, const float* state_in);
 
 // slab[dst_off[i] : +len] = 0
-void ops_zero_gpu(float* slab, const OpsBatch& b, void* stream);
-void ops_zero_cpu(float* slab, const OpsBatch& b);
+void ops_zero_gpu(const SlabBases& slab, const OpsBatch& b, void* stream);
+void ops_zero_cpu(const SlabBases& slab, const OpsBatch& b);
 
 // Identity-layout direct ops: key k (owned: k % world == rank) lives at
 // slab offset (k / world) * plen; out/in row i sits at i * len. Keys not
@@ -59,10 +69,10 @@ struct KeyBatch {
   int32_t plen;   // padded slot length
   int world, rank;
 };
-void ops_gather_keys_gpu(const float* slab, const KeyBatch& b, float* out, void* stream);
-void ops_gather_keys_cpu(const float* slab, const KeyBatch& b, float* out);
-void ops_scatter_keys_gpu(float* slab, const KeyBatch& b, const float* in, bool set, void* stream);
-void ops_scatter_keys_cpu(float* slab, const KeyBatch& b, const float* in, bool set);
+void ops_gather_keys_gpu(const SlabBases& slab, const KeyBatch& b, float* out, void* stream);
+void ops_gather_keys_cpu(const SlabBases& slab, const KeyBatch& b, float* out);
+void ops_scatter_keys_gpu(const SlabBases& slab, const KeyBatch& b, const float* in, bool set, void* stream);
+void ops_scatter_keys_cpu(const SlabBases& slab, const KeyBatch& b, const float* in, bool set);
 
 bool hip_available();
 

@@ -7,18 +7,29 @@
 
 namespace adapm {
 
-void ops_gather_cpu(const float* slab, const OpsBatch& b, float* out) {
+static inline const float* sel_base_c(const SlabBases& sb, int64_t& off) {
+  if (off & SPILL_BIT) { off &= ~SPILL_BIT; return sb.host; }
+  return sb.dev;
+}
+static inline float* sel_base(const SlabBases& sb, int64_t& off) {
+  if (off & SPILL_BIT) { off &= ~SPILL_BIT; return sb.host; }
+  return sb.dev;
+}
+
+void ops_gather_cpu(const SlabBases& sb, const OpsBatch& b, float* out) {
   for (int i = 0; i < b.n; ++i) {
     int64_t s = b.src_off[i];
     if (s < 0) continue;
+    const float* slab = sel_base_c(sb, s);
     std::memcpy(out + b.dst_off[i], slab + s, sizeof(float) * b.lens[i]);
   }
 }
 
-void ops_scatter_cpu(float* slab, const OpsBatch& b, const float* in, bool set) {
+void ops_scatter_cpu(const SlabBases& sb, const OpsBatch& b, const float* in, bool set) {
   for (int i = 0; i < b.n; ++i) {
     int64_t s = b.src_off[i];
     if (s < 0) continue;
+    float* slab = sel_base(sb, s);
     int64_t d = b.dst_off[i];
     int32_t len = b.lens[i];
     if (set) {
@@ -29,48 +40,52 @@ void ops_scatter_cpu(float* slab, const OpsBatch& b, const float* in, bool set) 
   }
 }
 
-void ops_extract_cpu(float* slab, const OpsBatch& b, const int64_t* sync_off, float* out) {
+void ops_extract_cpu(const SlabBases& sb, const OpsBatch& b, const int64_t* sync_off, float* out) {
   for (int i = 0; i < b.n; ++i) {
     int64_t v = b.src_off[i];
     if (v < 0) continue;
     int64_t o = b.dst_off[i], sy = sync_off[i];
+    float* vb = sel_base(sb, v);
+    float* syb = sel_base(sb, sy);
     int32_t len = b.lens[i];
     for (int e = 0; e < len; ++e) {
-      float cur = slab[v + e];
-      out[o + e] = cur - slab[sy + e];
-      slab[sy + e] = cur;
+      float cur = vb[v + e];
+      out[o + e] = cur - syb[sy + e];
+      syb[sy + e] = cur;
     }
   }
 }
 
-void ops_refresh_cpu(float* slab, const OpsBatch& b, const int64_t* sync_off, const float* state_in) {
+void ops_refresh_cpu(const SlabBases& sb, const OpsBatch& b, const int64_t* sync_off, const float* state_in) {
   for (int i = 0; i < b.n; ++i) {
     int64_t v = b.src_off[i];
     if (v < 0) continue;
     int64_t o = b.dst_off[i], sy = sync_off[i];
+    float* vb = sel_base(sb, v);
+    float* syb = sel_base(sb, sy);
     int32_t len = b.lens[i];
     for (int e = 0; e < len; ++e) {
       float s = state_in[o + e];
-      slab[v + e] += s - slab[sy + e];
-      slab[sy + e] = s;
+      vb[v + e] += s - syb[sy + e];
+      syb[sy + e] = s;
     }
   }
 }
 
-void ops_gather_keys_cpu(const float* slab, const KeyBatch& b, float* out) {
+void ops_gather_keys_cpu(const SlabBases& sb, const KeyBatch& b, float* out) {
   for (int i = 0; i < b.n; ++i) {
     int64_t k = b.keys[i];
     if ((int)(k % b.world) != b.rank) continue;
-    std::memcpy(out + (int64_t)i * b.len, slab + (k / b.world) * (int64_t)b.plen,
+    std::memcpy(out + (int64_t)i * b.len, sb.dev + (k / b.world) * (int64_t)b.plen,
                 sizeof(float) * b.len);
   }
 }
 
-void ops_scatter_keys_cpu(float* slab, const KeyBatch& b, const float* in, bool set) {
+void ops_scatter_keys_cpu(const SlabBases& sb, const KeyBatch& b, const float* in, bool set) {
   for (int i = 0; i < b.n; ++i) {
     int64_t k = b.keys[i];
     if ((int)(k % b.world) != b.rank) continue;
-    float* s = slab + (k / b.world) * (int64_t)b.plen;
+    float* s = sb.dev + (k / b.world) * (int64_t)b.plen;
     const float* d = in + (int64_t)i * b.len;
     if (set) {
       std::memcpy(s, d, sizeof(float) * b.len);
@@ -80,10 +95,11 @@ void ops_scatter_keys_cpu(float* slab, const KeyBatch& b, const float* in, bool 
   }
 }
 
-void ops_zero_cpu(float* slab, const OpsBatch& b) {
+void ops_zero_cpu(const SlabBases& sb, const OpsBatch& b) {
   for (int i = 0; i < b.n; ++i) {
     int64_t d = b.dst_off[i];
     if (d < 0) continue;
+    float* slab = sel_base(sb, d);
     std::memset(slab + d, 0, sizeof(float) * b.lens[i]);
   }
 }

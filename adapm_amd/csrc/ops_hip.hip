@@ -13,16 +13,23 @@ namespace adapm {
 
 #define THREADS 256
 
+__device__ inline float* sel_base(float* dev, float* host, int64_t& off) {
+  if (off & SPILL_BIT) { off &= ~SPILL_BIT; return host; }
+  return dev;
+}
+
 __device__ inline bool vec_ok(int64_t a, int64_t b, int32_t len) {
   return ((a | b) & 3) == 0 && (len & 3) == 0;
 }
 
-__global__ void k_gather(const float* __restrict__ slab, const int64_t* __restrict__ src_off,
+__global__ void k_gather(float* __restrict__ sdev, float* __restrict__ shost,
+                         const int64_t* __restrict__ src_off,
                          const int64_t* __restrict__ dst_off, const int32_t* __restrict__ lens,
                          int n, float* __restrict__ out) {
   for (int i = blockIdx.x; i < n; i += gridDim.x) {
     int64_t s = src_off[i];
     if (s < 0) continue;
+    const float* slab = sel_base(sdev, shost, s);
     int64_t d = dst_off[i];
     int32_t len = lens[i];
     if (vec_ok(s, d, len)) {
@@ -35,24 +42,28 @@ __global__ void k_gather(const float* __restrict__ slab, const int64_t* __restri
   }
 }
 
-__global__ void k_scatter_add(float* __restrict__ slab, const int64_t* __restrict__ src_off,
+__global__ void k_scatter_add(float* __restrict__ sdev, float* __restrict__ shost,
+                              const int64_t* __restrict__ src_off,
                               const int64_t* __restrict__ dst_off, const int32_t* __restrict__ lens,
                               int n, const float* __restrict__ in) {
   for (int i = blockIdx.x; i < n; i += gridDim.x) {
     int64_t s = src_off[i];
     if (s < 0) continue;
+    float* slab = sel_base(sdev, shost, s);
     int64_t d = dst_off[i];
     int32_t len = lens[i];
     for (int e = threadIdx.x; e < len; e += THREADS) atomicAdd(&slab[s + e], in[d + e]);
   }
 }
 
-__global__ void k_scatter_set(float* __restrict__ slab, const int64_t* __restrict__ src_off,
+__global__ void k_scatter_set(float* __restrict__ sdev, float* __restrict__ shost,
+                              const int64_t* __restrict__ src_off,
                               const int64_t* __restrict__ dst_off, const int32_t* __restrict__ lens,
                               int n, const float* __restrict__ in) {
   for (int i = blockIdx.x; i < n; i += gridDim.x) {
     int64_t s = src_off[i];
     if (s < 0) continue;
+    float* slab = sel_base(sdev, shost, s);
     int64_t d = dst_off[i];
     int32_t len = lens[i];
     if (vec_ok(s, d, len)) {
@@ -65,23 +76,27 @@ __global__ void k_scatter_set(float* __restrict__ slab, const int64_t* __restric
   }
 }
 
-__global__ void k_extract(float* __restrict__ slab, const int64_t* __restrict__ val_off,
+__global__ void k_extract(float* __restrict__ sdev, float* __restrict__ shost,
+                          const int64_t* __restrict__ val_off,
                           const int64_t* __restrict__ out_off, const int32_t* __restrict__ lens,
                           int n, const int64_t* __restrict__ sync_off, float* __restrict__ out) {
   for (int i = blockIdx.x; i < n; i += gridDim.x) {
     int64_t v = val_off[i];
     if (v < 0) continue;
     int64_t o = out_off[i], sy = sync_off[i];
+    float* vb = sel_base(sdev, shost, v);
+    float* sb = sel_base(sdev, shost, sy);
     int32_t len = lens[i];
     for (int e = threadIdx.x; e < len; e += THREADS) {
-      float cur = slab[v + e];        // single read: a concurrent atomicAdd
-      out[o + e] = cur - slab[sy + e];  // after this read stays in val and is
-      slab[sy + e] = cur;             // captured by the next round's extract
+      float cur = vb[v + e];        // single read: a concurrent atomicAdd
+      out[o + e] = cur - sb[sy + e];  // after this read stays in val and is
+      sb[sy + e] = cur;             // captured by the next round's extract
     }
   }
 }
 
-__global__ void k_refresh(float* __restrict__ slab, const int64_t* __restrict__ val_off,
+__global__ void k_refresh(float* __restrict__ sdev, float* __restrict__ shost,
+                          const int64_t* __restrict__ val_off,
                           const int64_t* __restrict__ in_off, const int32_t* __restrict__ lens,
                           int n, const int64_t* __restrict__ sync_off,
                           const float* __restrict__ state_in) {
@@ -89,20 +104,24 @@ __global__ void k_refresh(float* __restrict__ slab, const int64_t* __restrict__ 
     int64_t v = val_off[i];
     if (v < 0) continue;
     int64_t o = in_off[i], sy = sync_off[i];
+    float* vb = sel_base(sdev, shost, v);
+    float* sb = sel_base(sdev, shost, sy);
     int32_t len = lens[i];
     for (int e = threadIdx.x; e < len; e += THREADS) {
       float s = state_in[o + e];
-      atomicAdd(&slab[v + e], s - slab[sy + e]);  // delta form: preserves
-      slab[sy + e] = s;                           // concurrent pushes
+      atomicAdd(&vb[v + e], s - sb[sy + e]);  // delta form: preserves
+      sb[sy + e] = s;                         // concurrent pushes
     }
   }
 }
 
-__global__ void k_zero(float* __restrict__ slab, const int64_t* __restrict__ dst_off,
+__global__ void k_zero(float* __restrict__ sdev, float* __restrict__ shost,
+                       const int64_t* __restrict__ dst_off,
                        const int32_t* __restrict__ lens, int n) {
   for (int i = blockIdx.x; i < n; i += gridDim.x) {
     int64_t d = dst_off[i];
     if (d < 0) continue;
+    float* slab = sel_base(sdev, shost, d);
     int32_t len = lens[i];
     for (int e = threadIdx.x; e < len; e += THREADS) slab[d + e] = 0.f;
   }
@@ -162,49 +181,49 @@ static inline int grid_for(int n) {
   return g > 16384 ? 16384 : g;
 }
 
-void ops_gather_gpu(const float* slab, const OpsBatch& b, float* out, void* stream) {
+void ops_gather_gpu(const SlabBases& slab, const OpsBatch& b, float* out, void* stream) {
   if (b.n == 0) return;
   hipLaunchKernelGGL(k_gather, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
-                     slab, b.src_off, b.dst_off, b.lens, b.n, out);
+                     slab.dev, slab.host, b.src_off, b.dst_off, b.lens, b.n, out);
 }
-void ops_scatter_gpu(float* slab, const OpsBatch& b, const float* in, bool set, void* stream) {
+void ops_scatter_gpu(const SlabBases& slab, const OpsBatch& b, const float* in, bool set, void* stream) {
   if (b.n == 0) return;
   if (set)
     hipLaunchKernelGGL(k_scatter_set, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
-                       slab, b.src_off, b.dst_off, b.lens, b.n, in);
+                       slab.dev, slab.host, b.src_off, b.dst_off, b.lens, b.n, in);
   else
     hipLaunchKernelGGL(k_scatter_add, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
-                       slab, b.src_off, b.dst_off, b.lens, b.n, in);
+                       slab.dev, slab.host, b.src_off, b.dst_off, b.lens, b.n, in);
 }
-void ops_extract_gpu(float* slab, const OpsBatch& b, const int64_t* sync_off, float* out, void* stream) {
+void ops_extract_gpu(const SlabBases& slab, const OpsBatch& b, const int64_t* sync_off, float* out, void* stream) {
   if (b.n == 0) return;
   hipLaunchKernelGGL(k_extract, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
-                     slab, b.src_off, b.dst_off, b.lens, b.n, sync_off, out);
+                     slab.dev, slab.host, b.src_off, b.dst_off, b.lens, b.n, sync_off, out);
 }
-void ops_refresh_gpu(float* slab, const OpsBatch& b, const int64_t* sync_off, const float* state_in, void* stream) {
+void ops_refresh_gpu(const SlabBases& slab, const OpsBatch& b, const int64_t* sync_off, const float* state_in, void* stream) {
   if (b.n == 0) return;
   hipLaunchKernelGGL(k_refresh, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
-                     slab, b.src_off, b.dst_off, b.lens, b.n, sync_off, state_in);
+                     slab.dev, slab.host, b.src_off, b.dst_off, b.lens, b.n, sync_off, state_in);
 }
-void ops_zero_gpu(float* slab, const OpsBatch& b, void* stream) {
+void ops_zero_gpu(const SlabBases& slab, const OpsBatch& b, void* stream) {
   if (b.n == 0) return;
   hipLaunchKernelGGL(k_zero, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
-                     slab, b.dst_off, b.lens, b.n);
+                     slab.dev, slab.host, b.dst_off, b.lens, b.n);
 }
 
-void ops_gather_keys_gpu(const float* slab, const KeyBatch& b, float* out, void* stream) {
-  if (b.n == 0) return;
+void ops_gather_keys_gpu(const SlabBases& slab, const KeyBatch& b, float* out, void* stream) {
+  if (b.n == 0) return;  // identity layout never spills
   hipLaunchKernelGGL(k_gather_keys, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
-                     slab, b.keys, b.n, b.len, b.plen, b.world, b.rank, out);
+                     slab.dev, b.keys, b.n, b.len, b.plen, b.world, b.rank, out);
 }
-void ops_scatter_keys_gpu(float* slab, const KeyBatch& b, const float* in, bool set, void* stream) {
+void ops_scatter_keys_gpu(const SlabBases& slab, const KeyBatch& b, const float* in, bool set, void* stream) {
   if (b.n == 0) return;
   if (set)
     hipLaunchKernelGGL(k_scatter_set_keys, dim3(grid_for(b.n)), dim3(THREADS), 0,
-                       (hipStream_t)stream, slab, b.keys, b.n, b.len, b.plen, b.world, b.rank, in);
+                       (hipStream_t)stream, slab.dev, b.keys, b.n, b.len, b.plen, b.world, b.rank, in);
   else
     hipLaunchKernelGGL(k_scatter_add_keys, dim3(grid_for(b.n)), dim3(THREADS), 0,
-                       (hipStream_t)stream, slab, b.keys, b.n, b.len, b.plen, b.world, b.rank, in);
+                       (hipStream_t)stream, slab.dev, b.keys, b.n, b.len, b.plen, b.world, b.rank, in);
 }
 
 bool hip_available() {

@@ -27,6 +27,7 @@ def torch_paths():
 
 
 SOURCES = [
+    "slab_host.cpp",
     "ops_cpu.cpp",
     "ops_hip.hip",
     "core.cpp",
