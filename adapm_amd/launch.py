@@ -42,6 +42,7 @@ def main():
         env = dict(os.environ)
         env.update(RANK=str(rank), LOCAL_RANK=str(rank), WORLD_SIZE=str(args.num_ranks),
                    MASTER_ADDR=args.master_addr, MASTER_PORT=str(port))
+        env["PYTHONPATH"] = os.getcwd() + os.pathsep + env.get("PYTHONPATH", "")
         return subprocess.Popen([sys.executable] + args.cmd, env=env)
 
     procs = {r: spawn(r) for r in range(args.num_ranks)}
