@@ -69,6 +69,8 @@ def init_runtime(num_channels: int = 2, device: str | None = None,
         torch.cuda.set_device(dev)
 
     backend = "nccl" if dev.type == "cuda" else "gloo"
+    if os.environ.get("ADAPM_FORCE_GLOO", "0") == "1":
+        backend = "gloo"  # e.g. several ranks sharing one GPU for testing
     if world > 1:
         if not dist.is_initialized():
             os.environ.setdefault("MASTER_ADDR", _env("DMLC_PS_ROOT_URI", default="127.0.0.1"))

@@ -37,6 +37,8 @@ def main():
     ap.add_argument("--triples", type=int, default=90_000_000,
                     help="nominal epoch size (for epoch-time reporting)")
     ap.add_argument("--lookahead", type=int, default=4)
+    ap.add_argument("--no-intent", action="store_true",
+                    help="disable intent signaling (pure remote-op mode)")
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--channels", type=int, default=2)
     ap.add_argument("--capacity-factor", type=float, default=3.0)
@@ -81,9 +83,10 @@ def main():
     batches = [make_batch(i) for i in range(total_steps + args.lookahead)]
 
     def run_step(i, sync_loss=False):
-        model.signal_intent(batches[i + args.lookahead],
-                            worker.current_clock() + args.lookahead,
-                            worker.current_clock() + args.lookahead + 2)
+        if not args.no_intent:
+            model.signal_intent(batches[i + args.lookahead],
+                                worker.current_clock() + args.lookahead,
+                                worker.current_clock() + args.lookahead + 2)
         loss = model.train_batch(batches[i], sync_loss=sync_loss)
         worker.advance_clock()
         return loss
