@@ -248,9 +248,10 @@ class Worker:
         vt, _ = self._vals_tensor(vals)
         self._check_len(kt, vt)
         ts = self._s.pull(self.wid, kt, vt)
+        if ts != -1:
+            self.server._sync.kick_event.set()
         if not async_:
             self._s.wait(ts)
-            return ts
         return ts
 
     def push(self, keys, vals, async_: bool = False, **kw):
@@ -259,6 +260,8 @@ class Worker:
         vt, _ = self._vals_tensor(vals)
         self._check_len(kt, vt)
         ts = self._s.push(self.wid, kt, vt, False)
+        if ts != -1:
+            self.server._sync.kick_event.set()
         if not async_:
             self._s.wait(ts)
         return ts
@@ -269,6 +272,8 @@ class Worker:
         vt, _ = self._vals_tensor(vals)
         self._check_len(kt, vt)
         ts = self._s.push(self.wid, kt, vt, True)
+        if ts != -1:
+            self.server._sync.kick_event.set()
         if not async_:
             self._s.wait(ts)
         return ts

@@ -45,6 +45,13 @@ enum MsgCode : int64_t {
   M_PULL_RESP = 11,  // f0=req_id, f1=out_index. payload: len floats
   M_PUSH_ACK = 12,   // f0=req_id, f1=count acked
   M_RESIDENCE = 13,  // ->manager. f0=new_owner, f1=relocation counter
+  // Bulk records (uniform-length stores): the 5-word header is followed
+  // IN THE META STREAM by extra int64 words (keys, and out-indices for
+  // pulls). They collapse thousands of per-key records into one record +
+  // one batched kernel on each side — the multi-GPU hot path.
+  M_PULL_REQ_BULK = 4,   // hdr(code, nkeys, origin, req_id, hops) + keys + out_idx
+  M_PUSH_REQ_BULK = 5,   // hdr(code, nkeys, origin, req_id, set<<32|hops) + keys. payload: nkeys rows
+  M_PULL_RESP_BULK = 14, // hdr(code, nkeys, req_id, 0, 0) + keys + out_idx. payload: nkeys rows
 };
 
 enum DeltaFlags : int64_t {

@@ -38,6 +38,7 @@
 #include <condition_variable>
 #include <cstring>
 #include <deque>
+#include <map>
 #include <memory>
 #include <mutex>
 #include <thread>
@@ -183,17 +184,22 @@ struct Ticket {
 struct OutRec {
   int dest;
   int64_t code, key, f0, f1, f2;
-  torch::Tensor payload;  // may be undefined
+  torch::Tensor payload;       // may be undefined
+  std::vector<Key> keys;       // bulk records: keys (key field unused)
+  std::vector<int64_t> aux;    // bulk pulls: out indices
 };
 
 // phase-B pending response with deferred payload gather from the slab
 struct RespRec {
-  int dest;
-  int64_t code, key, f0, f1, f2;
+  int dest = 0;
+  int64_t code = 0, key = 0, f0 = 0, f1 = 0, f2 = 0;
   int64_t slab_off = -1;    // gather source (refresh / pull resp / relocate)
   int32_t len = 0;
   bool free_after = false;  // relocation: free the slot after gathering
   torch::Tensor payload;    // alternative payload source
+  std::vector<Key> keys;    // bulk responses
+  std::vector<int64_t> aux;       // bulk pulls: out indices
+  std::vector<int64_t> slab_offs; // bulk responses: per-key gather sources
 };
 
 struct IntentReq {
@@ -568,8 +574,25 @@ class Server {
       t->out_len = std::move(out_len);
       tickets_[ts] = std::move(t);
     }
-    for (auto& r : remote) {
-      enqueue_out(channel_of(r.k), OutRec{directions(r.k), M_PULL_REQ, r.k, rank_, ts, r.out_index, {}});
+    if (uniform_len_ >= 0) {
+      // bulk requests: one record per (channel, destination)
+      std::map<std::pair<int, int>, std::pair<std::vector<Key>, std::vector<int64_t>>> groups;
+      for (auto& r : remote) {
+        auto& g = groups[{channel_of(r.k), directions(r.k)}];
+        g.first.push_back(r.k);
+        g.second.push_back(r.out_index);
+      }
+      for (auto& [cd, g] : groups) {
+        OutRec rec{cd.second, M_PULL_REQ_BULK, 0, rank_, ts, 0, {}};
+        rec.keys = std::move(g.first);
+        rec.aux = std::move(g.second);
+        enqueue_out(cd.first, std::move(rec));
+      }
+    } else {
+      for (auto& r : remote) {
+        enqueue_out(channel_of(r.k),
+                    OutRec{directions(r.k), M_PULL_REQ, r.k, rank_, ts, r.out_index, {}});
+      }
     }
     return ts;
   }
@@ -689,10 +712,31 @@ class Server {
       t->expected = (int)remote.size();
       tickets_[ts] = std::move(t);
     }
-    for (auto& r : remote) {
-      torch::Tensor pay = flat.narrow(0, r.off, r.len).clone();
-      enqueue_out(channel_of(r.k),
-                  OutRec{directions(r.k), set_mode ? M_SET_REQ : M_PUSH_REQ, r.k, rank_, ts, 0, pay});
+    if (uniform_len_ >= 0) {
+      const int32_t l = uniform_len_;
+      std::map<std::pair<int, int>, std::pair<std::vector<Key>, std::vector<int64_t>>> groups;
+      for (auto& r : remote) {
+        auto& g = groups[{channel_of(r.k), directions(r.k)}];
+        g.first.push_back(r.k);
+        g.second.push_back(r.off / l);  // row index into flat
+      }
+      auto rows = flat.view({n, (int64_t)l});
+      for (auto& [cd, g] : groups) {
+        auto idx = torch::from_blob(g.second.data(), {(int64_t)g.second.size()},
+                                    torch::TensorOptions().dtype(torch::kInt64))
+                       .clone();
+        if (dev_.is_cuda()) idx = idx.to(dev_, true);
+        OutRec rec{cd.second, M_PUSH_REQ_BULK, 0, rank_, ts,
+                   set_mode ? (1LL << 32) : 0, rows.index_select(0, idx).reshape({-1})};
+        rec.keys = std::move(g.first);
+        enqueue_out(cd.first, std::move(rec));
+      }
+    } else {
+      for (auto& r : remote) {
+        torch::Tensor pay = flat.narrow(0, r.off, r.len).clone();
+        enqueue_out(channel_of(r.k),
+                    OutRec{directions(r.k), set_mode ? M_SET_REQ : M_PUSH_REQ, r.k, rank_, ts, 0, pay});
+      }
     }
     return ts;
   }
@@ -964,6 +1008,26 @@ class Server {
       }
     }
     for (auto& r : ops_out) {
+      if (r.code == M_PULL_REQ_BULK || r.code == M_PUSH_REQ_BULK) {
+        if (r.dest == rank_) {  // stale direction: handle locally
+          if (r.code == M_PULL_REQ_BULK)
+            handle_pull_bulk(ch, C, (int)r.f0, r.f1, (int)r.f2, r.keys.data(), r.aux.data(),
+                             (int64_t)r.keys.size());
+          else
+            handle_push_bulk(ch, C, (int)r.f0, r.f1, r.f2, r.keys.data(),
+                             (int64_t)r.keys.size(), r.payload);
+          continue;
+        }
+        Msg& m = add_rec(r.dest, r.code, (int64_t)r.keys.size(), r.f0, r.f1, r.f2);
+        m.meta.insert(m.meta.end(), r.keys.begin(), r.keys.end());
+        if (r.code == M_PULL_REQ_BULK)
+          m.meta.insert(m.meta.end(), r.aux.begin(), r.aux.end());
+        if (r.payload.defined()) {
+          m.copies.push_back({r.payload, m.payload_floats});
+          m.payload_floats += r.payload.numel();
+        }
+        continue;
+      }
       int dest = directions(r.key);  // re-resolve at send time
       if (dest == rank_) {
         apply_local_record(ch, C, r);
@@ -979,8 +1043,7 @@ class Server {
     // 5. materialize tensors; extract kernels write into the payloads
     std::vector<std::tuple<int, torch::Tensor, torch::Tensor>> out;
     for (auto& [dest, m] : msgs) {
-      int64_t n_rec = (int64_t)m.meta.size() / REC_I64;
-      auto meta = torch::from_blob(m.meta.data(), {n_rec, REC_I64},
+      auto meta = torch::from_blob(m.meta.data(), {(int64_t)m.meta.size()},
                                    torch::TensorOptions().dtype(torch::kInt64))
                       .clone();
       auto payload = torch::empty({m.payload_floats},
@@ -1100,24 +1163,138 @@ class Server {
     enqueue_out(ch, std::move(nr));
   }
 
+  // bulk pull request: serve owned keys from the slab (response gathered
+  // at respond time, after quiesce), forward the rest per destination.
+  void handle_pull_bulk(int ch, ChannelState& C, int origin, int64_t req_id, int hops,
+                        const Key* keys, const int64_t* oidx, int64_t nk) {
+    const int32_t l = uniform_len_;
+    TORCH_CHECK(l >= 0, "bulk records need a uniform-length store");
+    RespRec resp;
+    resp.dest = origin;
+    resp.code = M_PULL_RESP_BULK;
+    resp.f0 = req_id;
+    std::map<int, std::pair<std::vector<Key>, std::vector<int64_t>>> fwd;
+    for (int64_t i = 0; i < nk; ++i) {
+      Key k = keys[i];
+      uint8_t f = flags_[k].load(std::memory_order_acquire);
+      if (f & F_OWNER) {
+        resp.keys.push_back(k);
+        resp.aux.push_back(oidx[i]);
+        resp.slab_offs.push_back(loc_[k].load(std::memory_order_acquire));
+      } else {
+        auto& g = fwd[directions(k)];
+        g.first.push_back(k);
+        g.second.push_back(oidx[i]);
+      }
+    }
+    if (!resp.keys.empty()) {
+      stat_remote_pulls_served_ += (int64_t)resp.keys.size();
+      resp.key = (int64_t)resp.keys.size();
+      std::lock_guard<std::mutex> g(C.mu);
+      C.responses.push_back(std::move(resp));
+    }
+    for (auto& [d, g] : fwd) {
+      if (hops >= 64) {
+        stat_dropped_records_ += (int64_t)g.first.size();
+        continue;
+      }
+      stat_forwards_ += (int64_t)g.first.size();
+      OutRec rec{d == rank_ ? manager_of(g.first[0]) : d, M_PULL_REQ_BULK, 0, origin, req_id,
+                 hops + 1, {}};
+      rec.keys = std::move(g.first);
+      rec.aux = std::move(g.second);
+      enqueue_out(ch, std::move(rec));
+    }
+  }
+
+  // bulk push request: batched merge into owned rows, ack count, forward
+  // the rest (with their payload rows subset).
+  void handle_push_bulk(int ch, ChannelState& C, int origin, int64_t req_id, int64_t f2,
+                        const Key* keys, int64_t nk, torch::Tensor rows_flat) {
+    const int32_t l = uniform_len_;
+    TORCH_CHECK(l >= 0, "bulk records need a uniform-length store");
+    bool set_mode = (f2 >> 32) & 1;
+    int hops = (int)(f2 & 0xffffffff);
+    HostBatch apply;
+    std::map<int, std::pair<std::vector<Key>, std::vector<int64_t>>> fwd;  // keys, row idx
+    int64_t applied = 0;
+    for (int64_t i = 0; i < nk; ++i) {
+      Key k = keys[i];
+      uint8_t f = flags_[k].load(std::memory_order_acquire);
+      if (f & F_OWNER) {
+        apply.add(loc_[k].load(std::memory_order_acquire), i * (int64_t)l, l);
+        version_[k].fetch_add(1, std::memory_order_relaxed);
+        applied++;
+      } else {
+        auto& g = fwd[directions(k)];
+        g.first.push_back(k);
+        g.second.push_back(i);
+      }
+    }
+    run_scatter(apply, rows_flat, set_mode);
+    if (applied > 0) {
+      stat_remote_pushes_served_ += applied;
+      if (origin == rank_) {
+        complete_ticket(req_id, (int)applied);
+      } else {
+        std::lock_guard<std::mutex> g(C.mu);
+        C.responses.push_back(RespRec{origin, M_PUSH_ACK, 0, req_id, applied, 0, -1, 0, false, {}});
+      }
+    }
+    auto rows = rows_flat.view({nk, (int64_t)l});
+    for (auto& [d, g] : fwd) {
+      if (hops >= 64) {
+        stat_dropped_records_ += (int64_t)g.first.size();
+        if (origin == rank_) complete_ticket(req_id, (int)g.first.size());
+        continue;
+      }
+      stat_forwards_ += (int64_t)g.first.size();
+      auto idx = torch::from_blob(g.second.data(), {(int64_t)g.second.size()},
+                                  torch::TensorOptions().dtype(torch::kInt64))
+                     .clone();
+      if (dev_.is_cuda()) idx = idx.to(dev_, true);
+      OutRec rec{d == rank_ ? manager_of(g.first[0]) : d, M_PUSH_REQ_BULK, 0, origin, req_id,
+                 (set_mode ? (1LL << 32) : 0) | (hops + 1), rows.index_select(0, idx).reshape({-1})};
+      rec.keys = std::move(g.first);
+      enqueue_out(ch, std::move(rec));
+    }
+  }
+
   // ------------------------------------------------ sync round: phase A in
 
   void sync_process(int ch, int src, torch::Tensor meta, torch::Tensor payload) {
     ChannelState& C = channels_[ch];
     meta = meta.contiguous();
-    int64_t n_rec = meta.numel() / REC_I64;
+    int64_t n_words = meta.numel();
     const int64_t* mp = meta.data_ptr<int64_t>();
     stat_bytes_recv_ += meta.numel() * 8 + payload.numel() * 4;
 
     HostBatch merges, assigns;
     int64_t poff = 0;
+    int64_t pos = 0;
 
-    for (int64_t i = 0; i < n_rec; ++i) {
-      int64_t code = mp[i * REC_I64 + 0];
-      Key k = mp[i * REC_I64 + 1];
-      int64_t f0 = mp[i * REC_I64 + 2];
-      int64_t f1 = mp[i * REC_I64 + 3];
-      int64_t f2 = mp[i * REC_I64 + 4];
+    while (pos + REC_I64 <= n_words) {
+      int64_t code = mp[pos + 0];
+      Key k = mp[pos + 1];
+      int64_t f0 = mp[pos + 2];
+      int64_t f1 = mp[pos + 3];
+      int64_t f2 = mp[pos + 4];
+      pos += REC_I64;
+      if (code == M_PULL_REQ_BULK) {
+        int64_t nk = k;
+        handle_pull_bulk(ch, C, (int)f0, f1, (int)f2, mp + pos, mp + pos + nk, nk);
+        pos += 2 * nk;
+        continue;
+      }
+      if (code == M_PUSH_REQ_BULK) {
+        int64_t nk = k;
+        int64_t rows_floats = nk * (int64_t)uniform_len_;
+        handle_push_bulk(ch, C, (int)f0, f1, f2, mp + pos, nk,
+                         payload.narrow(0, poff, rows_floats));
+        pos += nk;
+        poff += rows_floats;
+        continue;
+      }
       int32_t l = len_of(k);
 
       switch (code) {
@@ -1323,6 +1500,16 @@ class Server {
     for (auto& r : resp) {
       Msg& m = msgs[r.dest];
       m.meta.insert(m.meta.end(), {r.code, r.key, r.f0, r.f1, r.f2});
+      if (r.code == M_PULL_RESP_BULK) {
+        m.meta.insert(m.meta.end(), r.keys.begin(), r.keys.end());
+        m.meta.insert(m.meta.end(), r.aux.begin(), r.aux.end());
+        const int32_t l = uniform_len_;
+        for (size_t i = 0; i < r.slab_offs.size(); ++i) {
+          m.gathers.add(r.slab_offs[i], m.payload_floats, l);
+          m.payload_floats += l;
+        }
+        continue;
+      }
       if (r.slab_off >= 0) {
         m.gathers.add(r.slab_off, m.payload_floats, r.len);
         m.payload_floats += r.len;
@@ -1334,8 +1521,7 @@ class Server {
     }
     std::vector<std::tuple<int, torch::Tensor, torch::Tensor>> out;
     for (auto& [dest, m] : msgs) {
-      int64_t n_rec = (int64_t)m.meta.size() / REC_I64;
-      auto meta = torch::from_blob(m.meta.data(), {n_rec, REC_I64},
+      auto meta = torch::from_blob(m.meta.data(), {(int64_t)m.meta.size()},
                                    torch::TensorOptions().dtype(torch::kInt64))
                       .clone();
       auto payload = torch::empty({m.payload_floats},
@@ -1352,7 +1538,7 @@ class Server {
   void sync_apply(int ch, int src, torch::Tensor meta, torch::Tensor payload) {
     ChannelState& C = channels_[ch];
     meta = meta.contiguous();
-    int64_t n_rec = meta.numel() / REC_I64;
+    int64_t n_words = meta.numel();
     const int64_t* mp = meta.data_ptr<int64_t>();
     stat_bytes_recv_ += meta.numel() * 8 + payload.numel() * 4;
 
@@ -1373,12 +1559,21 @@ class Server {
     };
     std::vector<Post> posts;
     int64_t poff = 0;
+    int64_t pos = 0;
 
-    for (int64_t i = 0; i < n_rec; ++i) {
-      int64_t code = mp[i * REC_I64 + 0];
-      Key k = mp[i * REC_I64 + 1];
-      int64_t f0 = mp[i * REC_I64 + 2];
-      int64_t f1 = mp[i * REC_I64 + 3];
+    while (pos + REC_I64 <= n_words) {
+      int64_t code = mp[pos + 0];
+      Key k = mp[pos + 1];
+      int64_t f0 = mp[pos + 2];
+      int64_t f1 = mp[pos + 3];
+      pos += REC_I64;
+      if (code == M_PULL_RESP_BULK) {
+        int64_t nk = k;
+        apply_pull_resp_bulk(f0, mp + pos, mp + pos + nk, nk, payload, poff, src);
+        pos += 2 * nk;
+        poff += nk * (int64_t)uniform_len_;
+        continue;
+      }
       int32_t l = len_of(k);
 
       switch (code) {
@@ -1416,8 +1611,9 @@ class Server {
           break;
         }
         case M_PUSH_ACK: {
+          // NOTE: no location-cache update here — bulk acks aggregate
+          // many keys and carry no meaningful key field.
           complete_ticket(f0, (int)f1);
-          if (use_loc_cache_) loc_cache_[k] = src;
           break;
         }
         case M_RESIDENCE: {
@@ -1461,6 +1657,60 @@ class Server {
       }
     }
     for (auto& [off, len] : local_frees) slab_.free_(off, len);
+  }
+
+  // bulk pull response: ONE batched copy kernel from the payload rows into
+  // the ticket's output tensor (reusing the gather kernel with the payload
+  // as the source arena).
+  void apply_pull_resp_bulk(int64_t ts, const Key* keys, const int64_t* oidx, int64_t nk,
+                            torch::Tensor payload, int64_t poff, int src) {
+    const int32_t l = uniform_len_;
+    torch::Tensor out;
+    HostBatch hb;
+    int delivered = 0;
+    bool complete = false;
+    torch::Tensor caller_out;
+    {
+      std::lock_guard<std::mutex> g(tickets_mu_);
+      auto it = tickets_.find(ts);
+      if (it == tickets_.end()) return;
+      Ticket& t = *it->second;
+      out = t.out;
+      for (int64_t i = 0; i < nk; ++i) {
+        int64_t oi = oidx[i];
+        if (t.out_off[oi] < 0) continue;  // duplicate
+        hb.add(poff + i * (int64_t)l, t.out_off[oi], l);
+        t.out_off[oi] = -1 - t.out_off[oi];
+        delivered++;
+      }
+      t.received += delivered;
+      if (t.received >= t.expected) {
+        complete = true;
+        caller_out = t.caller_out;
+      }
+    }
+    if (hb.size()) {
+      SlabBases pb{payload.data_ptr<float>(), nullptr};
+      if (dev_.is_cuda()) {
+        auto d = to_dev(hb);
+        ops_gather_gpu(pb, d.b, out.data_ptr<float>(), current_stream(dev_));
+      } else {
+        auto d = to_dev(hb);
+        ops_gather_cpu(pb, d.b, out.data_ptr<float>());
+      }
+    }
+    if (use_loc_cache_) {
+      for (int64_t i = 0; i < nk; ++i) loc_cache_[keys[i]] = src;
+    }
+    if (complete) {
+      std::lock_guard<std::mutex> g(tickets_mu_);
+      auto it = tickets_.find(ts);
+      if (it != tickets_.end()) {
+        if (caller_out.defined()) caller_out.view({-1}).copy_(it->second->out.view({-1}));
+        tickets_.erase(it);
+        tickets_cv_.notify_all();
+      }
+    }
   }
 
   void deliver_pull(int64_t ts, int64_t out_index, torch::Tensor data) {

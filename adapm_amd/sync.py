@@ -35,14 +35,27 @@ import torch.distributed as dist
 _TRACE_KEY = int(os.environ.get("ADAPM_TRACE_KEY", "-1"))
 
 
+_BULK_CODES = {4: 2, 5: 1, 14: 2}  # code -> extra words per key
+
+
 def _trace(rank, ch, direction, peer, meta):
     if _TRACE_KEY == -1:
         return
-    m = meta.reshape(-1, 5)
-    for r in m.tolist():
-        if _TRACE_KEY == -2 or r[1] == _TRACE_KEY:
-            print(f"[trace r{rank} ch{ch} {direction} peer{peer}] code={r[0]} key={r[1]} "
-                  f"f0={r[2]} f1={r[3]} f2={r[4]}", flush=True)
+    m = meta.reshape(-1).tolist()
+    pos = 0
+    while pos + 5 <= len(m):
+        code, a, f0, f1, f2 = m[pos:pos + 5]
+        pos += 5
+        if code in _BULK_CODES:
+            nk = a
+            keys = m[pos:pos + nk]
+            pos += _BULK_CODES[code] * nk
+            if _TRACE_KEY == -2 or _TRACE_KEY in keys:
+                print(f"[trace r{rank} ch{ch} {direction} peer{peer}] BULK code={code} "
+                      f"nk={nk} f0={f0} f1={f1} f2={f2} keys={keys[:8]}...", flush=True)
+        elif _TRACE_KEY == -2 or a == _TRACE_KEY:
+            print(f"[trace r{rank} ch{ch} {direction} peer{peer}] code={code} key={a} "
+                  f"f0={f0} f1={f1} f2={f2}", flush=True)
 
 
 class ActionTimer:
@@ -81,6 +94,7 @@ class SyncManager:
         self.min_period = 1.0 / max_per_sec if max_per_sec > 0 else 0.0
         self.time_intent_actions = time_intent_actions
         self.stop_requested = threading.Event()
+        self.kick_event = threading.Event()
         self.threads = []
         self.timer = ActionTimer()
         if not time_intent_actions:
@@ -97,6 +111,14 @@ class SyncManager:
 
     def request_stop(self):
         self.stop_requested.set()
+        self.kick_event.set()
+
+    def kick(self):
+        """Wake the sync loops early: a worker enqueued remote ops and is
+        (or will be) waiting on the round. All ranks under symmetric load
+        kick at similar times, so the collective rounds speed up together;
+        an early kicker just reaches the size all-gather sooner."""
+        self.kick_event.set()
 
     def join(self):
         for t in self.threads:
@@ -128,23 +150,27 @@ class SyncManager:
                 self.server.set_intent_ahead(self.timer.update(self.server.worker_clocks()))
 
             stop = self.stop_requested.is_set()
-            all_stopped = self._round(ch, group, world, rank, dev, stop)
+            all_stopped, any_work = self._round(ch, group, world, rank, dev, stop)
             self.server.sync_finish(ch)
             n_rounds += 1
             if all_stopped:
                 return
-            if self.min_period > 0:
-                dt = time.monotonic() - t0
-                if dt < self.min_period:
-                    time.sleep(self.min_period - dt)
+            # pacing: kick-able sleep; when NO rank had work this round,
+            # back off harder (idle rounds are pure overhead)
+            period = self.min_period if any_work else max(self.min_period, 0.005)
+            dt = time.monotonic() - t0
+            if dt < period:
+                self.kick_event.wait(timeout=period - dt)
+                self.kick_event.clear()
 
-    def _round(self, ch, group, world, rank, dev, stop_flag) -> bool:
+    def _round(self, ch, group, world, rank, dev, stop_flag):
         out_a = self.server.sync_collect(ch)
-        all_stopped = self._exchange(ch, group, world, rank, dev, out_a,
-                                     self.server.sync_process, stop_flag)
+        all_stopped, work_a = self._exchange(ch, group, world, rank, dev, out_a,
+                                             self.server.sync_process, stop_flag)
         out_b = self.server.sync_respond(ch)
-        self._exchange(ch, group, world, rank, dev, out_b, self.server.sync_apply, stop_flag)
-        return all_stopped
+        _, work_b = self._exchange(ch, group, world, rank, dev, out_b,
+                                   self.server.sync_apply, stop_flag)
+        return all_stopped, (work_a or work_b)
 
     def _exchange(self, ch, group, world, rank, dev, outgoing, handler, stop_flag) -> bool:
         # size matrix: row = this rank's (n_meta_i64, n_payload_f32) per dest
@@ -166,6 +192,7 @@ class SyncManager:
         dist.all_gather(gathered, sizes_d, group=group)
         gathered = [g.cpu() for g in gathered]
         all_stopped = all(int(g[0, 2]) == 1 for g in gathered)
+        any_work = any(int(g[:, :2].sum()) > 0 for g in gathered)
 
         # post sends/recvs (meta then payload per peer; order pairs them)
         p2p = []
@@ -199,8 +226,8 @@ class SyncManager:
         store_dev = self.rt.device
         for peer in sorted(recv_bufs):
             rm, rp = recv_bufs[peer]
-            meta = rm.cpu().reshape(-1, 5)
+            meta = rm.cpu()
             _trace(rank, ch, "in ", peer, meta)
             payload = rp if rp.device == store_dev else rp.to(store_dev)
             handler(ch, peer, meta, payload)
-        return all_stopped
+        return all_stopped, any_work
