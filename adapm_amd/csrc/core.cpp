@@ -39,6 +39,8 @@
 #include <cstring>
 #include <deque>
 #include <map>
+#include <queue>
+#include <tuple>
 #include <memory>
 #include <mutex>
 #include <thread>
@@ -213,6 +215,10 @@ struct ChannelState {
   std::deque<IntentReq> intent_queue;                                   // worker -> sync
   std::vector<IntentReq> future_intents;                                // not yet due
   std::unordered_map<Key, std::vector<std::pair<int, Clock>>> intents;  // active local intents
+  // expiry min-heap (end, wid, key): avoids sweeping the whole intents
+  // map every round — only due entries are popped
+  std::priority_queue<std::tuple<Clock, int, Key>, std::vector<std::tuple<Clock, int, Key>>,
+                      std::greater<>> intent_expiry;
   std::unordered_set<Key> replicas;                                     // local replica/stub keys
   std::unordered_map<Key, uint64_t> holders;    // owner side: ranks holding replicas
   std::deque<OutRec> out_queue;                 // pending remote ops + forwards
@@ -870,55 +876,59 @@ class Server {
         C.future_intents.push_back(std::move(req));
         continue;
       }
+      std::vector<Key> stubs;
       for (Key k : req.keys) {
-        bool need_stub = false;
-        int64_t v_off = -1, s_off = -1;
-        int32_t l = len_of(k);
-        {
+        uint8_t f0 = flags_[k].load(std::memory_order_acquire);
+        if (!(f0 & F_PRESENT)) {
+          int32_t l = len_of(k);
           std::lock_guard<std::mutex> lk(stripe(k));
           uint8_t f = flags_[k];
           if (!(f & F_PRESENT)) {
             // replica stub: zeroed val+sync, absorbs pushes until first
             // refresh (reference handle registerNewIntentsForKeyUnsafe)
             layout_identity_.store(false, std::memory_order_release);
-            v_off = slab_.alloc(l);
-            s_off = slab_.alloc(l);
+            int64_t v_off = slab_.alloc(l);
+            int64_t s_off = slab_.alloc(l);
             loc_[k] = v_off;
             sync_loc_[k] = s_off;
             flags_[k] = F_PRESENT | F_STUB;
             version_[k] = 0;
-            need_stub = true;
+            stubs.push_back(k);
+            zero_batch.add(0, v_off, l);   // freelist reuse leaves stale data
+            zero_batch.add(0, s_off, l);
+            trace_event(k, "REPLICA_SETUP");
           }
         }
-        {
-          std::lock_guard<std::mutex> g(C.mu);
+      }
+      {
+        std::lock_guard<std::mutex> g(C.mu);
+        for (Key k : req.keys) {
           C.intents[k].push_back({req.wid, req.end});
-          if (need_stub) C.replicas.insert(k);
+          C.intent_expiry.push({req.end, req.wid, k});
         }
-        trace_event(k, need_stub ? "REPLICA_SETUP" : "INTENT_START");
-        if (need_stub) {  // freelist reuse leaves stale data: zero it
-          zero_batch.add(0, v_off, l);
-          zero_batch.add(0, s_off, l);
-        }
+        for (Key k : stubs) C.replicas.insert(k);
       }
     }
     run_zero(zero_batch);
 
-    // 2. expire intents; snapshot replica set
+    // 2. expire due intents (heap pop — no full-map sweep) and snapshot
+    // the replica set
     std::vector<Key> replica_snapshot;
     {
       std::lock_guard<std::mutex> g(C.mu);
-      for (auto it = C.intents.begin(); it != C.intents.end();) {
+      while (!C.intent_expiry.empty()) {
+        auto [end, wid, k] = C.intent_expiry.top();
+        if (end > clocks_[wid].load()) break;
+        C.intent_expiry.pop();
+        auto it = C.intents.find(k);
+        if (it == C.intents.end()) continue;
         auto& vec = it->second;
         vec.erase(std::remove_if(vec.begin(), vec.end(),
                                  [&](const std::pair<int, Clock>& p) {
-                                   return p.second <= clocks_[p.first].load();
+                                   return p.first == wid && p.second == end;
                                  }),
                   vec.end());
-        if (vec.empty())
-          it = C.intents.erase(it);
-        else
-          ++it;
+        if (vec.empty()) C.intents.erase(it);
       }
       replica_snapshot.assign(C.replicas.begin(), C.replicas.end());
     }
