@@ -170,6 +170,57 @@ class ComplEx:
             self.worker.wait(t)
         self._pending.clear()
 
+    # ---------------------------------------------- prefetch pipeline
+
+    def prefetch(self, triples: np.ndarray):
+        """Start the pulls for a future batch (bounded async, like the
+        reference's max_concurrent_loops pipelining): sample negatives,
+        allocate buffers, issue one async pull. Returns a handle for
+        train_prefetched."""
+        cfg = self.cfg
+        w = self.worker
+        B = len(triples)
+        s_keys, r_keys, o_keys = self.keys_of(triples)
+        if self.server.sampling is not None:
+            sid = w.prepare_sample(B * cfg.neg_samples, w.current_clock(),
+                                   w.current_clock() + 2)
+            neg_keys = self.server.sampling.pull(w, sid, B * cfg.neg_samples)
+            w.finish_sample(sid)
+        else:
+            neg_keys = self.rng.integers(0, cfg.num_entities, size=B * cfg.neg_samples,
+                                         dtype=np.int64)
+        all_keys = np.concatenate([s_keys, r_keys, o_keys, neg_keys])
+        all_v = torch.empty(len(all_keys) * cfg.row, dtype=torch.float32, device=self.dev)
+        ts = w.pull(all_keys, all_v, async_=True)
+        return (B, all_keys, all_v, ts)
+
+    def train_prefetched(self, handle, sync_loss: bool = False):
+        """Finish a prefetched batch: wait for the pull, run the fused
+        kernel, push the deltas."""
+        cfg = self.cfg
+        w = self.worker
+        B, all_keys, all_v, ts = handle
+        w.wait(ts)
+        row = cfg.row
+        NN = B * cfg.neg_samples
+        s_v = all_v[: B * row].view(B, row)
+        r_v = all_v[B * row: 2 * B * row].view(B, row)
+        o_v = all_v[2 * B * row: 3 * B * row].view(B, row)
+        n_v = all_v[3 * B * row:].view(NN, row)
+        all_d = torch.empty_like(all_v)
+        loss = torch.empty(B, dtype=torch.float32, device=self.dev)
+        _C.kge_complex_step(s_v, r_v, o_v, n_v, all_d[: B * row].view(B, row),
+                            all_d[B * row: 2 * B * row].view(B, row),
+                            all_d[2 * B * row: 3 * B * row].view(B, row),
+                            all_d[3 * B * row:].view(NN, row), loss,
+                            cfg.neg_samples, cfg.dim, cfg.lr, cfg.eps)
+        pt = w.push(all_keys, all_d, async_=True)
+        if pt != -1:
+            self._pending.append(pt)
+        while len(self._pending) > 64:
+            w.wait(self._pending.pop(0))
+        return float(loss.mean().item()) if sync_loss else loss
+
     # ------------------------------------------------------------ eval
 
     @torch.no_grad()
