@@ -357,10 +357,10 @@ class Server {
     d.len_t = torch::from_blob((void*)hb.len.data(), {(int64_t)hb.size()}, i32).clone();
     if (aux) d.aux_t = torch::from_blob((void*)aux->data(), {(int64_t)aux->size()}, i64).clone();
     if (dev_.is_cuda()) {
-      d.src_t = d.src_t.to(dev_, /*non_blocking=*/true);
-      d.dst_t = d.dst_t.to(dev_, true);
-      d.len_t = d.len_t.to(dev_, true);
-      if (aux) d.aux_t = d.aux_t.to(dev_, true);
+      d.src_t = d.src_t.pin_memory().to(dev_, /*non_blocking=*/true);
+      d.dst_t = d.dst_t.pin_memory().to(dev_, true);
+      d.len_t = d.len_t.pin_memory().to(dev_, true);
+      if (aux) d.aux_t = d.aux_t.pin_memory().to(dev_, true);
     }
     d.b.src_off = d.src_t.data_ptr<int64_t>();
     d.b.dst_off = d.dst_t.data_ptr<int64_t>();
@@ -438,7 +438,9 @@ class Server {
   }
   void run_gather_keys(const torch::Tensor& keys_cpu, torch::Tensor out) {
     if (dev_.is_cuda()) {
-      auto kd = keys_cpu.to(dev_, /*non_blocking=*/true);
+      // pinned staging: a pageable H2D copy is synchronous with the whole
+      // stream queue — with a deep prefetch pipeline that stalls the host
+      auto kd = keys_cpu.pin_memory().to(dev_, /*non_blocking=*/true);
       ops_gather_keys_gpu(slab_.bases(), key_batch(kd), out.data_ptr<float>(), current_stream(dev_));
     } else {
       std::lock_guard<std::mutex> g(cpu_val_mu_);
@@ -447,7 +449,7 @@ class Server {
   }
   void run_scatter_keys(const torch::Tensor& keys_cpu, const torch::Tensor& in, bool set) {
     if (dev_.is_cuda()) {
-      auto kd = keys_cpu.to(dev_, /*non_blocking=*/true);
+      auto kd = keys_cpu.pin_memory().to(dev_, /*non_blocking=*/true);
       ops_scatter_keys_gpu(slab_.bases(), key_batch(kd), in.data_ptr<float>(), set,
                            current_stream(dev_));
     } else {
@@ -1752,6 +1754,24 @@ class Server {
     }
   }
 
+  // Failure path (reference heartbeat/dead-node handling is skeletal,
+  // van.cc:515-527: detection + surface to the app). When the transport
+  // fails (peer died), the Python sync loop calls this: every blocked
+  // Wait() returns, and later ops raise instead of hanging.
+  void fail(std::string reason) {
+    {
+      std::lock_guard<std::mutex> g(tickets_mu_);
+      failed_reason_ = std::move(reason);
+      tickets_.clear();
+    }
+    tickets_cv_.notify_all();
+    rounds_cv_.notify_all();
+  }
+  std::string failed_reason() {
+    std::lock_guard<std::mutex> g(tickets_mu_);
+    return failed_reason_;
+  }
+
   void sync_finish(int ch) {
     {
       std::lock_guard<std::mutex> g(rounds_mu_);
@@ -1947,6 +1967,7 @@ class Server {
   std::mutex tickets_mu_;
   std::condition_variable tickets_cv_;
   std::unordered_map<int64_t, std::unique_ptr<Ticket>> tickets_;
+  std::string failed_reason_;
   std::mutex rounds_mu_;
   std::condition_variable rounds_cv_;
 
@@ -2074,6 +2095,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("sync_respond", &Server::sync_respond, py::call_guard<py::gil_scoped_release>())
       .def("sync_apply", &Server::sync_apply, py::call_guard<py::gil_scoped_release>())
       .def("sync_finish", &Server::sync_finish, py::call_guard<py::gil_scoped_release>())
+      .def("fail", &Server::fail, py::call_guard<py::gil_scoped_release>())
+      .def("failed_reason", &Server::failed_reason)
       .def("scan_local", &Server::scan_local, py::call_guard<py::gil_scoped_release>())
       .def("get_len", &Server::get_len)
       .def("num_keys", &Server::num_keys)

@@ -95,6 +95,7 @@ class SyncManager:
         self.time_intent_actions = time_intent_actions
         self.stop_requested = threading.Event()
         self.kick_event = threading.Event()
+        self.failed = False
         self.threads = []
         self.timer = ActionTimer()
         if not time_intent_actions:
@@ -128,6 +129,17 @@ class SyncManager:
     # ---------------------------------------------------------------- loop
 
     def _loop(self, ch: int):
+        try:
+            self._loop_inner(ch)
+        except Exception as e:  # transport failure: a peer likely died
+            import sys
+
+            print(f"[adapm] rank {self.rt.rank}: sync channel {ch} failed: "
+                  f"{type(e).__name__}: {e}", file=sys.stderr, flush=True)
+            self.failed = True
+            self.server.fail(f"sync channel {ch}: {e}")
+
+    def _loop_inner(self, ch: int):
         rt = self.rt
         group = rt.channel_groups[ch]
         world, rank = rt.world, rt.rank
