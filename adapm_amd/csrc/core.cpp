@@ -357,10 +357,10 @@ class Server {
     d.len_t = torch::from_blob((void*)hb.len.data(), {(int64_t)hb.size()}, i32).clone();
     if (aux) d.aux_t = torch::from_blob((void*)aux->data(), {(int64_t)aux->size()}, i64).clone();
     if (dev_.is_cuda()) {
-      d.src_t = d.src_t.pin_memory().to(dev_, /*non_blocking=*/true);
-      d.dst_t = d.dst_t.pin_memory().to(dev_, true);
-      d.len_t = d.len_t.pin_memory().to(dev_, true);
-      if (aux) d.aux_t = d.aux_t.pin_memory().to(dev_, true);
+      d.src_t = d.src_t.to(dev_, /*non_blocking=*/true);
+      d.dst_t = d.dst_t.to(dev_, true);
+      d.len_t = d.len_t.to(dev_, true);
+      if (aux) d.aux_t = d.aux_t.to(dev_, true);
     }
     d.b.src_off = d.src_t.data_ptr<int64_t>();
     d.b.dst_off = d.dst_t.data_ptr<int64_t>();
@@ -438,9 +438,7 @@ class Server {
   }
   void run_gather_keys(const torch::Tensor& keys_cpu, torch::Tensor out) {
     if (dev_.is_cuda()) {
-      // pinned staging: a pageable H2D copy is synchronous with the whole
-      // stream queue — with a deep prefetch pipeline that stalls the host
-      auto kd = keys_cpu.pin_memory().to(dev_, /*non_blocking=*/true);
+      auto kd = keys_cpu.to(dev_, /*non_blocking=*/true);
       ops_gather_keys_gpu(slab_.bases(), key_batch(kd), out.data_ptr<float>(), current_stream(dev_));
     } else {
       std::lock_guard<std::mutex> g(cpu_val_mu_);
@@ -449,7 +447,7 @@ class Server {
   }
   void run_scatter_keys(const torch::Tensor& keys_cpu, const torch::Tensor& in, bool set) {
     if (dev_.is_cuda()) {
-      auto kd = keys_cpu.pin_memory().to(dev_, /*non_blocking=*/true);
+      auto kd = keys_cpu.to(dev_, /*non_blocking=*/true);
       ops_scatter_keys_gpu(slab_.bases(), key_batch(kd), in.data_ptr<float>(), set,
                            current_stream(dev_));
     } else {

@@ -39,8 +39,8 @@ def main():
     ap.add_argument("--lookahead", type=int, default=4)
     ap.add_argument("--no-intent", action="store_true",
                     help="disable intent signaling (pure remote-op mode)")
-    ap.add_argument("--pipeline", type=int, default=8,
-                    help="prefetch depth (bounded async, reference max_concurrent_loops)")
+    ap.add_argument("--pipeline", type=int, default=0,
+                    help="prefetch depth (bounded async, reference max_concurrent_loops); 0 = blocking per step")
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--channels", type=int, default=2)
     ap.add_argument("--capacity-factor", type=float, default=3.0)
@@ -84,7 +84,7 @@ def main():
     total_steps = args.warmup + args.steps
     batches = [make_batch(i) for i in range(total_steps + args.lookahead)]
 
-    # prefetch pipeline of depth args.pipeline (bounded async)
+    # optional prefetch pipeline (bounded async)
     from collections import deque
 
     handles = deque()
@@ -96,15 +96,25 @@ def main():
                                 worker.current_clock() + args.lookahead + args.pipeline + 2)
         handles.append(model.prefetch(batches[i]))
 
-    def run_step(i, sync_loss=False):
-        if i + args.pipeline < total_steps + args.lookahead - args.lookahead:
-            issue(i + args.pipeline)
-        loss = model.train_prefetched(handles.popleft(), sync_loss=sync_loss)
-        worker.advance_clock()
-        return loss
+    if args.pipeline > 0:
+        def run_step(i, sync_loss=False):
+            if i + args.pipeline < total_steps:
+                issue(i + args.pipeline)
+            loss = model.train_prefetched(handles.popleft(), sync_loss=sync_loss)
+            worker.advance_clock()
+            return loss
 
-    for i in range(min(args.pipeline, total_steps)):
-        issue(i)
+        for i in range(min(args.pipeline, total_steps)):
+            issue(i)
+    else:
+        def run_step(i, sync_loss=False):
+            if not args.no_intent:
+                model.signal_intent(batches[i + args.lookahead],
+                                    worker.current_clock() + args.lookahead,
+                                    worker.current_clock() + args.lookahead + 2)
+            loss = model.train_batch(batches[i], sync_loss=sync_loss)
+            worker.advance_clock()
+            return loss
 
     prof = None
     if os.environ.get("ADAPM_PROFILE") == "1":
