@@ -96,6 +96,9 @@ class SyncManager:
         self.stop_requested = threading.Event()
         self.kick_event = threading.Event()
         self.failed = False
+        from collections import defaultdict
+
+        self.phase_totals = defaultdict(float)
         self.threads = []
         self.timer = ActionTimer()
         if not time_intent_actions:
@@ -125,6 +128,15 @@ class SyncManager:
         for t in self.threads:
             t.join()
         self.threads = []
+        import os
+        import sys
+
+        if os.environ.get("ADAPM_VERBOSE", "0") != "0" and self.phase_totals.get("rounds"):
+            pt = dict(self.phase_totals)
+            n = pt.pop("rounds")
+            print(f"[adapm sync r{self.rt.rank}] {int(n)} rounds; per-round ms: " +
+                  ", ".join(f"{k}={1000*v/n:.2f}" for k, v in sorted(pt.items())),
+                  file=sys.stderr, flush=True)
 
     # ---------------------------------------------------------------- loop
 
@@ -176,12 +188,23 @@ class SyncManager:
                 self.kick_event.clear()
 
     def _round(self, ch, group, world, rank, dev, stop_flag):
+        t0 = time.perf_counter()
         out_a = self.server.sync_collect(ch)
+        t1 = time.perf_counter()
         all_stopped, work_a = self._exchange(ch, group, world, rank, dev, out_a,
                                              self.server.sync_process, stop_flag)
+        t2 = time.perf_counter()
         out_b = self.server.sync_respond(ch)
+        t3 = time.perf_counter()
         _, work_b = self._exchange(ch, group, world, rank, dev, out_b,
                                    self.server.sync_apply, stop_flag)
+        t4 = time.perf_counter()
+        pt = self.phase_totals
+        pt["collect"] += t1 - t0
+        pt["exchange_a"] += t2 - t1
+        pt["respond"] += t3 - t2
+        pt["exchange_b"] += t4 - t3
+        pt["rounds"] += 1
         return all_stopped, (work_a or work_b)
 
     def _exchange(self, ch, group, world, rank, dev, outgoing, handler, stop_flag) -> bool:
