@@ -1,0 +1,98 @@
+#!/usr/bin/env python3
+"""CTR wide-and-deep benchmark (BASELINE config 5: 100M x 64 embedding
+table; with --device-cap-gb the table exceeds the HBM budget and spills to
+pinned host memory). Metric = examples/s."""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--features", type=int, default=100_000_000)
+    ap.add_argument("--dim", type=int, default=64)
+    ap.add_argument("--fields", type=int, default=16)
+    ap.add_argument("--batch", type=int, default=16384)
+    ap.add_argument("--device", type=str, default=None)
+    ap.add_argument("--device-cap-gb", type=float, default=0.0,
+                    help=">0: cap HBM arena; the rest spills to pinned host")
+    ap.add_argument("--host-spill-gb", type=float, default=0.0)
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+
+    import adapm_amd
+    from adapm_amd.models.ctr import CTRConfig, WideAndDeep, make_synthetic_ctr
+
+    cfg = CTRConfig(num_features=args.features, dim=args.dim, fields=args.fields,
+                    batch_size=args.batch)
+    adapm_amd.setup(num_keys=cfg.num_features, num_threads=1, device=args.device,
+                    capacity_factor=1.3, max_sync_per_sec=2000.0,
+                    device_cap_gb=args.device_cap_gb, host_spill_gb=args.host_spill_gb)
+    server = adapm_amd.Server(cfg.row)
+    worker = adapm_amd.Worker(0, server)
+    model = WideAndDeep(cfg, server, worker)
+    t_init0 = time.time()
+    model.init_embeddings()
+    init_s = time.time() - t_init0
+
+    rng = np.random.default_rng(4000 + rank)
+    total = args.warmup + args.steps
+    batches = [make_synthetic_ctr(args.batch, args.features, args.fields, seed=1000 * rank + i)
+               for i in range(total)]
+
+    is_cuda = server.rt.device.type == "cuda"
+    for i in range(args.warmup):
+        model.train_batch(*batches[i], sync_loss=False)
+    model.drain()
+    worker.barrier()
+    if is_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.warmup, total):
+        if i + 1 < total:
+            model.signal_intent(batches[i + 1][0], worker.current_clock() + 1,
+                                worker.current_clock() + 3)
+        model.train_batch(*batches[i], sync_loss=False)
+        worker.advance_clock()
+    model.drain()
+    if is_cuda:
+        torch.cuda.synchronize()
+    el = time.perf_counter() - t0
+    worker.barrier()
+    if world > 1:
+        el = worker.allreduce(el, op="max")
+
+    st = server.stats()
+    if rank == 0:
+        print(json.dumps({
+            "metric": "ctr_examples_per_s", "value": args.batch * args.steps * world / el,
+            "unit": "examples/s", "n_gpus": world, "steps": args.steps,
+            "warmup": args.warmup, "ms_per_step": 1000 * el / args.steps,
+            "higher_is_better": True, "scaling": "weak", "vs_baseline": None,
+            "dtype": "fp32", "data": "synthetic",
+            "config": {"model": "ctr_wide_and_deep_100Mx64", "features": args.features,
+                       "dim": args.dim, "fields": args.fields,
+                       "global_batch": args.batch * world, "init_s": init_s,
+                       "device_cap_gb": args.device_cap_gb,
+                       "host_spill_in_use_gb": st["host_spill_in_use"] * 4 / (1 << 30),
+                       "pull_push_ops_per_s": 2 * args.batch * args.fields * args.steps * world / el,
+                       "parallelism": f"ps-async-dp{world}"},
+        }), flush=True)
+    worker.finalize()
+    server.shutdown()
+
+
+if __name__ == "__main__":
+    main()
