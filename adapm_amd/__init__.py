@@ -133,7 +133,7 @@ class Server:
     def enable_sampling_support(self, scheme: str, with_replacement: bool, distribution: str,
                                 min: int, max: int, counts=None, power: float = 0.75):
         d = make_distribution(distribution, min, max, seed=self.rt.rank, counts=counts,
-                              power=power)
+                              power=power, device=self.rt.device)
         self.sampling = SamplingManager(self._s, scheme, with_replacement, d, min, max)
 
     def barrier(self):

@@ -2057,12 +2057,28 @@ void mf_update_step(torch::Tensor w, torch::Tensor h, torch::Tensor x, torch::Te
   }
 }
 
+// alias-table draw: prob/alias on the op device; returns int64 keys there
+torch::Tensor alias_draw(torch::Tensor prob, torch::Tensor alias, int64_t seed, int64_t N) {
+  TORCH_CHECK(prob.is_contiguous() && prob.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(alias.is_contiguous() && alias.scalar_type() == torch::kInt32);
+  auto out = torch::empty({N}, torch::TensorOptions().dtype(torch::kInt64).device(prob.device()));
+  if (prob.is_cuda()) {
+    alias_draw_gpu(prob.data_ptr<float>(), alias.data_ptr<int32_t>(), prob.numel(),
+                   (uint64_t)seed, N, out.data_ptr<int64_t>(), current_stream(prob.device()));
+  } else {
+    alias_draw_cpu(prob.data_ptr<float>(), alias.data_ptr<int32_t>(), prob.numel(),
+                   (uint64_t)seed, N, out.data_ptr<int64_t>());
+  }
+  return out;
+}
+
 }  // namespace adapm
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   using namespace adapm;
   m.def("hip_available", &hip_available);
   m.def("kge_complex_step", &kge_complex_step, py::call_guard<py::gil_scoped_release>());
+  m.def("alias_draw", &alias_draw, py::call_guard<py::gil_scoped_release>());
   m.def("kge_complex_score", &kge_complex_score, py::call_guard<py::gil_scoped_release>());
   m.def("w2v_sgns_step", &w2v_sgns_step, py::call_guard<py::gil_scoped_release>());
   m.def("mf_update_step", &mf_update_step, py::call_guard<py::gil_scoped_release>());

@@ -54,4 +54,13 @@ void mf_update_step_gpu(const float* w, const float* h, const float* x, float* d
 void mf_update_step_cpu(const float* w, const float* h, const float* x, float* dw, float* dh,
                         float* loss, int B, int R, float lr, float lambda, float eps);
 
+// Alias-table sampling draw (negative sampling; reference unigram table,
+// word2vec.cc:125-146 — rebuilt as an O(1)-per-draw alias table):
+//  prob[n] f32, alias[n] i32 built host-side; out[N] int64 drawn keys.
+//  Counter-based RNG (PCG hash of (seed, index)): reproducible, stateless.
+void alias_draw_gpu(const float* prob, const int32_t* alias, int64_t n, uint64_t seed,
+                    int64_t N, int64_t* out, void* stream);
+void alias_draw_cpu(const float* prob, const int32_t* alias, int64_t n, uint64_t seed,
+                    int64_t N, int64_t* out);
+
 }  // namespace adapm

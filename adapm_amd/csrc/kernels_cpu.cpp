@@ -139,4 +139,21 @@ void mf_update_step_cpu(const float* w, const float* h, const float* x, float* d
   }
 }
 
+static inline uint64_t pcg_hash64_c(uint64_t x) {
+  x ^= x >> 33; x *= 0xff51afd7ed558ccdULL;
+  x ^= x >> 33; x *= 0xc4ceb9fe1a85ec53ULL;
+  x ^= x >> 33;
+  return x;
+}
+
+void alias_draw_cpu(const float* prob, const int32_t* alias, int64_t n, uint64_t seed,
+                    int64_t N, int64_t* out) {
+  for (int64_t i = 0; i < N; ++i) {
+    uint64_t h = pcg_hash64_c(seed ^ (uint64_t)i * 0x9e3779b97f4a7c15ULL);
+    int64_t slot = (int64_t)(h % (uint64_t)n);
+    float u = (float)((h >> 40) & 0xffffff) * (1.0f / 16777216.0f);
+    out[i] = (u < prob[slot]) ? slot : (int64_t)alias[slot];
+  }
+}
+
 }  // namespace adapm

@@ -5,6 +5,7 @@
 // goal is coalesced float4 row traffic and cheap block reductions
 // (wave __shfl_xor + one LDS hop across the 4 waves).
 #include <hip/hip_runtime.h>
+#include <algorithm>
 
 #include "kernels.h"
 
@@ -253,6 +254,27 @@ __global__ void k_mf_step(const float* __restrict__ w, const float* __restrict__
   }
 }
 
+// --------------------------------------------------------------- alias draw
+
+__device__ __host__ inline uint64_t pcg_hash64(uint64_t x) {
+  x ^= x >> 33; x *= 0xff51afd7ed558ccdULL;
+  x ^= x >> 33; x *= 0xc4ceb9fe1a85ec53ULL;
+  x ^= x >> 33;
+  return x;
+}
+
+__global__ void k_alias_draw(const float* __restrict__ prob, const int32_t* __restrict__ alias,
+                             int64_t n, uint64_t seed, int64_t N, int64_t* __restrict__ out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < N; i += stride) {
+    uint64_t h = pcg_hash64(seed ^ (uint64_t)i * 0x9e3779b97f4a7c15ULL);
+    int64_t slot = (int64_t)(h % (uint64_t)n);
+    float u = (float)((h >> 40) & 0xffffff) * (1.0f / 16777216.0f);
+    out[i] = (u < prob[slot]) ? slot : (int64_t)alias[slot];
+  }
+}
+
 // --------------------------------------------------------------- launchers
 
 static inline int grid_for(int64_t n) {
@@ -301,4 +323,14 @@ void mf_update_step_gpu(const float* w, const float* h, const float* x, float* d
                      dh, loss, B, R, lr, lambda, eps);
 }
 
+}  // namespace adapm
+
+namespace adapm {
+void alias_draw_gpu(const float* prob, const int32_t* alias, int64_t n, uint64_t seed,
+                    int64_t N, int64_t* out, void* stream) {
+  if (N == 0) return;
+  int blocks = (int)std::min<int64_t>((N + 255) / 256, 4096);
+  hipLaunchKernelGGL(k_alias_draw, dim3(blocks), dim3(256), 0, (hipStream_t)stream, prob, alias,
+                     n, seed, N, out);
+}
 }  // namespace adapm

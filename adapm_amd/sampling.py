@@ -45,20 +45,48 @@ class LogUniform(Distribution):
 
 
 class Unigram(Distribution):
-    """unigram^power table sampling (word2vec negative sampling,
-    reference apps/word2vec.cc:125-146). Alias-free: inverse-CDF via
-    searchsorted on the power-smoothed cumulative counts."""
+    """unigram^power sampling (word2vec negative sampling, reference
+    apps/word2vec.cc:125-146) via an O(1)-per-draw ALIAS TABLE. On GPU
+    stores the draw is a HIP kernel (_C.alias_draw) with the table
+    resident on-device; on CPU it is the native C++ loop."""
 
-    def __init__(self, counts: np.ndarray, key_of: Optional[np.ndarray], power: float, seed: int):
+    def __init__(self, counts: np.ndarray, key_of: Optional[np.ndarray], power: float, seed: int,
+                 device=None):
         p = counts.astype(np.float64) ** power
-        self.cdf = np.cumsum(p / p.sum())
+        p = p / p.sum()
+        self.prob, self.alias = self._build_alias(p)
         self.key_of = key_of  # optional map index->key
-        self.rng = np.random.default_rng(seed)
+        self.seed = np.random.default_rng(seed).integers(1, 2 ** 62)
+        self._calls = 0
+        self.device = device
+        if device is not None and str(device).startswith("cuda"):
+            self.prob = self.prob.to(device)
+            self.alias = self.alias.to(device)
+
+    @staticmethod
+    def _build_alias(p: np.ndarray):
+        import torch as _t
+
+        n = len(p)
+        prob = (p * n).astype(np.float64)
+        alias = np.zeros(n, dtype=np.int32)
+        small = list(np.where(prob < 1.0)[0][::-1])
+        large = list(np.where(prob >= 1.0)[0][::-1])
+        while small and large:
+            s, l = small.pop(), large.pop()
+            alias[s] = l
+            prob[l] = prob[l] - (1.0 - prob[s])
+            (small if prob[l] < 1.0 else large).append(l)
+        return (_t.from_numpy(prob.clip(0, 1).astype(np.float32)),
+                _t.from_numpy(alias))
 
     def draw(self, n):
-        u = self.rng.random(n)
-        idx = np.searchsorted(self.cdf, u, side="right").astype(np.int64)
-        return idx if self.key_of is None else self.key_of[idx]
+        from . import _C
+
+        self._calls += 1
+        out = _C.alias_draw(self.prob, self.alias, int(self.seed + self._calls), n)
+        keys = out.cpu().numpy() if out.is_cuda else out.numpy()
+        return keys if self.key_of is None else self.key_of[keys]
 
 
 @dataclass
@@ -177,7 +205,7 @@ class SamplingManager:
 
 
 def make_distribution(name: str, lo: int, hi: int, seed: int, counts=None,
-                      power: float = 0.75) -> Distribution:
+                      power: float = 0.75, device=None) -> Distribution:
     if name == "uniform":
         return Uniform(lo, hi, seed)
     if name == "log-uniform":
@@ -185,5 +213,5 @@ def make_distribution(name: str, lo: int, hi: int, seed: int, counts=None,
     if name == "unigram":
         if counts is None:
             raise ValueError("unigram distribution needs counts")
-        return Unigram(np.asarray(counts), None, power, seed)
+        return Unigram(np.asarray(counts), None, power, seed, device=device)
     raise ValueError(f"unknown sampling distribution '{name}'")

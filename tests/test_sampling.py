@@ -112,3 +112,15 @@ def test_unigram_distribution():
     ks = d.draw(20000)
     freq = np.bincount(ks, minlength=4) / 20000
     assert freq[0] > freq[1] > freq[2] * 2
+
+
+def test_alias_table_distribution():
+    """Alias-table draws match the target unigram^power distribution."""
+    from adapm_amd.sampling import Unigram
+
+    counts = np.array([100, 50, 10, 5, 1], dtype=np.float64)
+    d = Unigram(counts, None, 1.0, seed=9)
+    ks = d.draw(200000)
+    freq = np.bincount(ks, minlength=5) / 200000
+    target = counts / counts.sum()
+    assert np.allclose(freq, target, atol=0.01), (freq, target)
