@@ -177,3 +177,63 @@ def make_synthetic_sentences(n_sentences, vocab_size, mean_len=12, zipf_a=1.2, s
         words = rng.zipf(zipf_a, size=L)
         out.append(np.minimum(words - 1, vocab_size - 1).astype(np.int64))
     return out
+
+
+def main():
+    """CLI (rebuild of reference apps/word2vec.cc): SGNS on a synthetic
+    corpus with subsampling + unigram negatives; exports embeddings."""
+    import argparse
+    import time
+
+    import adapm_amd as _a
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--vocab", type=int, default=100_000)
+    ap.add_argument("--dim", type=int, default=100)
+    ap.add_argument("--window", type=int, default=5)
+    ap.add_argument("--negative", type=int, default=5)
+    ap.add_argument("--sentences", type=int, default=20_000)
+    ap.add_argument("--epochs", type=int, default=2)
+    ap.add_argument("--batch-pairs", type=int, default=16384)
+    ap.add_argument("--output", type=str, default="")
+    ap.add_argument("--device", type=str, default=None)
+    a = ap.parse_args()
+
+    cfg = W2VConfig(vocab_size=a.vocab, dim=a.dim, window=a.window, negative=a.negative,
+                    batch_pairs=a.batch_pairs)
+    _a.setup(num_keys=cfg.num_keys, num_threads=1, device=a.device)
+    server = _a.Server(cfg.row)
+    counts = (1.0 / np.arange(1, a.vocab + 1)) ** 0.75 * 1e9
+    server.enable_sampling_support("local", True, "unigram", 0, a.vocab, counts=counts)
+    worker = _a.Worker(0, server)
+    model = Word2Vec(cfg, server, worker)
+    model.set_vocab_counts(counts)
+    model.init_embeddings()
+    rank = server.my_rank()
+    world = server.rt.world
+    sents = make_synthetic_sentences(a.sentences // world, a.vocab, seed=rank)
+    for ep in range(a.epochs):
+        t0 = time.time()
+        ctr, ctx = model.pairs_from_sentences(sents)
+        losses = []
+        for i in range(0, len(ctr), a.batch_pairs):
+            model.signal_intent([np.unique(ctr[i:i + a.batch_pairs])],
+                                worker.current_clock() + 1, worker.current_clock() + 3)
+            losses.append(model.train_pairs(ctr[i:i + a.batch_pairs],
+                                            ctx[i:i + a.batch_pairs], sync_loss=True))
+            worker.advance_clock()
+        model.drain()
+        total = worker.allreduce(float(np.mean(losses)))
+        if rank == 0:
+            print(f"[w2v] epoch {ep}: loss {total / world:.4f} "
+                  f"({len(ctr)} pairs, {time.time()-t0:.1f}s)")
+    if a.output:
+        model.export_text(a.output, max_words=min(a.vocab, 10000))
+        if rank == 0:
+            print(f"[w2v] embeddings -> {a.output}")
+    worker.finalize()
+    server.shutdown()
+
+
+if __name__ == "__main__":
+    main()
