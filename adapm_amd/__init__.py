@@ -360,6 +360,24 @@ class Worker:
         self.wait_sync()
         self.barrier()
 
+    def staggered_push(self, keys, vals, chunk: int = 65536):
+        """Memory-bounded bulk push: chunked pushes, each waited before
+        the next chunk's values are touched (reference StaggeredPush,
+        coloc_kv_worker.h:556-580)."""
+        kt = _to_key_tensor(keys)
+        vt, _ = self._vals_tensor(vals)
+        flat = vt.reshape(vt.shape[0], -1) if vt.dim() > 1 else vt.reshape(kt.numel(), -1)
+        prev = -1
+        for i in range(0, kt.numel(), chunk):
+            ks = kt[i:i + chunk]
+            vs = flat[i:i + chunk]
+            if prev != -1:
+                self._s.wait(prev)
+            prev = self._s.push(self.wid, ks.contiguous(), vs.contiguous(), False)
+            if prev != -1:
+                self.server._sync.kick_event.set()
+        self._s.wait(prev)
+
     # ---- collectives (replaces reference utils.h ps_allreduce)
 
     def allreduce(self, value, op: str = "sum"):

@@ -111,3 +111,15 @@ def test_exact_sum_async_single_rank():
     w.pull(key, out)
     assert out[0, 0].item() == runs and out[0, 1].item() == 2 * runs
     s.shutdown()
+
+
+def test_staggered_push():
+    s = make_server(num_keys=1000, lens=4)
+    w = adapm_amd.Worker(0, s)
+    keys = torch.arange(1000)
+    vals = torch.ones(1000, 4)
+    w.staggered_push(keys, vals, chunk=128)
+    out = torch.zeros(1000, 4)
+    w.pull(keys, out)
+    assert torch.equal(out, vals)
+    s.shutdown()
