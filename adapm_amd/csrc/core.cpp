@@ -448,6 +448,16 @@ class Server {
   void run_scatter_keys(const torch::Tensor& keys_cpu, const torch::Tensor& in, bool set) {
     if (dev_.is_cuda()) {
       auto kd = keys_cpu.to(dev_, /*non_blocking=*/true);
+      if (!set) {
+        // sorted chunked scatter: duplicates pre-summed, no hot-row
+        // atomic contention (Zipf pushes)
+        auto [sorted, perm] = kd.sort();
+        ops_scatter_sorted_gpu(slab_.bases(), sorted.data_ptr<int64_t>(),
+                               perm.data_ptr<int64_t>(), (int)kd.numel(), uniform_len_,
+                               Slab::padded(uniform_len_), world_, rank_,
+                               in.data_ptr<float>(), current_stream(dev_));
+        return;
+      }
       ops_scatter_keys_gpu(slab_.bases(), key_batch(kd), in.data_ptr<float>(), set,
                            current_stream(dev_));
     } else {

@@ -74,6 +74,14 @@ void ops_gather_keys_cpu(const SlabBases& slab, const KeyBatch& b, float* out);
 void ops_scatter_keys_gpu(const SlabBases& slab, const KeyBatch& b, const float* in, bool set, void* stream);
 void ops_scatter_keys_cpu(const SlabBases& slab, const KeyBatch& b, const float* in, bool set);
 
+// Sorted scatter-add (identity layout): keys pre-sorted on device with the
+// permutation into `in` rows. Consecutive duplicate keys are pre-summed in
+// registers, so hot keys (Zipf pushes) cost one atomic per 16-row chunk
+// instead of one per row — removes atomic contention.
+void ops_scatter_sorted_gpu(const SlabBases& slab, const int64_t* sorted_keys,
+                            const int64_t* perm, int n, int32_t len, int32_t plen, int world,
+                            int rank, const float* in, void* stream);
+
 bool hip_available();
 
 }  // namespace adapm

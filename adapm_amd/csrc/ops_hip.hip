@@ -232,3 +232,53 @@ bool hip_available() {
 }
 
 }  // namespace adapm
+
+namespace adapm {
+
+#define SCHUNK 16  // sorted rows per workgroup
+
+__global__ void k_scatter_add_sorted(float* __restrict__ slab, const int64_t* __restrict__ skeys,
+                                     const int64_t* __restrict__ perm, int n, int32_t len,
+                                     int32_t plen, int world, int rank,
+                                     const float* __restrict__ in) {
+  __shared__ int64_t offs[SCHUNK];
+  __shared__ int32_t rows[SCHUNK];
+  for (int c0 = blockIdx.x * SCHUNK; c0 < n; c0 += gridDim.x * SCHUNK) {
+    int cn = min(SCHUNK, n - c0);
+    if (threadIdx.x < (unsigned)cn) {
+      int64_t k = skeys[c0 + threadIdx.x];
+      offs[threadIdx.x] =
+          ((int)(k % world) == rank) ? (k / world) * (int64_t)plen : (int64_t)-1;
+      rows[threadIdx.x] = (int32_t)perm[c0 + threadIdx.x];
+    }
+    __syncthreads();
+    for (int e = threadIdx.x; e < len; e += THREADS) {
+      float acc = 0.f;
+      int64_t cur = -1;
+      for (int j = 0; j < cn; ++j) {
+        int64_t o = offs[j];
+        if (o < 0) continue;
+        float v = in[(int64_t)rows[j] * len + e];
+        if (o != cur) {
+          if (cur >= 0) atomicAdd(&slab[cur + e], acc);
+          cur = o;
+          acc = 0.f;
+        }
+        acc += v;
+      }
+      if (cur >= 0) atomicAdd(&slab[cur + e], acc);
+    }
+    __syncthreads();
+  }
+}
+
+void ops_scatter_sorted_gpu(const SlabBases& slab, const int64_t* sorted_keys,
+                            const int64_t* perm, int n, int32_t len, int32_t plen, int world,
+                            int rank, const float* in, void* stream) {
+  if (n == 0) return;
+  int blocks = (int)std::min<int64_t>(((int64_t)n + SCHUNK - 1) / SCHUNK, 16384);
+  hipLaunchKernelGGL(k_scatter_add_sorted, dim3(blocks), dim3(THREADS), 0, (hipStream_t)stream,
+                     slab.dev, sorted_keys, perm, n, len, plen, world, rank, in);
+}
+
+}  // namespace adapm
