@@ -448,16 +448,10 @@ class Server {
   void run_scatter_keys(const torch::Tensor& keys_cpu, const torch::Tensor& in, bool set) {
     if (dev_.is_cuda()) {
       auto kd = keys_cpu.to(dev_, /*non_blocking=*/true);
-      if (!set) {
-        // sorted chunked scatter: duplicates pre-summed, no hot-row
-        // atomic contention (Zipf pushes)
-        auto [sorted, perm] = kd.sort();
-        ops_scatter_sorted_gpu(slab_.bases(), sorted.data_ptr<int64_t>(),
-                               perm.data_ptr<int64_t>(), (int)kd.numel(), uniform_len_,
-                               Slab::padded(uniform_len_), world_, rank_,
-                               in.data_ptr<float>(), current_stream(dev_));
-        return;
-      }
+      // NOTE: a sorted chunked scatter (ops_scatter_sorted_gpu) was
+      // measured here and LOST to plain HW atomics on every workload
+      // (the device sort + permuted reads cost more than hot-row atomic
+      // contention saves) — keep the straightforward atomic path.
       ops_scatter_keys_gpu(slab_.bases(), key_batch(kd), in.data_ptr<float>(), set,
                            current_stream(dev_));
     } else {
