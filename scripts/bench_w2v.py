@@ -44,9 +44,19 @@ def main():
     model.init_embeddings()
 
     rng = np.random.default_rng(2000 + rank)
-    zipf = lambda n: np.minimum(rng.zipf(1.3, size=n) - 1, args.vocab - 1).astype(np.int64)
+    # training pairs follow the SUBSAMPLED word distribution (reference
+    # word2vec 'sample' parameter drops frequent words before pairing —
+    # the same subsampling Word2Vec.set_vocab_counts implements), drawn
+    # via the alias table
+    from adapm_amd.sampling import Unigram
+
+    f = counts / counts.sum()
+    keep = np.minimum(1.0, np.sqrt(1e-3 / f) + 1e-3 / f)
+    pair_dist = Unigram(counts * keep, None, 1.0, seed=2000 + rank,
+                        device=server.rt.device)
     total = args.warmup + args.steps
-    batches = [(zipf(args.pairs), zipf(args.pairs)) for _ in range(total)]
+    batches = [(pair_dist.draw(args.pairs), pair_dist.draw(args.pairs))
+               for _ in range(total)]
 
     is_cuda = server.rt.device.type == "cuda"
     for i in range(args.warmup):
