@@ -185,6 +185,41 @@ class MF:
         return self.worker.allreduce(se) / self.worker.allreduce(float(B))
 
 
+def _chunks(n, c):
+    for i in range(0, n, c):
+        yield i, min(i + c, n)
+
+
+def save_factors(model: "MF", path: str, chunk: int = 65536):
+    """rank-0 full pull of row+column factors (incl. AdaGrad state) ->
+    .npz (reference matrix_factorization.cc:256-272 factor export;
+    loading them back is the resume path, :238-241)."""
+    model.drain()
+    model.worker.wait_sync()
+    model.worker.barrier()
+    if model.rank_id == 0:
+        cfg = model.cfg
+        out = np.empty((cfg.num_keys, cfg.row), dtype=np.float32)
+        for a, b in _chunks(cfg.num_keys, chunk):
+            buf = np.zeros((b - a, cfg.row), dtype=np.float32)
+            model.worker.pull(np.arange(a, b, dtype=np.int64), buf)
+            out[a:b] = buf
+        np.savez(path, w=out[:cfg.num_rows], h=out[cfg.num_rows:],
+                 rank=cfg.rank, rows=cfg.num_rows, cols=cfg.num_cols)
+    model.worker.barrier()
+
+
+def load_factors(model: "MF", path: str, chunk: int = 65536):
+    if model.rank_id == 0:
+        data = np.load(path)
+        vals = np.concatenate([data["w"], data["h"]])
+        for a, b in _chunks(len(vals), chunk):
+            model.worker.set(np.arange(a, b, dtype=np.int64),
+                             np.ascontiguousarray(vals[a:b]))
+    model.worker.wait_sync()
+    model.worker.barrier()
+
+
 def make_synthetic_ratings(n, num_rows, num_cols, rank_true=8, seed=0):
     """Low-rank synthetic ratings so MF can actually fit them."""
     rng = np.random.default_rng(seed)

@@ -177,3 +177,27 @@ def test_ctr_cpu():
 @pytest.mark.gpu
 def test_ctr_gpu():
     _ctr("cuda:0")
+
+
+def test_mf_checkpoint_roundtrip(tmp_path):
+    adapm = _fresh()
+    from adapm_amd.models.mf import (MF, MFConfig, load_factors,
+                                     make_synthetic_ratings, save_factors)
+
+    cfg = MFConfig(num_rows=100, num_cols=50, rank=8)
+    adapm.setup(num_keys=cfg.num_keys, num_threads=1, device="cpu")
+    server = adapm.Server(cfg.row)
+    worker = adapm.Worker(0, server)
+    model = MF(cfg, server, worker)
+    model.init_factors()
+    path = str(tmp_path / "mf.npz")
+    save_factors(model, path)
+    before = np.zeros((5, cfg.row), dtype=np.float32)
+    worker.pull(np.arange(5, dtype=np.int64), before)
+    worker.set(np.arange(5, dtype=np.int64), np.zeros((5, cfg.row), dtype=np.float32))
+    load_factors(model, path)
+    after = np.zeros((5, cfg.row), dtype=np.float32)
+    worker.pull(np.arange(5, dtype=np.int64), after)
+    assert np.allclose(before, after)
+    worker.finalize()
+    server.shutdown()
