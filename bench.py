@@ -39,7 +39,7 @@ def main():
     ap.add_argument("--lookahead", type=int, default=4)
     ap.add_argument("--no-intent", action="store_true",
                     help="disable intent signaling (pure remote-op mode)")
-    ap.add_argument("--pipeline", type=int, default=8,
+    ap.add_argument("--pipeline", type=int, default=0,
                     help="prefetch depth (bounded async, reference max_concurrent_loops); 0 = blocking per step")
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--channels", type=int, default=2)
