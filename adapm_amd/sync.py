@@ -259,10 +259,12 @@ class SyncManager:
                 r.wait()
         # handle incoming in fixed rank order for determinism
         store_dev = self.rt.device
+        th0 = time.perf_counter()
         for peer in sorted(recv_bufs):
             rm, rp = recv_bufs[peer]
             meta = rm.cpu()
             _trace(rank, ch, "in ", peer, meta)
             payload = rp if rp.device == store_dev else rp.to(store_dev)
             handler(ch, peer, meta, payload)
+        self.phase_totals["handlers"] += time.perf_counter() - th0
         return all_stopped, any_work
