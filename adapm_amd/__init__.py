@@ -222,7 +222,6 @@ class Worker:
         self.server = server
         self.wid = customer_id
         self._s = server._s
-        self._np_staging = {}
 
     # ---- data ops (torch / numpy overloads, like bindings.cc:160-290)
 

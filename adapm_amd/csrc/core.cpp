@@ -2018,6 +2018,23 @@ void kge_complex_step(torch::Tensor s, torch::Tensor r, torch::Tensor o, torch::
   }
 }
 
+void rescal_step(torch::Tensor s, torch::Tensor r, torch::Tensor o, torch::Tensor neg,
+                 torch::Tensor ds, torch::Tensor drl, torch::Tensor do_, torch::Tensor dneg,
+                 torch::Tensor loss, int64_t N, int64_t D, double lr, double eps) {
+  for (auto* t : {&s, &r, &o, &neg, &ds, &drl, &do_, &dneg, &loss}) check_f32(*t, "rescal");
+  int B = (int)loss.numel();
+  TORCH_CHECK(D <= 196, "rescal_step: dim must be <= 196 (R tile staged in LDS)");
+  if (s.is_cuda()) {
+    TORCH_CHECK(hip_available(), "rescal_step: CUDA tensor but no HIP device");
+    rescal_step_gpu(cfp(s), cfp(r), cfp(o), cfp(neg), fp(ds), fp(drl), fp(do_), fp(dneg),
+                    fp(loss), B, (int)N, (int)D, (float)lr, (float)eps,
+                    current_stream(s.device()));
+  } else {
+    rescal_step_cpu(cfp(s), cfp(r), cfp(o), cfp(neg), fp(ds), fp(drl), fp(do_), fp(dneg),
+                    fp(loss), B, (int)N, (int)D, (float)lr, (float)eps);
+  }
+}
+
 void kge_complex_score(torch::Tensor s, torch::Tensor r, torch::Tensor cand,
                        torch::Tensor scores, int64_t D) {
   for (auto* t : {&s, &r, &cand, &scores}) check_f32(*t, "kge tensor");
@@ -2083,6 +2100,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hip_available", &hip_available);
   m.def("kge_complex_step", &kge_complex_step, py::call_guard<py::gil_scoped_release>());
   m.def("alias_draw", &alias_draw, py::call_guard<py::gil_scoped_release>());
+  m.def("rescal_step", &rescal_step, py::call_guard<py::gil_scoped_release>());
   m.def("kge_complex_score", &kge_complex_score, py::call_guard<py::gil_scoped_release>());
   m.def("w2v_sgns_step", &w2v_sgns_step, py::call_guard<py::gil_scoped_release>());
   m.def("mf_update_step", &mf_update_step, py::call_guard<py::gil_scoped_release>());

@@ -34,6 +34,18 @@ void kge_complex_score_gpu(const float* s, const float* r, const float* cand, fl
 void kge_complex_score_cpu(const float* s, const float* r, const float* cand, float* scores,
                            int B, int E, int D);
 
+// RESCAL training step (reference knowledge_graph_embeddings.cc:895-922):
+// score = e_s^T R e_o, R is D x D. Entity rows [emb(D) | accum(D)],
+// relation rows [R(D*D) | accum(D*D)]. Math uses the factorization
+// u = R^T e_s (once), w = sum_o c_o e_o, de_s = R w, dR = e_s w^T — so a
+// triple with N negatives costs ~3 D^2 + N D instead of N D^2.
+void rescal_step_gpu(const float* s, const float* r, const float* o, const float* neg,
+                     float* ds, float* drl, float* do_, float* dneg, float* loss, int B, int N,
+                     int D, float lr, float eps, void* stream);
+void rescal_step_cpu(const float* s, const float* r, const float* o, const float* neg,
+                     float* ds, float* drl, float* do_, float* dneg, float* loss, int B, int N,
+                     int D, float lr, float eps);
+
 // word2vec SGNS step: per center/context pair with N negatives.
 //  ctr:  [B][2D] center (syn0) rows;  ctx: [B][2D] context (syn1) rows
 //  neg:  [B*N][2D] negative (syn1) rows

@@ -157,3 +157,63 @@ void alias_draw_cpu(const float* prob, const int32_t* alias, int64_t n, uint64_t
 }
 
 }  // namespace adapm
+
+namespace adapm {
+
+void rescal_step_cpu(const float* s, const float* r, const float* o, const float* neg,
+                     float* ds, float* drl, float* do_, float* dneg, float* loss, int B, int N,
+                     int D, float lr, float eps) {
+  const int erow = 2 * D;
+  const int64_t rrow = 2LL * D * D;
+  std::vector<float> u(D), w(D), t(D);
+  for (int b = 0; b < B; ++b) {
+    const float* sb = s + (int64_t)b * erow;
+    const float* rb = r + (int64_t)b * rrow;
+    // u = R^T e_s
+    for (int j = 0; j < D; ++j) {
+      double a = 0;
+      for (int i = 0; i < D; ++i) a += rb[(int64_t)i * D + j] * sb[i];
+      u[j] = (float)a;
+    }
+    std::fill(w.begin(), w.end(), 0.f);
+    float lsum = 0.f;
+    for (int j = 0; j <= N; ++j) {
+      const float* ob = (j == 0) ? o + (int64_t)b * erow : neg + ((int64_t)b * N + j - 1) * erow;
+      float* dob = (j == 0) ? do_ + (int64_t)b * erow : dneg + ((int64_t)b * N + j - 1) * erow;
+      float y = (j == 0) ? 1.f : -1.f;
+      double dot = 0;
+      for (int k = 0; k < D; ++k) dot += u[k] * ob[k];
+      float c = -y * sigmoidf_(-y * (float)dot);
+      lsum += softplusf_(-y * (float)dot);
+      for (int k = 0; k < D; ++k) {
+        w[k] += c * ob[k];
+        float g = c * u[k];
+        dob[k] = -lr * g / std::sqrt(ob[D + k] + g * g + eps);
+        dob[D + k] = g * g;
+      }
+    }
+    loss[b] = lsum;
+    // de_s = R w
+    float* dsb = ds + (int64_t)b * erow;
+    for (int i = 0; i < D; ++i) {
+      double a = 0;
+      for (int k = 0; k < D; ++k) a += rb[(int64_t)i * D + k] * w[k];
+      t[i] = (float)a;
+      float g = t[i];
+      dsb[i] = -lr * g / std::sqrt(sb[D + i] + g * g + eps);
+      dsb[D + i] = g * g;
+    }
+    // dR = e_s w^T
+    float* drb = drl + (int64_t)b * rrow;
+    for (int i = 0; i < D; ++i) {
+      for (int k = 0; k < D; ++k) {
+        float g = sb[i] * w[k];
+        int64_t idx = (int64_t)i * D + k;
+        drb[idx] = -lr * g / std::sqrt(rb[(int64_t)D * D + idx] + g * g + eps);
+        drb[(int64_t)D * D + idx] = g * g;
+      }
+    }
+  }
+}
+
+}  // namespace adapm

@@ -334,3 +334,93 @@ void alias_draw_gpu(const float* prob, const int32_t* alias, int64_t n, uint64_t
                      n, seed, N, out);
 }
 }  // namespace adapm
+
+namespace adapm {
+
+// RESCAL step: one workgroup per positive triple; R staged in LDS
+// (dim <= 128 -> 64 KiB). See kernels.h for the math factorization.
+__global__ void k_rescal_step(const float* __restrict__ s, const float* __restrict__ r,
+                              const float* __restrict__ o, const float* __restrict__ neg,
+                              float* __restrict__ ds, float* __restrict__ drl,
+                              float* __restrict__ do_, float* __restrict__ dneg,
+                              float* __restrict__ loss, int B, int N, int D, float lr,
+                              float eps) {
+  extern __shared__ __attribute__((aligned(16))) float smem[];
+  float* Rm = smem;            // D*D
+  float* u = Rm + D * D;       // D
+  float* w = u + D;            // D
+  float* es = w + D;           // D
+  float* red = es + D;         // KT/64 reduction scratch
+  const int erow = 2 * D;
+  const int64_t rrow = 2LL * D * D;
+
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    const float* sb = s + (int64_t)b * erow;
+    const float* rb = r + (int64_t)b * rrow;
+    for (int i = threadIdx.x; i < D * D; i += KT) Rm[i] = rb[i];
+    for (int i = threadIdx.x; i < D; i += KT) {
+      es[i] = sb[i];
+      w[i] = 0.f;
+    }
+    __syncthreads();
+    // u = R^T e_s: thread j computes column j
+    for (int j = threadIdx.x; j < D; j += KT) {
+      float a = 0.f;
+      for (int i = 0; i < D; ++i) a += Rm[i * D + j] * es[i];
+      u[j] = a;
+    }
+    __syncthreads();
+
+    float lsum = 0.f;
+    for (int j = 0; j <= N; ++j) {
+      const float* ob = (j == 0) ? o + (int64_t)b * erow
+                                 : neg + ((int64_t)b * N + j - 1) * erow;
+      float* dob = (j == 0) ? do_ + (int64_t)b * erow
+                            : dneg + ((int64_t)b * N + j - 1) * erow;
+      float y = (j == 0) ? 1.f : -1.f;
+      float part = 0.f;
+      for (int k = threadIdx.x; k < D; k += KT) part += u[k] * ob[k];
+      float dot = block_reduce_sum(part, red);
+      float c = -y * sigmoidf(-y * dot);
+      if (threadIdx.x == 0) lsum += softplusf(-y * dot);
+      for (int k = threadIdx.x; k < D; k += KT) {
+        w[k] += c * ob[k];  // single writer per k: thread-private index
+        float g = c * u[k];
+        dob[k] = -lr * g * __frsqrt_rn(ob[D + k] + g * g + eps);
+        dob[D + k] = g * g;
+      }
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) loss[b] = lsum;
+
+    // de_s = R w
+    float* dsb = ds + (int64_t)b * erow;
+    for (int i = threadIdx.x; i < D; i += KT) {
+      float a = 0.f;
+      for (int k = 0; k < D; ++k) a += Rm[i * D + k] * w[k];
+      float g = a;
+      dsb[i] = -lr * g * __frsqrt_rn(sb[D + i] + g * g + eps);
+      dsb[D + i] = g * g;
+    }
+    // dR = e_s w^T
+    float* drb = drl + (int64_t)b * rrow;
+    for (int idx = threadIdx.x; idx < D * D; idx += KT) {
+      int i = idx / D, k = idx % D;
+      float g = es[i] * w[k];
+      drb[idx] = -lr * g * __frsqrt_rn(rb[(int64_t)D * D + idx] + g * g + eps);
+      drb[(int64_t)D * D + idx] = g * g;
+    }
+    __syncthreads();
+  }
+}
+
+void rescal_step_gpu(const float* s, const float* r, const float* o, const float* neg,
+                     float* ds, float* drl, float* do_, float* dneg, float* loss, int B, int N,
+                     int D, float lr, float eps, void* stream) {
+  size_t smem = (size_t)(D * D + 3 * D + KT / 64) * sizeof(float);
+  int blocks = (int)std::min<int64_t>(B, 8192);
+  hipLaunchKernelGGL(k_rescal_step, dim3(blocks), dim3(KT), smem, (hipStream_t)stream, s, r, o,
+                     neg, ds, drl, do_, dneg, loss, B, N, D, lr, eps);
+}
+
+}  // namespace adapm

@@ -17,7 +17,6 @@ Rebuild of the reference app (reference apps/knowledge_graph_embeddings.cc):
 from __future__ import annotations
 
 import dataclasses
-import math
 import os
 import time
 from collections import defaultdict
@@ -427,3 +426,86 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+class Rescal:
+    """RESCAL scorer (reference knowledge_graph_embeddings.cc:895-922):
+    psi = e_s^T R e_o with a D x D matrix per relation. Exercises the
+    store's NON-UNIFORM value lengths: entity rows 2*D floats, relation
+    rows 2*D*D floats."""
+
+    def __init__(self, cfg: ComplExConfig, server, worker):
+        self.cfg = cfg
+        self.server = server
+        self.worker = worker
+        self.dev = server.rt.device
+        self.rank = server.rt.rank
+        self.world = server.rt.world
+        self.rng = np.random.default_rng(cfg.seed + self.rank)
+        self._pending = []
+
+    @staticmethod
+    def value_lengths(num_entities, num_relations, dim):
+        lens = np.full(num_entities + num_relations, 2 * dim, dtype=np.int64)
+        lens[num_entities:] = 2 * dim * dim
+        return lens
+
+    def init_embeddings(self):
+        cfg = self.cfg
+        my_keys = np.arange(self.rank, cfg.num_keys, self.world, dtype=np.int64)
+        ents = my_keys[my_keys < cfg.num_entities]
+        rels = my_keys[my_keys >= cfg.num_entities]
+        chunk = max(1, 2 ** 24 // (2 * cfg.dim))
+        for i in range(0, len(ents), chunk):
+            ks = ents[i:i + chunk]
+            v = torch.zeros(len(ks), 2 * cfg.dim, dtype=torch.float32, device=self.dev)
+            v[:, :cfg.dim].normal_(0, cfg.init_scale)
+            self.worker.set(ks, v)
+        rchunk = max(1, 2 ** 24 // (2 * cfg.dim * cfg.dim))
+        for i in range(0, len(rels), rchunk):
+            ks = rels[i:i + rchunk]
+            v = torch.zeros(len(ks), 2 * cfg.dim * cfg.dim, dtype=torch.float32,
+                            device=self.dev)
+            v[:, :cfg.dim * cfg.dim].normal_(0, cfg.init_scale / cfg.dim ** 0.5)
+            self.worker.set(ks, v)
+        self.worker.wait_sync()
+        self.worker.barrier()
+
+    def train_batch(self, triples: np.ndarray, sync_loss: bool = True):
+        cfg = self.cfg
+        w = self.worker
+        B = len(triples)
+        D = cfg.dim
+        s_keys = triples[:, 0].astype(np.int64)
+        r_keys = (cfg.num_entities + triples[:, 1]).astype(np.int64)
+        o_keys = triples[:, 2].astype(np.int64)
+        if self.server.sampling is not None:
+            sid = w.prepare_sample(B * cfg.neg_samples, w.current_clock(), w.current_clock() + 2)
+            neg_keys = self.server.sampling.pull(w, sid, B * cfg.neg_samples)
+            w.finish_sample(sid)
+        else:
+            neg_keys = self.rng.integers(0, cfg.num_entities, size=B * cfg.neg_samples,
+                                         dtype=np.int64)
+        opts = dict(dtype=torch.float32, device=self.dev)
+        s_v = torch.empty(B, 2 * D, **opts)
+        r_v = torch.empty(B, 2 * D * D, **opts)
+        o_v = torch.empty(B, 2 * D, **opts)
+        n_v = torch.empty(B * cfg.neg_samples, 2 * D, **opts)
+        for kt, vt in ((s_keys, s_v), (r_keys, r_v), (o_keys, o_v), (neg_keys, n_v)):
+            w.wait(w.pull(kt, vt, async_=True))
+        ds, drl, do, dn = (torch.empty_like(t) for t in (s_v, r_v, o_v, n_v))
+        loss = torch.empty(B, dtype=torch.float32, device=self.dev)
+        _C.rescal_step(s_v, r_v, o_v, n_v, ds, drl, do, dn, loss,
+                       cfg.neg_samples, D, cfg.lr, cfg.eps)
+        for kt, dt in ((s_keys, ds), (r_keys, drl), (o_keys, do), (neg_keys, dn)):
+            pt = w.push(kt, dt, async_=True)
+            if pt != -1:
+                self._pending.append(pt)
+        while len(self._pending) > 64:
+            w.wait(self._pending.pop(0))
+        return float(loss.mean().item()) if sync_loss else loss
+
+    def drain(self):
+        for t in self._pending:
+            self.worker.wait(t)
+        self._pending.clear()

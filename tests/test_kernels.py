@@ -209,3 +209,84 @@ def test_mf_step_cpu():
 @pytest.mark.gpu
 def test_mf_step_gpu():
     run_mf("cuda:0", B=16, R=128)
+
+
+# ------------------------------------------------------------ RESCAL
+
+def torch_rescal_ref(s, r, o, neg, N, D, lr, eps):
+    B = s.shape[0]
+    se = s[:, :D].clone().detach().requires_grad_(True)
+    Re = r[:, :D * D].clone().detach().requires_grad_(True)
+    oe = o[:, :D].clone().detach().requires_grad_(True)
+    ne = neg[:, :D].clone().detach().requires_grad_(True)
+    R = Re.view(B, D, D)
+    pos = torch.einsum("bi,bij,bj->b", se, R, oe)
+    loss = torch.nn.functional.softplus(-pos).sum()
+    nv = ne.view(B, N, D)
+    for j in range(N):
+        sc = torch.einsum("bi,bij,bj->b", se, R, nv[:, j, :])
+        loss = loss + torch.nn.functional.softplus(sc).sum()
+    loss.backward()
+
+    def delta(full, leaf, L):
+        g = leaf.grad.reshape(full.shape[0], L)
+        G = full[:, L:] + g * g
+        return torch.cat([-lr * g / torch.sqrt(G + eps), g * g], dim=1)
+
+    return (delta(s, se, D), delta(r, Re, D * D), delta(o, oe, D), delta(neg, ne, D))
+
+
+def run_rescal(device, B=3, N=2, D=16, tol=3e-4):
+    g = torch.Generator().manual_seed(5)
+    mke = lambda n: torch.cat([torch.randn(n, D, generator=g) * 0.3,
+                               torch.rand(n, D, generator=g) * 0.1], dim=1)
+    s, o, neg = mke(B), mke(B), mke(B * N)
+    r = torch.cat([torch.randn(B, D * D, generator=g) * 0.1,
+                   torch.rand(B, D * D, generator=g) * 0.1], dim=1)
+    lr, eps = 0.05, 1e-6
+    sd, rd, od, nd = (t.to(device) for t in (s, r, o, neg))
+    ds, dr, do, dn = (torch.empty_like(t) for t in (sd, rd, od, nd))
+    loss = torch.empty(B, dtype=torch.float32, device=device)
+    from adapm_amd import _C as C2
+
+    C2.rescal_step(sd, rd, od, nd, ds, dr, do, dn, loss, N, D, lr, eps)
+    eds, edr, edo, edn = torch_rescal_ref(s, r, o, neg, N, D, lr, eps)
+    for got, exp, name in [(ds, eds, "ds"), (dr, edr, "dR"), (do, edo, "do"), (dn, edn, "dn")]:
+        err = (got.cpu() - exp).abs().max().item()
+        assert err < tol, f"{name} max err {err}"
+
+
+def test_rescal_cpu():
+    run_rescal("cpu")
+
+
+@pytest.mark.gpu
+def test_rescal_gpu():
+    run_rescal("cuda:0", B=8, N=4, D=128)
+
+
+def test_rescal_model_cpu():
+    """End-to-end RESCAL on the store with NON-UNIFORM value lengths."""
+    import adapm_amd
+    from adapm_amd.models.kge import ComplExConfig, Rescal, make_synthetic_triples
+
+    adapm_amd._SETUP.clear()
+    adapm_amd.runtime._RUNTIME = None
+    E, R, D = 200, 10, 16
+    adapm_amd.setup(num_keys=E + R, num_threads=1, device="cpu")
+    lens = Rescal.value_lengths(E, R, D)
+    server = adapm_amd.Server(torch.from_numpy(lens))
+    worker = adapm_amd.Worker(0, server)
+    cfg = ComplExConfig(num_entities=E, num_relations=R, dim=D, neg_samples=2,
+                        batch_size=64, lr=0.1)
+    model = Rescal(cfg, server, worker)
+    model.init_embeddings()
+    triples = make_synthetic_triples(256, E, R, seed=2)
+    losses = []
+    for ep in range(6):
+        for i in range(0, len(triples), 64):
+            losses.append(model.train_batch(triples[i:i + 64]))
+    model.drain()
+    assert losses[-1] < losses[0], losses[:3] + losses[-3:]
+    worker.finalize()
+    server.shutdown()
