@@ -1793,22 +1793,26 @@ class Server {
     auto out = torch::empty({n}, torch::TensorOptions().dtype(torch::kInt64));
     const int64_t* cp = c.data_ptr<int64_t>();
     int64_t* op = out.data_ptr<int64_t>();
-    int64_t checks = 0;
-    for (int64_t i = 0; i < n; ++i) {
-      Key k = cp[i];
-      Key start = k;
-      while (true) {
-        checks++;
-        uint8_t f = flags_[k];
-        if ((f & F_PRESENT) && !(f & F_STUB)) break;
-        k++;
-        if (k >= hi) k = lo;
-        if (k == start) break;  // nothing local in range: keep candidate
+    std::atomic<int64_t> checks{0};
+    at::parallel_for(0, n, 4096, [&](int64_t b, int64_t e) {
+      int64_t local_checks = 0;
+      for (int64_t i = b; i < e; ++i) {
+        Key k = cp[i];
+        Key start = k;
+        while (true) {
+          local_checks++;
+          uint8_t f = flags_[k].load(std::memory_order_relaxed);
+          if ((f & F_PRESENT) && !(f & F_STUB)) break;
+          k++;
+          if (k >= hi) k = lo;
+          if (k == start) break;  // nothing local in range: keep candidate
+        }
+        op[i] = k;
       }
-      op[i] = k;
-    }
-    stat_sampling_checks_ += checks;
-    return {out, checks};
+      checks += local_checks;
+    });
+    stat_sampling_checks_ += checks.load();
+    return {out, checks.load()};
   }
 
   // ------------------------------------------------ info / stats

@@ -186,6 +186,16 @@ class SamplingManager:
         if self.server.world() == 1:
             keys = cands  # single rank: every key is local, no scan needed
         else:
+            # snap each candidate to the nearest key this rank OWNS BY
+            # CONSTRUCTION (manager = key % world): the scan below then
+            # hits on the first probe for everything that has not been
+            # relocated away. This is the NuPS "local" substitution —
+            # distribution preserved approximately (reference sampling.h
+            # Local scheme does the same kind of substitution).
+            w, r = self.server.world(), self.server.rank()
+            cands = cands - (cands % w) + r
+            cands = np.where(cands >= self.hi, cands - w, cands)
+            cands = np.where(cands < self.lo, cands + w, cands)
             keys, _checks = self.server.scan_local(torch.from_numpy(cands), self.lo, self.hi)
             keys = keys.numpy()
         if not self.with_replacement:
