@@ -131,10 +131,12 @@ class Server:
     # ---- reference Server API (bindings.cc:88-146)
 
     def enable_sampling_support(self, scheme: str, with_replacement: bool, distribution: str,
-                                min: int, max: int, counts=None, power: float = 0.75):
+                                min: int, max: int, counts=None, power: float = 0.75,
+                                pool_size: int = 5_000, reuse_factor: int = 4):
         d = make_distribution(distribution, min, max, seed=self.rt.rank, counts=counts,
                               power=power, device=self.rt.device)
-        self.sampling = SamplingManager(self._s, scheme, with_replacement, d, min, max)
+        self.sampling = SamplingManager(self._s, scheme, with_replacement, d, min, max,
+                                        pool_size=pool_size, reuse_factor=reuse_factor)
 
     def barrier(self):
         if self.rt.world > 1:
@@ -338,6 +340,13 @@ class Worker:
 
     def wait_sync(self):
         self.server.wait_sync()
+
+    def wait_replica_sync(self):  # deprecated alias (reference bindings.cc:355)
+        import warnings
+
+        warnings.warn("wait_replica_sync() is deprecated; use wait_sync()",
+                      DeprecationWarning)
+        self.wait_sync()
 
     def barrier(self):
         """Barrier across ALL worker threads of ALL ranks (reference

@@ -101,11 +101,10 @@ class SamplingManager:
     """Per-rank sampling state shared by all workers (reference sampling.h
     keeps it on the server; sample ids are (wid, counter))."""
 
-    POOL_SIZE = 5_000
-    POOL_REUSE = 4
-
     def __init__(self, server, scheme: str, with_replacement: bool, dist_: Distribution,
-                 lo: int, hi: int):
+                 lo: int, hi: int, pool_size: int = 5_000, reuse_factor: int = 4):
+        self.POOL_SIZE = pool_size    # reference --sampling.pool_size
+        self.POOL_REUSE = reuse_factor  # reference --sampling.reuse_factor
         self.server = server
         self.scheme = scheme
         self.with_replacement = with_replacement
