@@ -37,6 +37,7 @@
 #include <atomic>
 #include <condition_variable>
 #include <cstring>
+#include <limits>
 #include <deque>
 #include <map>
 #include <queue>
@@ -156,6 +157,8 @@ struct Slab {
     throw std::runtime_error("adapm slab out of capacity (device and host-spill)");
   }
 
+  bool poison = getenv("ADAPM_DEBUG_POISON") != nullptr;
+
   void free_(int64_t off, int32_t len) {
     int32_t p = padded(len);
     std::lock_guard<std::mutex> g(mu);
@@ -163,6 +166,11 @@ struct Slab {
       host_freelists[p].push_back(off & ~SPILL_BIT);
       host_in_use -= p;
     } else {
+      if (poison) {
+        // debug mode: NaN-fill freed slots so use-after-free reads are
+        // loud (stream-ordered, so this poisons only after prior readers)
+        buf.narrow(0, off, p).fill_(std::numeric_limits<float>::quiet_NaN());
+      }
       freelists[p].push_back(off);
       in_use -= p;
     }
