@@ -502,10 +502,9 @@ class Server {
         // every offset from the key (see layout_identity_). The host only
         // scans for non-owned keys (pure arithmetic, nothing when world=1).
         const int32_t l = uniform_len_;
-        if (world_ > 1) {
-          for (int64_t i = 0; i < n; ++i) {
-            if (kp[i] % world_ != rank_) remote.push_back({kp[i], i});
-          }
+        for (int64_t i = 0; i < n; ++i) {
+          TORCH_CHECK((uint64_t)kp[i] < (uint64_t)num_keys_, "key out of range: ", kp[i]);
+          if (world_ > 1 && kp[i] % world_ != rank_) remote.push_back({kp[i], i});
         }
         if (!remote.empty()) {
           for (int64_t i = 0; i < n; ++i) {
@@ -522,9 +521,14 @@ class Server {
         // lock-free parallel metadata pass (atomic reads; see meta rules)
         if (uniform_len_ >= 0) {
           const int32_t l = uniform_len_;
-          for (int64_t i = 0; i < n; ++i) { out_off[i] = i * (int64_t)l; out_len[i] = l; }
+          for (int64_t i = 0; i < n; ++i) {
+            TORCH_CHECK((uint64_t)kp[i] < (uint64_t)num_keys_, "key out of range: ", kp[i]);
+            out_off[i] = i * (int64_t)l;
+            out_len[i] = l;
+          }
         } else {
           for (int64_t i = 0; i < n; ++i) {
+            TORCH_CHECK((uint64_t)kp[i] < (uint64_t)num_keys_, "key out of range: ", kp[i]);
             int32_t l = len_of(kp[i]);
             out_off[i] = cum;
             out_len[i] = l;
@@ -649,10 +653,9 @@ class Server {
         // layout_identity_ when the first replica of one of our keys is
         // granted, so while the flag holds versions are unobserved.
         const int32_t l = uniform_len_;
-        if (world_ > 1) {
-          for (int64_t i = 0; i < n; ++i) {
-            if (kp[i] % world_ != rank_) remote.push_back({kp[i], i * (int64_t)l, l});
-          }
+        for (int64_t i = 0; i < n; ++i) {
+          TORCH_CHECK((uint64_t)kp[i] < (uint64_t)num_keys_, "key out of range: ", kp[i]);
+          if (world_ > 1 && kp[i] % world_ != rank_) remote.push_back({kp[i], i * (int64_t)l, l});
         }
         if (cpp_timing_) { int64_t t1 = tick(); t_pass_ += t1 - tp0; tp0 = t1; }
         run_scatter_keys(keys, flat, set_mode);
@@ -663,9 +666,13 @@ class Server {
         // lock-free parallel metadata pass (atomic reads; see meta rules)
         std::vector<int64_t> offs(n);
         if (uniform_len_ >= 0) {
-          for (int64_t i = 0; i < n; ++i) offs[i] = i * (int64_t)uniform_len_;
+          for (int64_t i = 0; i < n; ++i) {
+            TORCH_CHECK((uint64_t)kp[i] < (uint64_t)num_keys_, "key out of range: ", kp[i]);
+            offs[i] = i * (int64_t)uniform_len_;
+          }
         } else {
           for (int64_t i = 0; i < n; ++i) {
+            TORCH_CHECK((uint64_t)kp[i] < (uint64_t)num_keys_, "key out of range: ", kp[i]);
             offs[i] = cum;
             cum += len_of(kp[i]);
           }
@@ -796,8 +803,14 @@ class Server {
   // Intent: announce access to keys in clock window [start, end)
   // (reference coloc_kv_worker.h:368-408). No-op on a single node.
   void intent(int wid, torch::Tensor keys, Clock start, Clock end) {
-    if (world_ == 1) return;
     check_keys(keys);
+    {
+      const int64_t* kp0 = keys.data_ptr<int64_t>();
+      for (int64_t i = 0; i < keys.numel(); ++i)
+        TORCH_CHECK((uint64_t)kp0[i] < (uint64_t)num_keys_, "key out of range: ", kp0[i]);
+    }
+    if (world_ == 1) return;  // single node: intents are no-ops (reference
+                              // coloc_kv_worker.h:728)
     if (end == 0) end = start + 1;
     int64_t n = keys.numel();
     const int64_t* kp = keys.data_ptr<int64_t>();

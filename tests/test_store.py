@@ -123,3 +123,15 @@ def test_staggered_push():
     w.pull(keys, out)
     assert torch.equal(out, vals)
     s.shutdown()
+
+
+def test_out_of_range_key_raises():
+    s = make_server(num_keys=10, lens=4)
+    w = adapm_amd.Worker(0, s)
+    with pytest.raises(RuntimeError):
+        w.push(torch.tensor([11]), torch.zeros(1, 4))
+    with pytest.raises(RuntimeError):
+        w.pull(torch.tensor([-1]), torch.zeros(1, 4))
+    with pytest.raises(RuntimeError):
+        w.intent(torch.tensor([99]), 1, 5)
+    s.shutdown()
