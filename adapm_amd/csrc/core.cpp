@@ -891,8 +891,20 @@ class Server {
       C.future_intents.clear();
     }
 
-    // 1. register new intents (reference sync_manager.h registerNewIntents)
+    // 1. register new intents (reference sync_manager.h registerNewIntents).
+    // Stub slots are ZEROED BEFORE the key is published (flags set): a
+    // worker push that lands right after publication would otherwise be
+    // wiped by a later zero (exact-sum loss, found by the stress hunt).
+    // Only this channel's thread creates stubs for its keys, so the
+    // absent-check before allocation cannot race another creator.
     HostBatch zero_batch;
+    struct NewStub {
+      Key k;
+      int64_t v_off, s_off;
+    };
+    std::vector<NewStub> new_stubs;
+    std::unordered_set<Key> new_stub_keys;
+    std::vector<IntentReq> due;
     for (auto& req : intents_in) {
       Clock now = clocks_[req.wid].load();
       if (req.end <= now) continue;  // already expired
@@ -901,40 +913,40 @@ class Server {
         C.future_intents.push_back(std::move(req));
         continue;
       }
-      std::vector<Key> stubs;
       for (Key k : req.keys) {
-        uint8_t f0 = flags_[k].load(std::memory_order_acquire);
-        if (!(f0 & F_PRESENT)) {
+        if (!(flags_[k].load(std::memory_order_acquire) & F_PRESENT)) {
+          if (!new_stub_keys.insert(k).second) continue;  // dedup within round
           int32_t l = len_of(k);
-          std::lock_guard<std::mutex> lk(stripe(k));
-          uint8_t f = flags_[k];
-          if (!(f & F_PRESENT)) {
-            // replica stub: zeroed val+sync, absorbs pushes until first
-            // refresh (reference handle registerNewIntentsForKeyUnsafe)
-            layout_identity_.store(false, std::memory_order_release);
-            int64_t v_off = slab_.alloc(l);
-            int64_t s_off = slab_.alloc(l);
-            loc_[k] = v_off;
-            sync_loc_[k] = s_off;
-            flags_[k] = F_PRESENT | F_STUB;
-            version_[k] = 0;
-            stubs.push_back(k);
-            zero_batch.add(0, v_off, l);   // freelist reuse leaves stale data
-            zero_batch.add(0, s_off, l);
-            trace_event(k, "REPLICA_SETUP");
-          }
+          layout_identity_.store(false, std::memory_order_release);
+          int64_t v_off = slab_.alloc(l);
+          int64_t s_off = slab_.alloc(l);
+          zero_batch.add(0, v_off, l);  // freelist reuse leaves stale data
+          zero_batch.add(0, s_off, l);
+          new_stubs.push_back({k, v_off, s_off});
         }
       }
-      {
-        std::lock_guard<std::mutex> g(C.mu);
+      due.push_back(std::move(req));
+    }
+    run_zero(zero_batch);  // CPU: inline; GPU: stream-ordered before any
+                           // later worker merge kernel on these slots
+    for (auto& ns : new_stubs) {
+      std::lock_guard<std::mutex> lk(stripe(ns.k));
+      loc_[ns.k] = ns.v_off;      // loc before flags (meta rules)
+      sync_loc_[ns.k] = ns.s_off;
+      version_[ns.k] = 0;
+      flags_[ns.k] = F_PRESENT | F_STUB;
+      trace_event(ns.k, "REPLICA_SETUP");
+    }
+    {
+      std::lock_guard<std::mutex> g(C.mu);
+      for (auto& req : due) {
         for (Key k : req.keys) {
           C.intents[k].push_back({req.wid, req.end});
           C.intent_expiry.push({req.end, req.wid, k});
         }
-        for (Key k : stubs) C.replicas.insert(k);
       }
+      for (auto& ns : new_stubs) C.replicas.insert(ns.k);
     }
-    run_zero(zero_batch);
 
     // 2. expire due intents (heap pop — no full-map sweep) and snapshot
     // the replica set
