@@ -71,17 +71,17 @@ def _gpu_kge_dist(rank, world):
     model = ComplEx(cfg, server, worker)
     model.init_embeddings()
     triples = make_synthetic_triples(1024, E, R, seed=rank)
-    first = last = None
-    for ep in range(3):
+    epoch_means = []
+    for ep in range(4):
+        losses = []
         for i in range(0, len(triples), cfg.batch_size):
             b = triples[i:i + cfg.batch_size]
             model.signal_intent(b, worker.current_clock() + 1, worker.current_clock() + 4)
-            loss = model.train_batch(b)
-            first = loss if first is None else first
-            last = loss
+            losses.append(model.train_batch(b))
             worker.advance_clock()
+        epoch_means.append(float(np.mean(losses)))
     model.drain()
-    assert worker.allreduce(last) < worker.allreduce(first)
+    assert worker.allreduce(epoch_means[-1]) < worker.allreduce(epoch_means[0]), epoch_means
     worker.barrier()
     worker.finalize()
     server.shutdown()
