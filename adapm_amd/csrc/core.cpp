@@ -1815,6 +1815,58 @@ class Server {
     rounds_cv_.notify_all();
   }
 
+  // ------------------------------------------------ fused app operator
+
+  // Fused ComplEx train step: the kernel reads rows straight from the
+  // HBM slab and atomically accumulates AdaGrad deltas back — no
+  // intermediate pull/push buffers. Requires: single rank (all keys
+  // local-owned), uniform lengths, identity layout, GPU store. The
+  // classic pull/kernel/push path remains the general case. Counts the
+  // same pull+push key-ops in the stats (the store IS read and updated
+  // per key).
+  torch::Tensor kge_step_fused(torch::Tensor keys_s, torch::Tensor keys_r, torch::Tensor keys_o,
+                               torch::Tensor keys_neg, int64_t N, int64_t D, double lr,
+                               double eps) {
+    TORCH_CHECK(world_ == 1, "kge_step_fused requires a single rank (all keys local)");
+    TORCH_CHECK(uniform_len_ == 2 * D, "kge_step_fused: store rows must be [emb|accum] = 2D");
+    TORCH_CHECK(layout_identity_.load(), "kge_step_fused requires the identity layout");
+    TORCH_CHECK(dev_.is_cuda(), "kge_step_fused is the GPU fast path");
+    for (auto* t : {&keys_s, &keys_r, &keys_o, &keys_neg}) check_keys(*t);
+    int64_t B = keys_s.numel();
+    TORCH_CHECK(keys_r.numel() == B && keys_o.numel() == B && keys_neg.numel() == B * N);
+    auto rng_check = [&](const torch::Tensor& t) {
+      const int64_t* kp = t.data_ptr<int64_t>();
+      for (int64_t i = 0; i < t.numel(); ++i)
+        TORCH_CHECK((uint64_t)kp[i] < (uint64_t)num_keys_, "key out of range: ", kp[i]);
+    };
+    rng_check(keys_s);
+    rng_check(keys_r);
+    rng_check(keys_o);
+    rng_check(keys_neg);
+
+    auto loss = torch::empty({B}, torch::TensorOptions().dtype(torch::kFloat32).device(dev_));
+    auto ks = keys_s.to(dev_, true);
+    auto kr = keys_r.to(dev_, true);
+    auto ko = keys_o.to(dev_, true);
+    auto kn = keys_neg.to(dev_, true);
+    {
+      InflightGuard g(this);
+      kge_complex_step_fused_gpu(slab_.data, ks.data_ptr<int64_t>(), kr.data_ptr<int64_t>(),
+                                 ko.data_ptr<int64_t>(), kn.data_ptr<int64_t>(),
+                                 loss.data_ptr<float>(), (int)B, (int)N, (int)D,
+                                 Slab::padded(uniform_len_), world_, rank_, (float)lr,
+                                 (float)eps, current_stream(dev_));
+    }
+    int64_t total = 3 * B + B * N;
+    stat_pull_keys_ += total;
+    stat_pull_local_ += total;
+    stat_push_keys_ += total;
+    stat_push_local_ += total;
+    stat_pulls_ += 1;
+    stat_pushes_ += 1;
+    return loss;
+  }
+
   // ------------------------------------------------ sampling support
 
   // "Local" sampling scheme scan: per candidate, scan upward (wrapping in
@@ -2171,6 +2223,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("fail", &Server::fail, py::call_guard<py::gil_scoped_release>())
       .def("failed_reason", &Server::failed_reason)
       .def("scan_local", &Server::scan_local, py::call_guard<py::gil_scoped_release>())
+      .def("kge_step_fused", &Server::kge_step_fused, py::call_guard<py::gil_scoped_release>())
       .def("get_len", &Server::get_len)
       .def("num_keys", &Server::num_keys)
       .def("rank", &Server::rank)

@@ -27,6 +27,19 @@ void kge_complex_step_cpu(const float* s, const float* r, const float* o, const 
                           float* ds, float* dr, float* do_, float* dneg, float* loss,
                           int B, int N, int D, float lr, float eps);
 
+// FUSED ComplEx step: reads entity/relation rows DIRECTLY from the slab
+// (identity layout: row k at (k/world)*plen) and atomically accumulates
+// the AdaGrad-transformed deltas back — no intermediate pull/push
+// buffers (the classic path pays a gather write + kernel read + delta
+// write + scatter read; this saves all four). Duplicate keys within the
+// batch see hogwild-style concurrent updates (async-PS semantics).
+// keys_* are DEVICE int64 pointers. Keys not owned here are skipped
+// (callers must ensure all-local, e.g. world==1).
+void kge_complex_step_fused_gpu(float* slab, const int64_t* keys_s, const int64_t* keys_r,
+                                const int64_t* keys_o, const int64_t* keys_neg, float* loss,
+                                int B, int N, int D, int32_t plen, int world, int rank,
+                                float lr, float eps, void* stream);
+
 // ComplEx scoring only (evaluation): score[b][e] = psi(s_b, r_b, cand_e)
 //  cand: [E][2D] candidate entity rows; scores: [B][E]
 void kge_complex_score_gpu(const float* s, const float* r, const float* cand, float* scores,

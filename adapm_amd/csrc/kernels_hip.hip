@@ -424,3 +424,117 @@ void rescal_step_gpu(const float* s, const float* r, const float* o, const float
 }
 
 }  // namespace adapm
+
+namespace adapm {
+
+// Fused ComplEx step — see kernels.h. Structure mirrors k_kge_step but
+// rows come from / return to the slab itself.
+template <int KPT>
+__global__ void k_kge_step_fused(float* __restrict__ slab, const int64_t* __restrict__ keys_s,
+                                 const int64_t* __restrict__ keys_r,
+                                 const int64_t* __restrict__ keys_o,
+                                 const int64_t* __restrict__ keys_neg, float* __restrict__ loss,
+                                 int B, int N, int D, int32_t plen, int world, int rank,
+                                 float lr, float eps) {
+  __shared__ float lds[KT / 64];
+  const int dc = D >> 1;
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    const float* sb = slab + (keys_s[b] / world) * (int64_t)plen;
+    const float* rb = slab + (keys_r[b] / world) * (int64_t)plen;
+
+    float s_re[KPT], s_im[KPT], r_re[KPT], r_im[KPT];
+    float a_sre[KPT], a_sim[KPT], a_rre[KPT], a_rim[KPT];
+    float u_re[KPT], u_im[KPT];
+#pragma unroll
+    for (int i = 0; i < KPT; ++i) {
+      int k = threadIdx.x + i * KT;
+      a_sre[i] = a_sim[i] = a_rre[i] = a_rim[i] = 0.f;
+      if (k < dc) {
+        s_re[i] = sb[k];
+        s_im[i] = sb[dc + k];
+        r_re[i] = rb[k];
+        r_im[i] = rb[dc + k];
+        u_re[i] = s_re[i] * r_re[i] - s_im[i] * r_im[i];
+        u_im[i] = s_im[i] * r_re[i] + s_re[i] * r_im[i];
+      } else {
+        s_re[i] = s_im[i] = r_re[i] = r_im[i] = u_re[i] = u_im[i] = 0.f;
+      }
+    }
+
+    float lsum = 0.f;
+    for (int j = 0; j <= N; ++j) {
+      int64_t ok = (j == 0) ? keys_o[b] : keys_neg[(int64_t)b * N + (j - 1)];
+      float* ob = slab + (ok / world) * (int64_t)plen;
+      float y = (j == 0) ? 1.f : -1.f;
+      float part = 0.f;
+      float o_re[KPT], o_im[KPT];
+#pragma unroll
+      for (int i = 0; i < KPT; ++i) {
+        int k = threadIdx.x + i * KT;
+        o_re[i] = k < dc ? ob[k] : 0.f;
+        o_im[i] = k < dc ? ob[dc + k] : 0.f;
+        part += u_re[i] * o_re[i] + u_im[i] * o_im[i];
+      }
+      float psi = block_reduce_sum(part, lds);
+      float c = -y * sigmoidf(-y * psi);
+      if (threadIdx.x == 0) lsum += softplusf(-y * psi);
+
+#pragma unroll
+      for (int i = 0; i < KPT; ++i) {
+        int k = threadIdx.x + i * KT;
+        if (k >= dc) continue;
+        a_sre[i] += c * (r_re[i] * o_re[i] + r_im[i] * o_im[i]);
+        a_sim[i] += c * (r_re[i] * o_im[i] - r_im[i] * o_re[i]);
+        a_rre[i] += c * (s_re[i] * o_re[i] + s_im[i] * o_im[i]);
+        a_rim[i] += c * (s_re[i] * o_im[i] - s_im[i] * o_re[i]);
+        float g_re = c * u_re[i];
+        float g_im = c * u_im[i];
+        float G_re = ob[D + k] + g_re * g_re;
+        float G_im = ob[D + dc + k] + g_im * g_im;
+        atomicAdd(&ob[k], -lr * g_re * __frsqrt_rn(G_re + eps));
+        atomicAdd(&ob[dc + k], -lr * g_im * __frsqrt_rn(G_im + eps));
+        atomicAdd(&ob[D + k], g_re * g_re);
+        atomicAdd(&ob[D + dc + k], g_im * g_im);
+      }
+    }
+    float* dsb = const_cast<float*>(sb);
+    float* drb = const_cast<float*>(rb);
+#pragma unroll
+    for (int i = 0; i < KPT; ++i) {
+      int k = threadIdx.x + i * KT;
+      if (k >= dc) continue;
+      float Gsr = sb[D + k] + a_sre[i] * a_sre[i];
+      float Gsi = sb[D + dc + k] + a_sim[i] * a_sim[i];
+      atomicAdd(&dsb[k], -lr * a_sre[i] * __frsqrt_rn(Gsr + eps));
+      atomicAdd(&dsb[dc + k], -lr * a_sim[i] * __frsqrt_rn(Gsi + eps));
+      atomicAdd(&dsb[D + k], a_sre[i] * a_sre[i]);
+      atomicAdd(&dsb[D + dc + k], a_sim[i] * a_sim[i]);
+      float Grr = rb[D + k] + a_rre[i] * a_rre[i];
+      float Gri = rb[D + dc + k] + a_rim[i] * a_rim[i];
+      atomicAdd(&drb[k], -lr * a_rre[i] * __frsqrt_rn(Grr + eps));
+      atomicAdd(&drb[dc + k], -lr * a_rim[i] * __frsqrt_rn(Gri + eps));
+      atomicAdd(&drb[D + k], a_rre[i] * a_rre[i]);
+      atomicAdd(&drb[D + dc + k], a_rim[i] * a_rim[i]);
+    }
+    if (threadIdx.x == 0) loss[b] = lsum;
+  }
+}
+
+void kge_complex_step_fused_gpu(float* slab, const int64_t* keys_s, const int64_t* keys_r,
+                                const int64_t* keys_o, const int64_t* keys_neg, float* loss,
+                                int B, int N, int D, int32_t plen, int world, int rank,
+                                float lr, float eps, void* stream) {
+  int dc = D >> 1;
+  dim3 g((unsigned)std::min<int64_t>(B, 16384)), t(KT);
+  auto st = (hipStream_t)stream;
+#define LAUNCHF(KPT) \
+  hipLaunchKernelGGL(k_kge_step_fused<KPT>, g, t, 0, st, slab, keys_s, keys_r, keys_o, \
+                     keys_neg, loss, B, N, D, plen, world, rank, lr, eps)
+  if (dc <= KT) LAUNCHF(1);
+  else if (dc <= 2 * KT) LAUNCHF(2);
+  else if (dc <= 4 * KT) LAUNCHF(4);
+  else LAUNCHF(8);
+#undef LAUNCHF
+}
+
+}  // namespace adapm

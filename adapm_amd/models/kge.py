@@ -164,6 +164,31 @@ class ComplEx:
             return out
         return loss
 
+    def train_batch_fused(self, triples: np.ndarray, sync_loss: bool = False):
+        """Single-rank GPU fast path: one fused kernel reads the rows
+        straight from the HBM slab and accumulates the AdaGrad deltas
+        back (Server.kge_step_fused) — no pull/push buffers. Falls back
+        to train_batch when the preconditions don't hold."""
+        cfg = self.cfg
+        w = self.worker
+        if self.world != 1 or self.dev.type != "cuda":
+            return self.train_batch(triples, sync_loss=sync_loss)
+        B = len(triples)
+        s_keys, r_keys, o_keys = self.keys_of(triples)
+        if self.server.sampling is not None:
+            sid = w.prepare_sample(B * cfg.neg_samples, w.current_clock(),
+                                   w.current_clock() + 2)
+            neg_keys = self.server.sampling.pull(w, sid, B * cfg.neg_samples)
+            w.finish_sample(sid)
+        else:
+            neg_keys = self.rng.integers(0, cfg.num_entities, size=B * cfg.neg_samples,
+                                         dtype=np.int64)
+        loss = self.server.raw.kge_step_fused(
+            torch.from_numpy(s_keys), torch.from_numpy(r_keys), torch.from_numpy(o_keys),
+            torch.from_numpy(np.ascontiguousarray(neg_keys, dtype=np.int64)),
+            cfg.neg_samples, cfg.dim, cfg.lr, cfg.eps)
+        return float(loss.mean().item()) if sync_loss else loss
+
     def drain(self):
         for t in self._pending:
             self.worker.wait(t)

@@ -41,6 +41,8 @@ def main():
                     help="disable intent signaling (pure remote-op mode)")
     ap.add_argument("--pipeline", type=int, default=0,
                     help="prefetch depth (bounded async, reference max_concurrent_loops); 0 = blocking per step")
+    ap.add_argument("--no-fused", action="store_true",
+                    help="disable the fused slab-direct kernel (single-rank GPU fast path)")
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--channels", type=int, default=2)
     ap.add_argument("--capacity-factor", type=float, default=3.0)
@@ -107,12 +109,17 @@ def main():
         for i in range(min(args.pipeline, total_steps)):
             issue(i)
     else:
+        use_fused = (not args.no_fused) and world == 1 and is_cuda
+
         def run_step(i, sync_loss=False):
             if not args.no_intent:
                 model.signal_intent(batches[i + args.lookahead],
                                     worker.current_clock() + args.lookahead,
                                     worker.current_clock() + args.lookahead + 2)
-            loss = model.train_batch(batches[i], sync_loss=sync_loss)
+            if use_fused:
+                loss = model.train_batch_fused(batches[i], sync_loss=sync_loss)
+            else:
+                loss = model.train_batch(batches[i], sync_loss=sync_loss)
             worker.advance_clock()
             return loss
 

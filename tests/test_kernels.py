@@ -290,3 +290,59 @@ def test_rescal_model_cpu():
     assert losses[-1] < losses[0], losses[:3] + losses[-3:]
     worker.finalize()
     server.shutdown()
+
+
+@pytest.mark.gpu
+def test_kge_fused_matches_classic():
+    """Fused slab-direct step == classic pull/kernel/push on UNIQUE keys
+    (duplicates are hogwild-updated by design)."""
+    import adapm_amd
+
+    adapm_amd._SETUP.clear()
+    adapm_amd.runtime._RUNTIME = None
+    E, R, D, N, B = 500, 20, 64, 3, 32
+    adapm_amd.setup(num_keys=E + R, num_threads=1, device="cuda:0")
+    server = adapm_amd.Server(2 * D)
+    worker = adapm_amd.Worker(0, server)
+    g = torch.Generator().manual_seed(0)
+    init = torch.cat([torch.randn(E + R, D, generator=g) * 0.2,
+                      torch.rand(E + R, D, generator=g) * 0.1], dim=1)
+    worker.set(np.arange(E + R, dtype=np.int64), init.cuda())
+
+    rng = np.random.default_rng(1)
+    s_k = rng.choice(E, B, replace=False).astype(np.int64)
+    r_k = (E + rng.choice(R, B, replace=True)).astype(np.int64)  # rels may repeat
+    o_k = rng.choice(np.setdiff1d(np.arange(E), s_k), B, replace=False).astype(np.int64)
+    neg_pool = np.setdiff1d(np.arange(E), np.concatenate([s_k, o_k]))
+    n_k = rng.choice(neg_pool, B * N, replace=False).astype(np.int64)
+    # make relations unique too so fused == classic exactly
+    r_k = (E + rng.choice(R, min(B, R), replace=False))[:B].astype(np.int64)
+    if len(r_k) < B:
+        pytest.skip("not enough unique relations")
+
+    # classic path on a snapshot
+    all_keys = np.concatenate([s_k, r_k, o_k, n_k])
+    rows = torch.empty(len(all_keys), 2 * D, device="cuda")
+    worker.pull(all_keys, rows)
+    sv, rv, ov, nv = (rows[:B], rows[B:2 * B], rows[2 * B:3 * B], rows[3 * B:])
+    from adapm_amd import _C as C3
+
+    ds, dr, do, dn = (torch.empty_like(t) for t in (sv, rv, ov, nv))
+    loss_c = torch.empty(B, device="cuda")
+    C3.kge_complex_step(sv.contiguous(), rv.contiguous(), ov.contiguous(), nv.contiguous(),
+                        ds, dr, do, dn, loss_c, N, D, 0.05, 1e-6)
+
+    # fused path mutates the store
+    loss_f = server.raw.kge_step_fused(torch.from_numpy(s_k), torch.from_numpy(r_k),
+                                       torch.from_numpy(o_k), torch.from_numpy(n_k),
+                                       N, D, 0.05, 1e-6)
+    torch.cuda.synchronize()
+    assert torch.allclose(loss_f, loss_c, atol=1e-4)
+
+    after = torch.empty(len(all_keys), 2 * D, device="cuda")
+    worker.pull(all_keys, after)
+    torch.cuda.synchronize()
+    expected = rows + torch.cat([ds, dr, do, dn])
+    err = (after - expected).abs().max().item()
+    assert err < 1e-4, f"fused store update mismatch: {err}"
+    server.shutdown()
