@@ -4,6 +4,7 @@
 import os
 
 import numpy as np
+import pytest
 import torch
 
 from dist_helper import run_dist
@@ -100,4 +101,20 @@ def test_techniques_aliases():
     assert adapm_amd._SETUP["techniques"] == adapm_amd.TECH_REPLICATION_ONLY
     adapm_amd.setup(num_keys=4, num_threads=1, use_techniques="ALL")
     assert adapm_amd._SETUP["techniques"] == adapm_amd.TECH_ALL
+    adapm_amd._SETUP.clear()
+
+
+def test_fused_step_guards_cpu():
+    """kge_step_fused must refuse a CPU store (it is the single-rank GPU
+    fast path; callers fall back to the classic pull/kernel/push path)."""
+    import adapm_amd
+
+    adapm_amd._SETUP.clear()
+    adapm_amd.runtime._RUNTIME = None
+    adapm_amd.setup(num_keys=16, num_threads=1, device="cpu")
+    s = adapm_amd.Server(8)
+    k = torch.zeros(2, dtype=torch.int64)
+    with pytest.raises(RuntimeError, match="GPU fast path"):
+        s.raw.kge_step_fused(k, k, k, torch.zeros(4, dtype=torch.int64), 2, 4, 0.1, 1e-6)
+    s.shutdown()
     adapm_amd._SETUP.clear()
