@@ -1867,6 +1867,83 @@ class Server {
     return loss;
   }
 
+  torch::Tensor w2v_step_fused(torch::Tensor keys_ctr, torch::Tensor keys_ctx,
+                               torch::Tensor keys_neg, int64_t N, int64_t D, double lr,
+                               double eps) {
+    TORCH_CHECK(world_ == 1, "w2v_step_fused requires a single rank (all keys local)");
+    TORCH_CHECK(uniform_len_ == 2 * D, "w2v_step_fused: store rows must be [emb|accum] = 2D");
+    TORCH_CHECK(layout_identity_.load(), "w2v_step_fused requires the identity layout");
+    TORCH_CHECK(dev_.is_cuda(), "w2v_step_fused is the GPU fast path");
+    for (auto* t : {&keys_ctr, &keys_ctx, &keys_neg}) check_keys(*t);
+    int64_t B = keys_ctr.numel();
+    TORCH_CHECK(keys_ctx.numel() == B && keys_neg.numel() == B * N);
+    auto rng_check = [&](const torch::Tensor& t) {
+      const int64_t* kp = t.data_ptr<int64_t>();
+      for (int64_t i = 0; i < t.numel(); ++i)
+        TORCH_CHECK((uint64_t)kp[i] < (uint64_t)num_keys_, "key out of range: ", kp[i]);
+    };
+    rng_check(keys_ctr);
+    rng_check(keys_ctx);
+    rng_check(keys_neg);
+    auto loss = torch::empty({B}, torch::TensorOptions().dtype(torch::kFloat32).device(dev_));
+    auto kc = keys_ctr.to(dev_, true);
+    auto kx = keys_ctx.to(dev_, true);
+    auto kn = keys_neg.to(dev_, true);
+    {
+      InflightGuard g(this);
+      w2v_sgns_step_fused_gpu(slab_.data, kc.data_ptr<int64_t>(), kx.data_ptr<int64_t>(),
+                              kn.data_ptr<int64_t>(), loss.data_ptr<float>(), (int)B, (int)N,
+                              (int)D, Slab::padded(uniform_len_), world_, (float)lr, (float)eps,
+                              current_stream(dev_));
+    }
+    int64_t total = 2 * B + B * N;
+    stat_pull_keys_ += total;
+    stat_pull_local_ += total;
+    stat_push_keys_ += total;
+    stat_push_local_ += total;
+    stat_pulls_ += 1;
+    stat_pushes_ += 1;
+    return loss;
+  }
+
+  torch::Tensor mf_step_fused(torch::Tensor keys_w, torch::Tensor keys_h, torch::Tensor x,
+                              int64_t R, double lr, double lambda, double eps) {
+    TORCH_CHECK(world_ == 1, "mf_step_fused requires a single rank (all keys local)");
+    TORCH_CHECK(uniform_len_ == 2 * R, "mf_step_fused: store rows must be [emb|accum] = 2R");
+    TORCH_CHECK(layout_identity_.load(), "mf_step_fused requires the identity layout");
+    TORCH_CHECK(dev_.is_cuda(), "mf_step_fused is the GPU fast path");
+    for (auto* t : {&keys_w, &keys_h}) check_keys(*t);
+    int64_t B = keys_w.numel();
+    TORCH_CHECK(keys_h.numel() == B && x.numel() == B);
+    TORCH_CHECK(x.scalar_type() == torch::kFloat32, "ratings must be float32");
+    auto rng_check = [&](const torch::Tensor& t) {
+      const int64_t* kp = t.data_ptr<int64_t>();
+      for (int64_t i = 0; i < t.numel(); ++i)
+        TORCH_CHECK((uint64_t)kp[i] < (uint64_t)num_keys_, "key out of range: ", kp[i]);
+    };
+    rng_check(keys_w);
+    rng_check(keys_h);
+    auto loss = torch::empty({B}, torch::TensorOptions().dtype(torch::kFloat32).device(dev_));
+    auto kw = keys_w.to(dev_, true);
+    auto kh = keys_h.to(dev_, true);
+    auto xd = x.contiguous().to(dev_, true);
+    {
+      InflightGuard g(this);
+      mf_update_step_fused_gpu(slab_.data, kw.data_ptr<int64_t>(), kh.data_ptr<int64_t>(),
+                               xd.data_ptr<float>(), loss.data_ptr<float>(), (int)B, (int)R,
+                               Slab::padded(uniform_len_), world_, (float)lr, (float)lambda,
+                               (float)eps, current_stream(dev_));
+    }
+    int64_t total = 2 * B;
+    stat_pull_keys_ += total;
+    stat_pull_local_ += total;
+    stat_push_keys_ += total;
+    stat_push_local_ += total;
+    stat_pulls_ += 1;
+    stat_pushes_ += 1;
+    return loss;
+  }
+
   // ------------------------------------------------ sampling support
 
   // "Local" sampling scheme scan: per candidate, scan upward (wrapping in
@@ -2224,6 +2301,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("failed_reason", &Server::failed_reason)
       .def("scan_local", &Server::scan_local, py::call_guard<py::gil_scoped_release>())
       .def("kge_step_fused", &Server::kge_step_fused, py::call_guard<py::gil_scoped_release>())
+      .def("w2v_step_fused", &Server::w2v_step_fused, py::call_guard<py::gil_scoped_release>())
+      .def("mf_step_fused", &Server::mf_step_fused, py::call_guard<py::gil_scoped_release>())
       .def("get_len", &Server::get_len)
       .def("num_keys", &Server::num_keys)
       .def("rank", &Server::rank)

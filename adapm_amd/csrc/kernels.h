@@ -40,6 +40,17 @@ void kge_complex_step_fused_gpu(float* slab, const int64_t* keys_s, const int64_
                                 int B, int N, int D, int32_t plen, int world, int rank,
                                 float lr, float eps, void* stream);
 
+// FUSED SGNS step (same contract as kge_complex_step_fused_gpu: slab-
+// direct reads + atomic AdaGrad writes, hogwild on duplicates).
+void w2v_sgns_step_fused_gpu(float* slab, const int64_t* keys_ctr, const int64_t* keys_ctx,
+                             const int64_t* keys_neg, float* loss, int B, int N, int D,
+                             int32_t plen, int world, float lr, float eps, void* stream);
+
+// FUSED MF step (slab-direct NZSL+L2; same contract as above).
+void mf_update_step_fused_gpu(float* slab, const int64_t* keys_w, const int64_t* keys_h,
+                              const float* x, float* loss, int B, int R, int32_t plen,
+                              int world, float lr, float lambda, float eps, void* stream);
+
 // ComplEx scoring only (evaluation): score[b][e] = psi(s_b, r_b, cand_e)
 //  cand: [E][2D] candidate entity rows; scores: [B][E]
 void kge_complex_score_gpu(const float* s, const float* r, const float* cand, float* scores,

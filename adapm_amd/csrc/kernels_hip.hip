@@ -538,3 +538,117 @@ void kge_complex_step_fused_gpu(float* slab, const int64_t* keys_s, const int64_
 }
 
 }  // namespace adapm
+
+namespace adapm {
+
+// Fused SGNS: mirror of k_w2v_step reading rows straight from the slab
+// at identity offsets and atomically applying the AdaGrad updates.
+template <int KPT>
+__global__ void k_w2v_step_fused(float* __restrict__ slab, const int64_t* __restrict__ keys_ctr,
+                                 const int64_t* __restrict__ keys_ctx,
+                                 const int64_t* __restrict__ keys_neg, float* __restrict__ loss,
+                                 int B, int N, int D, int32_t plen, int world, float lr,
+                                 float eps) {
+  __shared__ float lds[KT / 64];
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    float* cb = slab + (keys_ctr[b] / world) * (int64_t)plen;
+    float c_emb[KPT], a_c[KPT];
+#pragma unroll
+    for (int i = 0; i < KPT; ++i) {
+      int k = threadIdx.x + i * KT;
+      c_emb[i] = k < D ? cb[k] : 0.f;
+      a_c[i] = 0.f;
+    }
+    float lsum = 0.f;
+    for (int j = 0; j <= N; ++j) {
+      int64_t xk = (j == 0) ? keys_ctx[b] : keys_neg[(int64_t)b * N + (j - 1)];
+      float* xb = slab + (xk / world) * (int64_t)plen;
+      float y = (j == 0) ? 1.f : -1.f;
+      float part = 0.f;
+      float x_emb[KPT];
+#pragma unroll
+      for (int i = 0; i < KPT; ++i) {
+        int k = threadIdx.x + i * KT;
+        x_emb[i] = k < D ? xb[k] : 0.f;
+        part += c_emb[i] * x_emb[i];
+      }
+      float dot = block_reduce_sum(part, lds);
+      float g = -y * sigmoidf(-y * dot);
+      if (threadIdx.x == 0) lsum += softplusf(-y * dot);
+#pragma unroll
+      for (int i = 0; i < KPT; ++i) {
+        int k = threadIdx.x + i * KT;
+        if (k >= D) continue;
+        a_c[i] += g * x_emb[i];
+        float gx = g * c_emb[i];
+        float G = xb[D + k] + gx * gx;
+        atomicAdd(&xb[k], -lr * gx * __frsqrt_rn(G + eps));
+        atomicAdd(&xb[D + k], gx * gx);
+      }
+    }
+#pragma unroll
+    for (int i = 0; i < KPT; ++i) {
+      int k = threadIdx.x + i * KT;
+      if (k >= D) continue;
+      float G = cb[D + k] + a_c[i] * a_c[i];
+      atomicAdd(&cb[k], -lr * a_c[i] * __frsqrt_rn(G + eps));
+      atomicAdd(&cb[D + k], a_c[i] * a_c[i]);
+    }
+    if (threadIdx.x == 0) loss[b] = lsum;
+  }
+}
+
+// Fused MF: w/h row values are read into registers before any atomic
+// write in the same lane, so the NZSL+L2 gradients use a consistent
+// pre-update snapshot per nonzero (duplicates across the batch hogwild).
+__global__ void k_mf_step_fused(float* __restrict__ slab, const int64_t* __restrict__ keys_w,
+                                const int64_t* __restrict__ keys_h, const float* __restrict__ x,
+                                float* __restrict__ loss, int B, int R, int32_t plen, int world,
+                                float lr, float lambda, float eps) {
+  __shared__ float lds[KT / 64];
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    float* wb = slab + (keys_w[b] / world) * (int64_t)plen;
+    float* hb = slab + (keys_h[b] / world) * (int64_t)plen;
+    float part = 0.f;
+    for (int k = threadIdx.x; k < R; k += KT) part += wb[k] * hb[k];
+    float pred = block_reduce_sum(part, lds);
+    float e = x[b] - pred;
+    if (threadIdx.x == 0) loss[b] = e * e;
+    for (int k = threadIdx.x; k < R; k += KT) {
+      float wv = wb[k], hv = hb[k];
+      float gw = -2.f * e * hv + 2.f * lambda * wv;
+      float gh = -2.f * e * wv + 2.f * lambda * hv;
+      float Gw = wb[R + k] + gw * gw;
+      float Gh = hb[R + k] + gh * gh;
+      atomicAdd(&wb[k], -lr * gw * __frsqrt_rn(Gw + eps));
+      atomicAdd(&wb[R + k], gw * gw);
+      atomicAdd(&hb[k], -lr * gh * __frsqrt_rn(Gh + eps));
+      atomicAdd(&hb[R + k], gh * gh);
+    }
+  }
+}
+
+void w2v_sgns_step_fused_gpu(float* slab, const int64_t* keys_ctr, const int64_t* keys_ctx,
+                             const int64_t* keys_neg, float* loss, int B, int N, int D,
+                             int32_t plen, int world, float lr, float eps, void* stream) {
+  dim3 g((unsigned)std::min<int64_t>(B < 1 ? 1 : B, 16384)), t(KT);
+  auto st = (hipStream_t)stream;
+#define LAUNCHW(KPT) \
+  hipLaunchKernelGGL(k_w2v_step_fused<KPT>, g, t, 0, st, slab, keys_ctr, keys_ctx, keys_neg, \
+                     loss, B, N, D, plen, world, lr, eps)
+  if (D <= KT) LAUNCHW(1);
+  else if (D <= 2 * KT) LAUNCHW(2);
+  else if (D <= 4 * KT) LAUNCHW(4);
+  else LAUNCHW(8);
+#undef LAUNCHW
+}
+
+void mf_update_step_fused_gpu(float* slab, const int64_t* keys_w, const int64_t* keys_h,
+                              const float* x, float* loss, int B, int R, int32_t plen,
+                              int world, float lr, float lambda, float eps, void* stream) {
+  hipLaunchKernelGGL(k_mf_step_fused, dim3((unsigned)std::min<int64_t>(B < 1 ? 1 : B, 16384)),
+                     dim3(KT), 0, (hipStream_t)stream, slab, keys_w, keys_h, x, loss, B, R,
+                     plen, world, lr, lambda, eps);
+}
+
+}  // namespace adapm

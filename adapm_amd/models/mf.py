@@ -97,6 +97,18 @@ class MF:
             w.wait(self._pending.pop(0))
         return float(loss.mean().item()) if sync_loss else loss
 
+    def train_batch_fused(self, rows, cols, ratings, sync_loss=False):
+        """Single-rank GPU fast path: slab-direct fused kernel
+        (Server.mf_step_fused); falls back to train_batch otherwise."""
+        cfg = self.cfg
+        if self.world != 1 or self.dev.type != "cuda":
+            return self.train_batch(rows, cols, ratings, sync_loss=sync_loss)
+        k_w = torch.from_numpy(np.asarray(rows, dtype=np.int64))
+        k_h = torch.from_numpy(self.col_key(cols))
+        x = torch.from_numpy(np.asarray(ratings, dtype=np.float32))
+        loss = self.server.raw.mf_step_fused(k_w, k_h, x, cfg.rank, cfg.lr, cfg.lam, cfg.eps)
+        return float(loss.mean().item()) if sync_loss else loss
+
     def drain(self):
         for t in self._pending:
             self.worker.wait(t)

@@ -144,6 +144,26 @@ class Word2Vec:
             w.wait(self._pending.pop(0))
         return float(loss.mean().item()) if sync_loss else loss
 
+    def train_pairs_fused(self, ctr_words, ctx_words, sync_loss=False):
+        """Single-rank GPU fast path: slab-direct fused SGNS kernel
+        (Server.w2v_step_fused); falls back to train_pairs otherwise."""
+        cfg = self.cfg
+        w = self.worker
+        if self.world != 1 or self.dev.type != "cuda":
+            return self.train_pairs(ctr_words, ctx_words, sync_loss=sync_loss)
+        B = len(ctr_words)
+        if self.server.sampling is not None:
+            sid = w.prepare_sample(B * cfg.negative, w.current_clock(), w.current_clock() + 2)
+            neg_words = self.server.sampling.pull(w, sid, B * cfg.negative)
+            w.finish_sample(sid)
+        else:
+            neg_words = self.rng.integers(0, cfg.vocab_size, size=B * cfg.negative)
+        loss = self.server.raw.w2v_step_fused(
+            torch.from_numpy(syn0(ctr_words)), torch.from_numpy(syn1(ctx_words)),
+            torch.from_numpy(syn1(np.ascontiguousarray(neg_words, dtype=np.int64))),
+            cfg.negative, cfg.dim, cfg.lr, cfg.eps)
+        return float(loss.mean().item()) if sync_loss else loss
+
     def drain(self):
         for t in self._pending:
             self.worker.wait(t)

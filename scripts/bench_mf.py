@@ -23,6 +23,8 @@ def main():
     ap.add_argument("--rank", type=int, default=128)
     ap.add_argument("--batch-nnz", type=int, default=131072)
     ap.add_argument("--device", type=str, default=None)
+    ap.add_argument("--no-fused", action="store_true",
+                    help="disable the fused slab-direct kernel (single-rank GPU fast path)")
     args = ap.parse_args()
 
     rank_id = int(os.environ.get("RANK", "0"))
@@ -47,8 +49,10 @@ def main():
                 rng.normal(size=args.batch_nnz).astype(np.float32)) for _ in range(total)]
 
     is_cuda = server.rt.device.type == "cuda"
+    use_fused = (not args.no_fused) and world == 1 and is_cuda
+    step_fn = model.train_batch_fused if use_fused else model.train_batch
     for i in range(args.warmup):
-        model.train_batch(*batches[i])
+        step_fn(*batches[i])
     model.drain()
     worker.barrier()
     if is_cuda:
@@ -60,7 +64,7 @@ def main():
             r2, c2, _ = batches[i + 1]
             worker.intent(np.concatenate([r2.astype(np.int64), model.col_key(c2)]),
                           worker.current_clock() + 1, worker.current_clock() + 3)
-        model.train_batch(*batches[i])
+        step_fn(*batches[i])
         worker.advance_clock()
     model.drain()
     if is_cuda:

@@ -24,6 +24,8 @@ def main():
     ap.add_argument("--negative", type=int, default=5)
     ap.add_argument("--pairs", type=int, default=65536, help="pairs per step")
     ap.add_argument("--device", type=str, default=None)
+    ap.add_argument("--no-fused", action="store_true",
+                    help="disable the fused slab-direct kernel (single-rank GPU fast path)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -59,15 +61,17 @@ def main():
                for _ in range(total)]
 
     is_cuda = server.rt.device.type == "cuda"
+    use_fused = (not args.no_fused) and world == 1 and is_cuda
+    step_fn = model.train_pairs_fused if use_fused else model.train_pairs
     for i in range(args.warmup):
-        model.train_pairs(*batches[i])
+        step_fn(*batches[i])
     model.drain()
     worker.barrier()
     if is_cuda:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     for i in range(args.warmup, total):
-        model.train_pairs(*batches[i])
+        step_fn(*batches[i])
     model.drain()
     if is_cuda:
         torch.cuda.synchronize()

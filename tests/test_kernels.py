@@ -344,3 +344,93 @@ def test_kge_fused_matches_classic():
     err = (after - expected).abs().max().item()
     assert err < 1e-4, f"fused store update mismatch: {err}"
     server.shutdown()
+
+
+@pytest.mark.gpu
+def test_w2v_fused_matches_classic():
+    import adapm_amd
+
+    adapm_amd._SETUP.clear()
+    adapm_amd.runtime._RUNTIME = None
+    V, D, N, B = 600, 48, 3, 24
+    adapm_amd.setup(num_keys=2 * V, num_threads=1, device="cuda:0")
+    server = adapm_amd.Server(2 * D)
+    worker = adapm_amd.Worker(0, server)
+    g = torch.Generator().manual_seed(3)
+    init = torch.cat([torch.randn(2 * V, D, generator=g) * 0.2,
+                      torch.rand(2 * V, D, generator=g) * 0.1], dim=1)
+    worker.set(np.arange(2 * V, dtype=np.int64), init.cuda())
+
+    rng = np.random.default_rng(4)
+    # all keys unique (ctr on even keys, ctx+neg on odd keys)
+    ctr_k = 2 * rng.choice(V, B, replace=False).astype(np.int64)
+    rest = rng.choice(V, B + B * N, replace=False).astype(np.int64)
+    ctx_k = 2 * rest[:B] + 1
+    neg_k = 2 * rest[B:] + 1
+
+    all_keys = np.concatenate([ctr_k, ctx_k, neg_k])
+    rows = torch.empty(len(all_keys), 2 * D, device="cuda")
+    worker.pull(all_keys, rows)
+    cv, xv, nv = rows[:B], rows[B:2 * B], rows[2 * B:]
+    from adapm_amd import _C as C
+
+    dc, dx, dn = (torch.empty_like(t) for t in (cv, xv, nv))
+    loss_c = torch.empty(B, device="cuda")
+    C.w2v_sgns_step(cv.contiguous(), xv.contiguous(), nv.contiguous(),
+                    dc, dx, dn, loss_c, N, D, 0.05, 1e-6)
+
+    loss_f = server.raw.w2v_step_fused(torch.from_numpy(ctr_k), torch.from_numpy(ctx_k),
+                                       torch.from_numpy(neg_k), N, D, 0.05, 1e-6)
+    torch.cuda.synchronize()
+    assert torch.allclose(loss_f, loss_c, atol=1e-4)
+    after = torch.empty(len(all_keys), 2 * D, device="cuda")
+    worker.pull(all_keys, after)
+    torch.cuda.synchronize()
+    expected = rows + torch.cat([dc, dx, dn])
+    err = (after - expected).abs().max().item()
+    assert err < 1e-4, f"fused w2v store update mismatch: {err}"
+    server.shutdown()
+
+
+@pytest.mark.gpu
+def test_mf_fused_matches_classic():
+    import adapm_amd
+
+    adapm_amd._SETUP.clear()
+    adapm_amd.runtime._RUNTIME = None
+    NR, NC, R, B = 400, 300, 32, 48
+    adapm_amd.setup(num_keys=NR + NC, num_threads=1, device="cuda:0")
+    server = adapm_amd.Server(2 * R)
+    worker = adapm_amd.Worker(0, server)
+    g = torch.Generator().manual_seed(5)
+    init = torch.cat([torch.rand(NR + NC, R, generator=g) * 0.1,
+                      torch.full((NR + NC, R), 1.0)], dim=1)
+    worker.set(np.arange(NR + NC, dtype=np.int64), init.cuda())
+
+    rng = np.random.default_rng(6)
+    w_k = rng.choice(NR, B, replace=False).astype(np.int64)
+    h_k = (NR + rng.choice(NC, B, replace=False)).astype(np.int64)
+    x = rng.normal(size=B).astype(np.float32)
+
+    all_keys = np.concatenate([w_k, h_k])
+    rows = torch.empty(2 * B, 2 * R, device="cuda")
+    worker.pull(all_keys, rows)
+    wv, hv = rows[:B], rows[B:]
+    from adapm_amd import _C as C
+
+    dw, dh = torch.empty_like(wv), torch.empty_like(hv)
+    loss_c = torch.empty(B, device="cuda")
+    C.mf_update_step(wv.contiguous(), hv.contiguous(),
+                     torch.from_numpy(x).cuda(), dw, dh, loss_c, R, 0.05, 0.01, 1e-6)
+
+    loss_f = server.raw.mf_step_fused(torch.from_numpy(w_k), torch.from_numpy(h_k),
+                                      torch.from_numpy(x), R, 0.05, 0.01, 1e-6)
+    torch.cuda.synchronize()
+    assert torch.allclose(loss_f, loss_c, atol=1e-4)
+    after = torch.empty(2 * B, 2 * R, device="cuda")
+    worker.pull(all_keys, after)
+    torch.cuda.synchronize()
+    expected = rows + torch.cat([dw, dh])
+    err = (after - expected).abs().max().item()
+    assert err < 1e-4, f"fused mf store update mismatch: {err}"
+    server.shutdown()
