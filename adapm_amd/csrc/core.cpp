@@ -286,7 +286,13 @@ class Server {
     int64_t cap = (int64_t)((double)owned_floats * capacity_factor) + (1 << 20);
     if (device_cap_floats > 0) cap = device_cap_floats;  // explicit HBM budget
     slab_.init(cap, dev_);
-    if (host_spill_floats > 0) slab_.init_host(host_spill_floats);
+    if (host_spill_floats > 0) {
+      slab_.init_host(host_spill_floats);
+      // tiered store: track per-key access heat so rebalance_spill can
+      // keep the hot set HBM-resident (HBM as a cache over pinned host)
+      heat_.reset(new std::atomic<uint32_t>[num_keys_]);
+      for (Key k = 0; k < num_keys_; ++k) heat_[k].store(0, std::memory_order_relaxed);
+    }
     // if the initial allocation will not fit the device arena, identity
     // breaks immediately (spilled slots are not at identity offsets)
     if (owned_floats > cap) layout_identity_.store(false);
@@ -327,7 +333,15 @@ class Server {
   struct InflightGuard {
     Server* s;
     explicit InflightGuard(Server* sv) : s(sv) {
-      s->inflight_.fetch_add(1, std::memory_order_acquire);
+      // increment-then-check: if a spill rebalance is in progress, back
+      // out and wait (it quiesces after raising migrating_, so a guard
+      // acquired before the raise is waited for; one after waits here).
+      for (;;) {
+        s->inflight_.fetch_add(1, std::memory_order_acquire);
+        if (!s->migrating_.load(std::memory_order_acquire)) break;
+        s->inflight_.fetch_sub(1, std::memory_order_release);
+        std::this_thread::yield();
+      }
     }
     ~InflightGuard() { s->inflight_.fetch_sub(1, std::memory_order_release); }
   };
@@ -553,6 +567,7 @@ class Server {
               if ((f & F_PRESENT) && !(f & F_STUB)) {
                 P.local.add(loc_[k].load(std::memory_order_acquire), out_off[i], out_len[i]);
                 if (!(f & F_OWNER)) P.n_repl++;
+                if (heat_) heat_[k].fetch_add(1, std::memory_order_relaxed);
                 if (locality_stats_) {
                   key_accesses_[k].fetch_add(1, std::memory_order_relaxed);
                   key_local_[k].fetch_add(1, std::memory_order_relaxed);
@@ -697,6 +712,7 @@ class Server {
                 (set_mode ? P.assign : P.merge).add(loc_[k].load(std::memory_order_acquire),
                                                     offs[i], l);
                 version_[k].fetch_add(1, std::memory_order_relaxed);
+                if (heat_) heat_[k].fetch_add(1, std::memory_order_relaxed);
               } else if ((f & F_PRESENT) && !set_mode) {
                 // replica/stub: merge locally, flush at next sync round
                 P.merge.add(loc_[k].load(std::memory_order_acquire), offs[i], l);
@@ -1944,6 +1960,132 @@ class Server {
     return loss;
   }
 
+  // --------------------------------------------- spill-tier rebalance
+
+  // Which tier a key's value lives in: 0 = device (HBM), 1 = host-spill,
+  // -1 = not locally present.
+  int key_tier(int64_t k) {
+    TORCH_CHECK((uint64_t)k < (uint64_t)num_keys_, "key out of range: ", k);
+    uint8_t f = flags_[k].load(std::memory_order_acquire);
+    if (!(f & F_PRESENT) || (f & F_STUB)) return -1;
+    return (loc_[k].load(std::memory_order_acquire) & SPILL_BIT) ? 1 : 0;
+  }
+
+  // Swap the hottest host-spilled rows with the coldest HBM-resident
+  // rows ("HBM as a cache over pinned host memory" — the tier the
+  // reference cannot have: its store is all host DRAM). Heat = access
+  // count since the last rebalance (halved each call, so it is an EWMA).
+  // Stop-the-world for the swap batch: migrating_ gates InflightGuard
+  // entry, quiesce() drains ops that already read metadata, then the
+  // loc_ swaps + the staging kernels are enqueued before workers resume —
+  // single-stream order makes the swap invisible to them. Currently
+  // world==1 only (sync threads are not gated; at world==1 they carry
+  // no traffic).
+  int64_t rebalance_spill(int64_t max_moves = 4096) {
+    TORCH_CHECK(world_ == 1, "rebalance_spill currently requires world==1");
+    TORCH_CHECK(uniform_len_ >= 0, "rebalance_spill requires a uniform-length store");
+    if (!heat_ || slab_.host_capacity == 0 || max_moves <= 0) return 0;
+    const int32_t l = uniform_len_;
+
+    struct Cand { Key k; uint32_t h; };
+    // collect top-heat spilled keys and bottom-heat device keys with
+    // bounded per-chunk partial selections (full sort of 100M keys would
+    // dwarf the swap itself); heat is halved in the same pass.
+    constexpr int64_t G = 1 << 20;
+    int64_t nchunks = (num_keys_ + G - 1) / G;
+    std::vector<std::vector<Cand>> hot_parts(nchunks), cold_parts(nchunks);
+    at::parallel_for(0, nchunks, 1, [&](int64_t c0, int64_t c1) {
+      for (int64_t c = c0; c < c1; ++c) {
+        auto& hot = hot_parts[c];
+        auto& cold = cold_parts[c];
+        Key e = (Key)std::min<int64_t>(num_keys_, (c + 1) * G);
+        for (Key k = (Key)(c * G); k < e; ++k) {
+          uint32_t h = heat_[k].load(std::memory_order_relaxed);
+          if (h) heat_[k].store(h >> 1, std::memory_order_relaxed);
+          uint8_t f = flags_[k].load(std::memory_order_acquire);
+          if (!(f & F_PRESENT) || (f & F_STUB)) continue;
+          bool spilled = (loc_[k].load(std::memory_order_acquire) & SPILL_BIT) != 0;
+          if (spilled) {
+            if (h == 0) continue;
+            hot.push_back({k, h});
+            if ((int64_t)hot.size() >= 2 * max_moves) {
+              std::nth_element(hot.begin(), hot.begin() + max_moves - 1, hot.end(),
+                               [](const Cand& a, const Cand& b) { return a.h > b.h; });
+              hot.resize(max_moves);
+            }
+          } else {
+            cold.push_back({k, h});
+            if ((int64_t)cold.size() >= 2 * max_moves) {
+              std::nth_element(cold.begin(), cold.begin() + max_moves - 1, cold.end(),
+                               [](const Cand& a, const Cand& b) { return a.h < b.h; });
+              cold.resize(max_moves);
+            }
+          }
+        }
+      }
+    });
+    std::vector<Cand> hot, cold;
+    for (auto& p : hot_parts) hot.insert(hot.end(), p.begin(), p.end());
+    for (auto& p : cold_parts) cold.insert(cold.end(), p.begin(), p.end());
+    if (hot.empty()) return 0;
+    auto hotter = [](const Cand& a, const Cand& b) { return a.h > b.h; };
+    auto colder = [](const Cand& a, const Cand& b) { return a.h < b.h; };
+    if ((int64_t)hot.size() > max_moves) {
+      std::nth_element(hot.begin(), hot.begin() + max_moves - 1, hot.end(), hotter);
+      hot.resize(max_moves);
+    }
+    if ((int64_t)cold.size() > max_moves) {
+      std::nth_element(cold.begin(), cold.begin() + max_moves - 1, cold.end(), colder);
+      cold.resize(max_moves);
+    }
+    std::sort(hot.begin(), hot.end(), hotter);
+    std::sort(cold.begin(), cold.end(), colder);
+
+    // stop the world
+    migrating_.store(1, std::memory_order_release);
+    quiesce();
+    int64_t moves = 0;
+    HostBatch from, to;
+    int64_t pos = 0;
+    size_t ci = 0;
+    for (auto& hc : hot) {
+      if (moves >= max_moves || ci >= cold.size()) break;
+      // hysteresis: a swap must be clearly profitable or keys ping-pong
+      if ((int64_t)hc.h <= 2 * (int64_t)cold[ci].h + 1) break;
+      Key ks = hc.k, kd = cold[ci].k;
+      // re-validate (metadata may have changed since the lock-free scan)
+      uint8_t fs = flags_[ks].load(std::memory_order_acquire);
+      uint8_t fd = flags_[kd].load(std::memory_order_acquire);
+      if (!(fs & F_PRESENT) || (fs & F_STUB) || !(fd & F_PRESENT) || (fd & F_STUB)) {
+        ci++;
+        continue;
+      }
+      int64_t off_s = loc_[ks].load(std::memory_order_acquire);
+      int64_t off_d = loc_[kd].load(std::memory_order_acquire);
+      if (!(off_s & SPILL_BIT) || (off_d & SPILL_BIT)) {
+        ci++;
+        continue;
+      }
+      from.add(off_s, pos, l);
+      from.add(off_d, pos + l, l);
+      to.add(off_d, pos, l);
+      to.add(off_s, pos + l, l);
+      loc_[ks].store(off_d, std::memory_order_release);
+      loc_[kd].store(off_s, std::memory_order_release);
+      pos += 2 * (int64_t)l;
+      ci++;
+      moves++;
+    }
+    if (moves > 0) {
+      auto tmp = torch::empty({pos}, torch::TensorOptions().dtype(torch::kFloat32).device(dev_));
+      run_gather(from, tmp);
+      run_scatter(to, tmp, /*set=*/true);
+    }
+    migrating_.store(0, std::memory_order_release);
+    stat_spill_moves_ += moves;
+    return moves;
+  }
+
   // ------------------------------------------------ sampling support
 
   // "Local" sampling scheme scan: per candidate, scan upward (wrapping in
@@ -2060,6 +2202,7 @@ class Server {
     d["slab_in_use"] = slab_.in_use.load();
     d["slab_capacity"] = slab_.capacity;
     d["host_spill_in_use"] = slab_.host_in_use.load();
+    d["spill_rebalance_moves"] = stat_spill_moves_.load();
     d["host_spill_capacity"] = slab_.host_capacity;
     int64_t rounds = 0;
     for (auto& c : channels_) rounds += c.rounds.load();
@@ -2126,6 +2269,9 @@ class Server {
   std::atomic<bool> layout_identity_{true};
 
   std::atomic<int> inflight_{0};
+  std::atomic<int> migrating_{0};                       // spill-rebalance stop-the-world gate
+  std::unique_ptr<std::atomic<uint32_t>[]> heat_;       // per-key access heat (spill stores)
+  std::atomic<int64_t> stat_spill_moves_{0};
   std::atomic<int64_t> next_ts_{1};
  public:
   // env ADAPM_CPP_TIMING=1: nanosecond accounting of the worker-op host path
@@ -2303,6 +2449,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("kge_step_fused", &Server::kge_step_fused, py::call_guard<py::gil_scoped_release>())
       .def("w2v_step_fused", &Server::w2v_step_fused, py::call_guard<py::gil_scoped_release>())
       .def("mf_step_fused", &Server::mf_step_fused, py::call_guard<py::gil_scoped_release>())
+      .def("rebalance_spill", &Server::rebalance_spill, py::arg("max_moves") = 4096,
+           py::call_guard<py::gil_scoped_release>())
+      .def("key_tier", &Server::key_tier)
       .def("get_len", &Server::get_len)
       .def("num_keys", &Server::num_keys)
       .def("rank", &Server::rank)

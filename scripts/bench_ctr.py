@@ -27,6 +27,14 @@ def main():
     ap.add_argument("--device-cap-gb", type=float, default=0.0,
                     help=">0: cap HBM arena; the rest spills to pinned host")
     ap.add_argument("--host-spill-gb", type=float, default=0.0)
+    ap.add_argument("--hash-ids", action="store_true",
+                    help="hash feature ids so hotness is uncorrelated with the "
+                         "allocation order (realistic: ids are hashes) — otherwise "
+                         "the Zipf-hot low ids happen to allocate into HBM first")
+    ap.add_argument("--rebalance-every", type=int, default=0,
+                    help=">0: call rebalance_spill every N steps (tiered store "
+                         "keeps the hot set HBM-resident; world==1 only)")
+    ap.add_argument("--rebalance-moves", type=int, default=131072)
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -51,9 +59,17 @@ def main():
     total = args.warmup + args.steps
     batches = [make_synthetic_ctr(args.batch, args.features, args.fields, seed=1000 * rank + i)
                for i in range(total)]
+    if args.hash_ids:
+        batches = [((f * 2654435761) % args.features, y) for f, y in batches]
+    do_rebalance = args.rebalance_every > 0 and world == 1
+
+    def maybe_rebalance(i):
+        if do_rebalance and i % args.rebalance_every == 0:
+            server.raw.rebalance_spill(args.rebalance_moves)
 
     is_cuda = server.rt.device.type == "cuda"
     for i in range(args.warmup):
+        maybe_rebalance(i)
         model.train_batch(*batches[i], sync_loss=False)
     model.drain()
     worker.barrier()
@@ -64,6 +80,7 @@ def main():
         if i + 1 < total:
             model.signal_intent(batches[i + 1][0], worker.current_clock() + 1,
                                 worker.current_clock() + 3)
+        maybe_rebalance(i)
         model.train_batch(*batches[i], sync_loss=False)
         worker.advance_clock()
     model.drain()
@@ -87,6 +104,8 @@ def main():
                        "global_batch": args.batch * world, "init_s": init_s,
                        "device_cap_gb": args.device_cap_gb,
                        "host_spill_in_use_gb": st["host_spill_in_use"] * 4 / (1 << 30),
+                       "spill_rebalance_moves": st.get("spill_rebalance_moves", 0),
+                       "hash_ids": args.hash_ids,
                        "pull_push_ops_per_s": 2 * args.batch * args.fields * args.steps * world / el,
                        "parallelism": f"ps-async-dp{world}"},
         }), flush=True)

@@ -44,3 +44,70 @@ def test_host_spill_cpu():
 @pytest.mark.gpu
 def test_host_spill_gpu():
     _run_spill("cuda:0")
+
+
+def _mk_tiered(device):
+    """400 keys x 64 floats; device arena sized for ~100 rows, rest spill."""
+    import adapm_amd
+
+    adapm_amd._SETUP.clear()
+    adapm_amd.runtime._RUNTIME = None
+    adapm_amd.setup(num_keys=400, num_threads=1, device=device,
+                    device_cap_gb=100 * 64 * 4 / 2**30, host_spill_gb=0.01)
+    s = adapm_amd.Server(64)
+    w = adapm_amd.Worker(0, s)
+    return adapm_amd, s, w
+
+
+def _run_rebalance(device):
+    import torch as T
+
+    adapm_amd, s, w = _mk_tiered(device)
+    keys = np.arange(400, dtype=np.int64)
+    vals = torch.arange(400, dtype=torch.float32)[:, None].expand(400, 64).contiguous()
+    if device.startswith("cuda"):
+        vals = vals.cuda()
+    w.set(keys, vals)
+    tiers = np.array([s.raw.key_tier(int(k)) for k in keys])
+    assert (tiers == 0).sum() == 100 and (tiers == 1).sum() == 300, tiers
+
+    # hammer 20 spilled keys; touch device keys once so they have heat=1
+    hot = keys[tiers == 1][:20]
+    out = torch.zeros(20, 64, device=vals.device)
+    for _ in range(30):
+        w.pull(hot, out)
+    moved = s.raw.rebalance_spill(64)
+    assert moved == 20, moved
+    for k in hot:
+        assert s.raw.key_tier(int(k)) == 0, k
+    # total device-resident count unchanged (swaps, not growth)
+    tiers2 = np.array([s.raw.key_tier(int(k)) for k in keys])
+    assert (tiers2 == 0).sum() == 100, tiers2.sum()
+
+    # ALL values must be intact after the swap
+    out_all = torch.zeros(400, 64, device=vals.device)
+    w.pull(keys, out_all)
+    if device.startswith("cuda"):
+        torch.cuda.synchronize()
+    assert torch.equal(out_all.cpu(), vals.cpu())
+
+    # pushes after the swap land in the new slots
+    w.push(hot, torch.ones(20, 64, device=vals.device))
+    w.pull(hot, out)
+    if device.startswith("cuda"):
+        torch.cuda.synchronize()
+    assert torch.equal(out.cpu(), vals.cpu()[hot] + 1)
+
+    # a second rebalance with no new skew does nothing (hysteresis)
+    assert s.raw.rebalance_spill(64) == 0
+    assert s.stats()["spill_rebalance_moves"] == 20
+    s.shutdown()
+
+
+def test_spill_rebalance_cpu():
+    _run_rebalance("cpu")
+
+
+@pytest.mark.gpu
+def test_spill_rebalance_gpu():
+    _run_rebalance("cuda:0")
