@@ -44,6 +44,7 @@
 #include <tuple>
 #include <memory>
 #include <mutex>
+#include <random>
 #include <thread>
 #include <chrono>
 #include <unordered_map>
@@ -121,6 +122,26 @@ struct Slab {
   ~Slab() { adapm_host_arena_free(host_raw, is_cuda ? 1 : 0); }
 
   SlabBases bases() const { return SlabBases{data, is_cuda ? host_dev : host_host}; }
+
+  // device-arena-only allocation attempt (spill-tier promotion): -1 if full
+  int64_t try_alloc_device(int32_t len) {
+    int32_t p = padded(len);
+    std::lock_guard<std::mutex> g(mu);
+    auto it = freelists.find(p);
+    if (it != freelists.end() && !it->second.empty()) {
+      int64_t off = it->second.back();
+      it->second.pop_back();
+      in_use += p;
+      return off;
+    }
+    if (bump + p <= capacity) {
+      int64_t off = bump;
+      bump += p;
+      in_use += p;
+      return off;
+    }
+    return -1;
+  }
 
   int64_t alloc(int32_t len) {
     int32_t p = padded(len);
@@ -482,6 +503,23 @@ class Server {
     }
   }
 
+  // spill-tier promotion candidates: remember which spilled keys were
+  // touched (bounded; rebalance_spill consumes the list — this is what
+  // makes rebalance O(touched), not O(num_keys))
+  template <class Parts>
+  void record_spill_touches(const Parts& parts) {
+    if (!heat_) return;
+    size_t add = 0;
+    for (auto& P : parts) add += P.spilled.size();
+    if (add == 0) return;
+    std::lock_guard<std::mutex> g(spill_mu_);
+    constexpr size_t CAP = 1 << 22;
+    for (auto& P : parts) {
+      if (spill_touched_.size() >= CAP) break;
+      spill_touched_.insert(spill_touched_.end(), P.spilled.begin(), P.spilled.end());
+    }
+  }
+
   // ------------------------------------------------ worker API
 
   // Pull: local fast path returns -1 with the gather already enqueued on
@@ -554,6 +592,7 @@ class Server {
         struct Part {
           HostBatch local;
           std::vector<Remote> remote;
+          std::vector<Key> spilled;
           int64_t n_repl = 0;
         };
         std::vector<Part> parts(nchunks);
@@ -565,9 +604,13 @@ class Server {
               Key k = kp[i];
               uint8_t f = flags_[k].load(std::memory_order_acquire);
               if ((f & F_PRESENT) && !(f & F_STUB)) {
-                P.local.add(loc_[k].load(std::memory_order_acquire), out_off[i], out_len[i]);
+                int64_t off = loc_[k].load(std::memory_order_acquire);
+                P.local.add(off, out_off[i], out_len[i]);
                 if (!(f & F_OWNER)) P.n_repl++;
-                if (heat_) heat_[k].fetch_add(1, std::memory_order_relaxed);
+                if (heat_) {
+                  heat_[k].fetch_add(1, std::memory_order_relaxed);
+                  if (off & SPILL_BIT) P.spilled.push_back(k);
+                }
                 if (locality_stats_) {
                   key_accesses_[k].fetch_add(1, std::memory_order_relaxed);
                   key_local_[k].fetch_add(1, std::memory_order_relaxed);
@@ -586,6 +629,7 @@ class Server {
           remote.insert(remote.end(), P.remote.begin(), P.remote.end());
           stat_pull_replica_ += P.n_repl;
         }
+        record_spill_touches(parts);
         stat_pull_local_ += n - (int64_t)remote.size();
         stat_pull_keys_ += n;
         if (cpp_timing_) { t_pass_ += tick() - tp0; tp0 = tick(); }
@@ -697,6 +741,7 @@ class Server {
         struct Part {
           HostBatch merge, assign;
           std::vector<Remote> remote;
+          std::vector<Key> spilled;
           int64_t n_repl = 0;
         };
         std::vector<Part> parts(nchunks);
@@ -709,10 +754,13 @@ class Server {
               int32_t l = len_of(k);
               uint8_t f = flags_[k].load(std::memory_order_acquire);
               if ((f & F_PRESENT) && (f & F_OWNER)) {
-                (set_mode ? P.assign : P.merge).add(loc_[k].load(std::memory_order_acquire),
-                                                    offs[i], l);
+                int64_t off = loc_[k].load(std::memory_order_acquire);
+                (set_mode ? P.assign : P.merge).add(off, offs[i], l);
                 version_[k].fetch_add(1, std::memory_order_relaxed);
-                if (heat_) heat_[k].fetch_add(1, std::memory_order_relaxed);
+                if (heat_) {
+                  heat_[k].fetch_add(1, std::memory_order_relaxed);
+                  if (off & SPILL_BIT) P.spilled.push_back(k);
+                }
               } else if ((f & F_PRESENT) && !set_mode) {
                 // replica/stub: merge locally, flush at next sync round
                 P.merge.add(loc_[k].load(std::memory_order_acquire), offs[i], l);
@@ -734,6 +782,7 @@ class Server {
           remote.insert(remote.end(), P.remote.begin(), P.remote.end());
           stat_push_replica_ += P.n_repl;
         }
+        record_spill_touches(parts);
         stat_push_local_ += n - (int64_t)remote.size();
         stat_push_keys_ += n;
         if (cpp_timing_) { int64_t t1 = tick(); t_pass_ += t1 - tp0; tp0 = t1; }
@@ -1971,14 +2020,18 @@ class Server {
     return (loc_[k].load(std::memory_order_acquire) & SPILL_BIT) ? 1 : 0;
   }
 
-  // Swap the hottest host-spilled rows with the coldest HBM-resident
-  // rows ("HBM as a cache over pinned host memory" — the tier the
-  // reference cannot have: its store is all host DRAM). Heat = access
-  // count since the last rebalance (halved each call, so it is an EWMA).
-  // Stop-the-world for the swap batch: migrating_ gates InflightGuard
-  // entry, quiesce() drains ops that already read metadata, then the
-  // loc_ swaps + the staging kernels are enqueued before workers resume —
-  // single-stream order makes the swap invisible to them. Currently
+  // Promote the hottest host-spilled rows into HBM ("HBM as a cache
+  // over pinned host memory" — the tier the all-host reference cannot
+  // have). Heat = per-key access count (halved every 8th call → EWMA);
+  // candidates come from the touched-list the metadata pass records, so
+  // a call is O(touched + moves), NOT O(num_keys). If the device arena
+  // has headroom the row simply moves; otherwise it swaps with a cold
+  // HBM row found by random sampling (power-of-choices; a 2x-heat
+  // hysteresis stops ping-ponging). The move batch is stop-the-world:
+  // migrating_ gates InflightGuard entry, quiesce() drains ops that
+  // already read metadata, then the loc_ updates and one batched
+  // gather+scatter staging pass are enqueued before workers resume —
+  // single-stream order makes the move invisible to them. Currently
   // world==1 only (sync threads are not gated; at world==1 they carry
   // no traffic).
   int64_t rebalance_spill(int64_t max_moves = 4096) {
@@ -1986,86 +2039,90 @@ class Server {
     TORCH_CHECK(uniform_len_ >= 0, "rebalance_spill requires a uniform-length store");
     if (!heat_ || slab_.host_capacity == 0 || max_moves <= 0) return 0;
     const int32_t l = uniform_len_;
+    rebalance_calls_++;
+
+    std::vector<Key> touched;
+    {
+      std::lock_guard<std::mutex> g(spill_mu_);
+      touched.swap(spill_touched_);
+    }
+    if (touched.empty()) return 0;
+    std::sort(touched.begin(), touched.end());
+    touched.erase(std::unique(touched.begin(), touched.end()), touched.end());
 
     struct Cand { Key k; uint32_t h; };
-    // collect top-heat spilled keys and bottom-heat device keys with
-    // bounded per-chunk partial selections (full sort of 100M keys would
-    // dwarf the swap itself); heat is halved in the same pass.
-    constexpr int64_t G = 1 << 20;
-    int64_t nchunks = (num_keys_ + G - 1) / G;
-    std::vector<std::vector<Cand>> hot_parts(nchunks), cold_parts(nchunks);
-    at::parallel_for(0, nchunks, 1, [&](int64_t c0, int64_t c1) {
-      for (int64_t c = c0; c < c1; ++c) {
-        auto& hot = hot_parts[c];
-        auto& cold = cold_parts[c];
-        Key e = (Key)std::min<int64_t>(num_keys_, (c + 1) * G);
-        for (Key k = (Key)(c * G); k < e; ++k) {
-          uint32_t h = heat_[k].load(std::memory_order_relaxed);
-          if (h) heat_[k].store(h >> 1, std::memory_order_relaxed);
-          uint8_t f = flags_[k].load(std::memory_order_acquire);
-          if (!(f & F_PRESENT) || (f & F_STUB)) continue;
-          bool spilled = (loc_[k].load(std::memory_order_acquire) & SPILL_BIT) != 0;
-          if (spilled) {
-            if (h == 0) continue;
-            hot.push_back({k, h});
-            if ((int64_t)hot.size() >= 2 * max_moves) {
-              std::nth_element(hot.begin(), hot.begin() + max_moves - 1, hot.end(),
-                               [](const Cand& a, const Cand& b) { return a.h > b.h; });
-              hot.resize(max_moves);
-            }
-          } else {
-            cold.push_back({k, h});
-            if ((int64_t)cold.size() >= 2 * max_moves) {
-              std::nth_element(cold.begin(), cold.begin() + max_moves - 1, cold.end(),
-                               [](const Cand& a, const Cand& b) { return a.h < b.h; });
-              cold.resize(max_moves);
-            }
-          }
-        }
-      }
-    });
-    std::vector<Cand> hot, cold;
-    for (auto& p : hot_parts) hot.insert(hot.end(), p.begin(), p.end());
-    for (auto& p : cold_parts) cold.insert(cold.end(), p.begin(), p.end());
-    if (hot.empty()) return 0;
+    std::vector<Cand> hot;
+    hot.reserve(touched.size());
+    for (Key k : touched) hot.push_back({k, heat_[k].load(std::memory_order_relaxed)});
     auto hotter = [](const Cand& a, const Cand& b) { return a.h > b.h; };
-    auto colder = [](const Cand& a, const Cand& b) { return a.h < b.h; };
     if ((int64_t)hot.size() > max_moves) {
       std::nth_element(hot.begin(), hot.begin() + max_moves - 1, hot.end(), hotter);
       hot.resize(max_moves);
     }
-    if ((int64_t)cold.size() > max_moves) {
-      std::nth_element(cold.begin(), cold.begin() + max_moves - 1, cold.end(), colder);
-      cold.resize(max_moves);
-    }
     std::sort(hot.begin(), hot.end(), hotter);
-    std::sort(cold.begin(), cold.end(), colder);
+
+    // eviction candidates: random sample of device-resident keys
+    // (collected lazily below, only if promotion needs swaps)
+    std::mt19937_64 rng(0x9e3779b97f4a7c15ULL ^ (uint64_t)rebalance_calls_);
+    std::vector<Cand> cold;
+    auto sample_cold = [&]() {
+      int64_t want = 2 * max_moves;
+      for (int64_t tries = 0; tries < 8 * want && (int64_t)cold.size() < want; ++tries) {
+        Key k = (Key)(rng() % (uint64_t)num_keys_);
+        uint8_t f = flags_[k].load(std::memory_order_acquire);
+        if (!(f & F_PRESENT) || (f & F_STUB)) continue;
+        if (loc_[k].load(std::memory_order_acquire) & SPILL_BIT) continue;
+        cold.push_back({k, heat_[k].load(std::memory_order_relaxed)});
+      }
+      std::sort(cold.begin(), cold.end(),
+                [](const Cand& a, const Cand& b) { return a.h < b.h; });
+    };
 
     // stop the world
     migrating_.store(1, std::memory_order_release);
     quiesce();
     int64_t moves = 0;
     HostBatch from, to;
+    std::vector<std::pair<int64_t, int32_t>> frees;  // freed AFTER kernels are enqueued
     int64_t pos = 0;
     size_t ci = 0;
+    bool cold_sampled = false;
     for (auto& hc : hot) {
-      if (moves >= max_moves || ci >= cold.size()) break;
+      if (moves >= max_moves) break;
+      Key ks = hc.k;
+      uint8_t fs = flags_[ks].load(std::memory_order_acquire);
+      if (!(fs & F_PRESENT) || (fs & F_STUB)) continue;
+      int64_t off_s = loc_[ks].load(std::memory_order_acquire);
+      if (!(off_s & SPILL_BIT)) continue;
+      // free HBM headroom? plain move, no eviction
+      int64_t off_new = slab_.try_alloc_device(l);
+      if (off_new >= 0) {
+        from.add(off_s, pos, l);
+        to.add(off_new, pos, l);
+        loc_[ks].store(off_new, std::memory_order_release);
+        frees.push_back({off_s, l});
+        pos += l;
+        moves++;
+        continue;
+      }
+      if (!cold_sampled) {
+        sample_cold();
+        cold_sampled = true;
+      }
+      // skip cold entries invalidated since sampling
+      while (ci < cold.size()) {
+        Key kd = cold[ci].k;
+        uint8_t fd = flags_[kd].load(std::memory_order_acquire);
+        if ((fd & F_PRESENT) && !(fd & F_STUB) &&
+            !(loc_[kd].load(std::memory_order_acquire) & SPILL_BIT))
+          break;
+        ci++;
+      }
+      if (ci >= cold.size()) break;
       // hysteresis: a swap must be clearly profitable or keys ping-pong
       if ((int64_t)hc.h <= 2 * (int64_t)cold[ci].h + 1) break;
-      Key ks = hc.k, kd = cold[ci].k;
-      // re-validate (metadata may have changed since the lock-free scan)
-      uint8_t fs = flags_[ks].load(std::memory_order_acquire);
-      uint8_t fd = flags_[kd].load(std::memory_order_acquire);
-      if (!(fs & F_PRESENT) || (fs & F_STUB) || !(fd & F_PRESENT) || (fd & F_STUB)) {
-        ci++;
-        continue;
-      }
-      int64_t off_s = loc_[ks].load(std::memory_order_acquire);
+      Key kd = cold[ci].k;
       int64_t off_d = loc_[kd].load(std::memory_order_acquire);
-      if (!(off_s & SPILL_BIT) || (off_d & SPILL_BIT)) {
-        ci++;
-        continue;
-      }
       from.add(off_s, pos, l);
       from.add(off_d, pos + l, l);
       to.add(off_d, pos, l);
@@ -2080,9 +2137,21 @@ class Server {
       auto tmp = torch::empty({pos}, torch::TensorOptions().dtype(torch::kFloat32).device(dev_));
       run_gather(from, tmp);
       run_scatter(to, tmp, /*set=*/true);
+      for (auto& fr : frees) slab_.free_(fr.first, fr.second);
     }
     migrating_.store(0, std::memory_order_release);
     stat_spill_moves_ += moves;
+
+    // EWMA decay: a cheap full halving every 8th call keeps old heat
+    // from pinning stale residents (O(num_keys) but branch-free)
+    if ((rebalance_calls_ & 7) == 0) {
+      at::parallel_for(0, num_keys_, 1 << 20, [&](int64_t b, int64_t e) {
+        for (int64_t k = b; k < e; ++k) {
+          uint32_t h = heat_[k].load(std::memory_order_relaxed);
+          if (h) heat_[k].store(h >> 1, std::memory_order_relaxed);
+        }
+      });
+    }
     return moves;
   }
 
@@ -2271,6 +2340,9 @@ class Server {
   std::atomic<int> inflight_{0};
   std::atomic<int> migrating_{0};                       // spill-rebalance stop-the-world gate
   std::unique_ptr<std::atomic<uint32_t>[]> heat_;       // per-key access heat (spill stores)
+  std::mutex spill_mu_;
+  std::vector<Key> spill_touched_;                      // spilled keys accessed since last rebalance
+  int64_t rebalance_calls_ = 0;
   std::atomic<int64_t> stat_spill_moves_{0};
   std::atomic<int64_t> next_ts_{1};
  public:
