@@ -92,6 +92,9 @@ def main():
         el = worker.allreduce(el, op="max")
 
     st = server.stats()
+    if rank == 0 and os.environ.get("ADAPM_CPP_TIMING"):
+        print("cpp timing:", {k: v for k, v in st.items() if k.startswith("t_")},
+              file=sys.stderr, flush=True)
     if rank == 0:
         print(json.dumps({
             "metric": "ctr_examples_per_s", "value": args.batch * args.steps * world / el,
