@@ -435,6 +435,12 @@ class Server {
       ops_scatter_cpu(slab_.bases(), d.b, in_c.data_ptr<float>(), set);
     }
   }
+  void run_scatter_rmw(const HostBatch& hb, torch::Tensor in) {
+    if (hb.size() == 0) return;
+    auto d = to_dev(hb);
+    auto in_c = in.is_contiguous() ? in : in.contiguous();
+    ops_scatter_rmw_gpu(slab_.bases(), d.b, in_c.data_ptr<float>(), current_stream(dev_));
+  }
   void run_extract(const HostBatch& hb, const std::vector<int64_t>& sync_off, torch::Tensor out) {
     if (hb.size() == 0) return;
     auto d = to_dev(hb, &sync_off);
@@ -739,7 +745,7 @@ class Server {
         constexpr int64_t G = 8192;
         int64_t nchunks = (n + G - 1) / G;
         struct Part {
-          HostBatch merge, assign;
+          HostBatch merge, assign, merge_spill;
           std::vector<Remote> remote;
           std::vector<Key> spilled;
           int64_t n_repl = 0;
@@ -755,7 +761,13 @@ class Server {
               uint8_t f = flags_[k].load(std::memory_order_acquire);
               if ((f & F_PRESENT) && (f & F_OWNER)) {
                 int64_t off = loc_[k].load(std::memory_order_acquire);
-                (set_mode ? P.assign : P.merge).add(off, offs[i], l);
+                // spilled merges go through the non-atomic RMW kernel
+                // (PCIe atomics are ~144x slower; dups re-routed below)
+                bool spill_merge = heat_ && (off & SPILL_BIT) && !set_mode && dev_.is_cuda();
+                if (spill_merge)
+                  P.merge_spill.add(off, offs[i], l);
+                else
+                  (set_mode ? P.assign : P.merge).add(off, offs[i], l);
                 version_[k].fetch_add(1, std::memory_order_relaxed);
                 if (heat_) {
                   heat_[k].fetch_add(1, std::memory_order_relaxed);
@@ -785,9 +797,31 @@ class Server {
         record_spill_touches(parts);
         stat_push_local_ += n - (int64_t)remote.size();
         stat_push_keys_ += n;
+        // spilled merges: unique destinations use the RMW kernel; a slot
+        // hit twice in this batch falls back to the atomic path
+        HostBatch rmw;
+        {
+          HostBatch all_spill;
+          for (auto& P : parts) {
+            all_spill.src.insert(all_spill.src.end(), P.merge_spill.src.begin(), P.merge_spill.src.end());
+            all_spill.dst.insert(all_spill.dst.end(), P.merge_spill.dst.begin(), P.merge_spill.dst.end());
+            all_spill.len.insert(all_spill.len.end(), P.merge_spill.len.begin(), P.merge_spill.len.end());
+          }
+          if (all_spill.size()) {
+            std::unordered_set<int64_t> seen;
+            seen.reserve(all_spill.size() * 2);
+            for (size_t i = 0; i < all_spill.size(); ++i) {
+              if (seen.insert(all_spill.src[i]).second)
+                rmw.add(all_spill.src[i], all_spill.dst[i], all_spill.len[i]);
+              else
+                merge.add(all_spill.src[i], all_spill.dst[i], all_spill.len[i]);
+            }
+          }
+        }
         if (cpp_timing_) { int64_t t1 = tick(); t_pass_ += t1 - tp0; tp0 = t1; }
         run_scatter(merge, flat, false);
         run_scatter(assign, flat, true);
+        run_scatter_rmw(rmw, flat);
         if (cpp_timing_) { t_launch_ += tick() - tp0; t_calls_++; }
       }
     }

@@ -39,6 +39,12 @@ void ops_gather_cpu(const SlabBases& slab, const OpsBatch& b, float* out);
 // slab[dst(src)_off[i]] += in[...]   (atomic on GPU)  — or assign when set=true
 void ops_scatter_gpu(const SlabBases& slab, const OpsBatch& b, const float* in, bool set, void* stream);
 void ops_scatter_cpu(const SlabBases& slab, const OpsBatch& b, const float* in, bool set);
+// merge (+=) with plain load/store instead of atomics. ONLY safe when no
+// two batch entries share a destination slot. Exists for host-spilled
+// rows: an atomicAdd to pinned host memory is a non-posted PCIe
+// round-trip per dword (measured 144x slower than the zero-copy read),
+// while a read+add+posted-write streams at PCIe bandwidth.
+void ops_scatter_rmw_gpu(const SlabBases& slab, const OpsBatch& b, const float* in, void* stream);
 
 // replica delta extraction (src_off = val offsets, dst_off = out buffer
 // offsets, aux_off = sync_state offsets):

@@ -56,6 +56,32 @@ __global__ void k_scatter_add(float* __restrict__ sdev, float* __restrict__ shos
   }
 }
 
+// non-atomic merge for unique destination slots (host-spill fast path;
+// see ops.h) — vectorized read+add+posted-write
+__global__ void k_scatter_rmw(float* __restrict__ sdev, float* __restrict__ shost,
+                              const int64_t* __restrict__ src_off,
+                              const int64_t* __restrict__ dst_off, const int32_t* __restrict__ lens,
+                              int n, const float* __restrict__ in) {
+  for (int i = blockIdx.x; i < n; i += gridDim.x) {
+    int64_t s = src_off[i];
+    if (s < 0) continue;
+    float* slab = sel_base(sdev, shost, s);
+    int64_t d = dst_off[i];
+    int32_t len = lens[i];
+    if (vec_ok(s, d, len)) {
+      float4* sp = reinterpret_cast<float4*>(slab + s);
+      const float4* dp = reinterpret_cast<const float4*>(in + d);
+      for (int e = threadIdx.x; e < (len >> 2); e += THREADS) {
+        float4 v = sp[e], w = dp[e];
+        v.x += w.x; v.y += w.y; v.z += w.z; v.w += w.w;
+        sp[e] = v;
+      }
+    } else {
+      for (int e = threadIdx.x; e < len; e += THREADS) slab[s + e] += in[d + e];
+    }
+  }
+}
+
 __global__ void k_scatter_set(float* __restrict__ sdev, float* __restrict__ shost,
                               const int64_t* __restrict__ src_off,
                               const int64_t* __restrict__ dst_off, const int32_t* __restrict__ lens,
@@ -194,6 +220,11 @@ void ops_scatter_gpu(const SlabBases& slab, const OpsBatch& b, const float* in, 
   else
     hipLaunchKernelGGL(k_scatter_add, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
                        slab.dev, slab.host, b.src_off, b.dst_off, b.lens, b.n, in);
+}
+void ops_scatter_rmw_gpu(const SlabBases& slab, const OpsBatch& b, const float* in, void* stream) {
+  if (b.n == 0) return;
+  hipLaunchKernelGGL(k_scatter_rmw, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
+                     slab.dev, slab.host, b.src_off, b.dst_off, b.lens, b.n, in);
 }
 void ops_extract_gpu(const SlabBases& slab, const OpsBatch& b, const int64_t* sync_off, float* out, void* stream) {
   if (b.n == 0) return;
