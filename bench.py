@@ -194,6 +194,7 @@ def main():
                 "global_batch": args.batch * world,
                 "triples": args.triples,
                 "parallelism": f"ps-async-dp{world}",
+                "fused_step": (not args.no_fused) and world == 1 and dev.type == "cuda",
                 "epoch_time_s": epoch_time_s,
                 "triples_per_s": triples_per_s,
                 "init_s": init_s,
