@@ -2099,16 +2099,19 @@ class Server {
     // (collected lazily below, only if promotion needs swaps)
     std::mt19937_64 rng(0x9e3779b97f4a7c15ULL ^ (uint64_t)rebalance_calls_);
     std::vector<Cand> cold;
+    // incremental: each call samples one chunk, so sampling cost tracks
+    // the number of swaps actually performed, not max_moves
     auto sample_cold = [&]() {
-      int64_t want = 2 * max_moves;
-      for (int64_t tries = 0; tries < 8 * want && (int64_t)cold.size() < want; ++tries) {
+      constexpr int64_t CHUNK = 8192;
+      size_t before = cold.size();
+      for (int64_t tries = 0; tries < 8 * CHUNK && cold.size() < before + CHUNK; ++tries) {
         Key k = (Key)(rng() % (uint64_t)num_keys_);
         uint8_t f = flags_[k].load(std::memory_order_acquire);
         if (!(f & F_PRESENT) || (f & F_STUB)) continue;
         if (loc_[k].load(std::memory_order_acquire) & SPILL_BIT) continue;
         cold.push_back({k, heat_[k].load(std::memory_order_relaxed)});
       }
-      std::sort(cold.begin(), cold.end(),
+      std::sort(cold.begin() + before, cold.end(),
                 [](const Cand& a, const Cand& b) { return a.h < b.h; });
     };
 
@@ -2120,7 +2123,6 @@ class Server {
     std::vector<std::pair<int64_t, int32_t>> frees;  // freed AFTER kernels are enqueued
     int64_t pos = 0;
     size_t ci = 0;
-    bool cold_sampled = false;
     for (auto& hc : hot) {
       if (moves >= max_moves) break;
       Key ks = hc.k;
@@ -2139,9 +2141,10 @@ class Server {
         moves++;
         continue;
       }
-      if (!cold_sampled) {
+      if (ci >= cold.size()) {
+        size_t before = cold.size();
         sample_cold();
-        cold_sampled = true;
+        if (cold.size() == before) break;  // nothing evictable found
       }
       // skip cold entries invalidated since sampling
       while (ci < cold.size()) {
@@ -2152,7 +2155,7 @@ class Server {
           break;
         ci++;
       }
-      if (ci >= cold.size()) break;
+      if (ci >= cold.size()) continue;
       // hysteresis: a swap must be clearly profitable or keys ping-pong
       if ((int64_t)hc.h <= 2 * (int64_t)cold[ci].h + 1) break;
       Key kd = cold[ci].k;

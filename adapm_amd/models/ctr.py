@@ -68,17 +68,32 @@ class WideAndDeep:
         self.worker.intent(np.unique(feats.reshape(-1)), start, end)
 
     def train_batch(self, feats: np.ndarray, labels: np.ndarray, sync_loss=True):
-        """feats: [B, fields] int64 feature ids; labels: [B] {0,1}."""
+        """feats: [B, fields] int64 feature ids; labels: [B] {0,1}.
+
+        Keys are dedup'd per batch: the store sees each distinct key once
+        (Zipf batches repeat hot ids thousands of times), autograd
+        aggregates the per-occurrence gradients on-device via the
+        index_select backward (a scatter-add in HBM), and AdaGrad then
+        transforms the aggregated gradient — one pull row + one push row
+        per distinct key."""
         cfg = self.cfg
         w = self.worker
         B, F = feats.shape
         keys = feats.reshape(-1).astype(np.int64)
-        rows = torch.empty(B * F, cfg.row, dtype=torch.float32, device=self.dev)
-        w.wait(w.pull(keys, rows, async_=True))
+        if self.dev.type == "cuda":
+            keys_u_t, inv = torch.unique(torch.from_numpy(keys).to(self.dev),
+                                         return_inverse=True)
+            keys_u = keys_u_t.cpu().numpy()
+        else:
+            keys_u, inv = np.unique(keys, return_inverse=True)
+            inv = torch.from_numpy(inv)
+        U = len(keys_u)
+        rows = torch.empty(U, cfg.row, dtype=torch.float32, device=self.dev)
+        w.wait(w.pull(keys_u, rows, async_=True))
         emb = rows[:, :cfg.dim].detach().clone().requires_grad_(True)
         accum = rows[:, cfg.dim:]
 
-        x = emb.view(B, F * cfg.dim)
+        x = emb.index_select(0, inv.to(self.dev)).view(B, F * cfg.dim)
         logits = self.net(x).squeeze(1)
         y = torch.as_tensor(labels.astype(np.float32), device=self.dev)
         loss = torch.nn.functional.binary_cross_entropy_with_logits(logits, y)
@@ -86,13 +101,14 @@ class WideAndDeep:
         loss.backward()
         self.opt.step()
 
-        # AdaGrad-transform the embedding grads into a push-ready delta
+        # AdaGrad-transform the aggregated embedding grads into a
+        # push-ready delta (one row per distinct key)
         g = emb.grad
         delta = torch.empty_like(rows)
         g2 = g * g
         delta[:, :cfg.dim] = -cfg.lr * g / torch.sqrt(accum + g2 + cfg.eps)
         delta[:, cfg.dim:] = g2
-        pt = w.push(keys, delta, async_=True)
+        pt = w.push(keys_u, delta, async_=True)
         if pt != -1:
             self._pending.append(pt)
         while len(self._pending) > 64:

@@ -111,3 +111,48 @@ def test_spill_rebalance_cpu():
 @pytest.mark.gpu
 def test_spill_rebalance_gpu():
     _run_rebalance("cuda:0")
+
+
+def test_rebalance_concurrent_with_workers_cpu():
+    """The stop-the-world gate: worker threads hammer pulls/pushes while
+    the main thread rebalances repeatedly; no op may observe a torn
+    migration (values stay exact)."""
+    import threading
+
+    adapm_amd, s, w = _mk_tiered("cpu")
+    keys = np.arange(400, dtype=np.int64)
+    w.set(keys, torch.zeros(400, 64))
+    pushes = np.zeros(400)
+    stop = threading.Event()
+    errs = []
+
+    def hammer(seed):
+        rng = np.random.default_rng(seed)
+        local = np.zeros(400)
+        try:
+            while not stop.is_set():
+                ks = rng.choice(400, size=32, replace=False).astype(np.int64)
+                w.push(ks, np.ones((32, 64), dtype=np.float32))
+                local[ks] += 1
+                out = np.zeros((32, 64), dtype=np.float32)
+                w.pull(ks, out)
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+        results.append(local)
+
+    results = []
+    threads = [threading.Thread(target=hammer, args=(i,)) for i in range(2)]
+    for t in threads:
+        t.start()
+    for _ in range(30):
+        s.raw.rebalance_spill(256)
+    stop.set()
+    for t in threads:
+        t.join(timeout=60)
+    assert not errs, errs
+    total = sum(results)
+    out = np.zeros((400, 64), dtype=np.float32)
+    w.pull(keys, out)
+    assert np.allclose(out, total[:, None]), \
+        f"mismatch at {np.where(np.abs(out[:, 0] - total) > 1e-3)[0]}"
+    s.shutdown()
