@@ -62,10 +62,14 @@ def main():
     if args.hash_ids:
         batches = [((f * 2654435761) % args.features, y) for f, y in batches]
     do_rebalance = args.rebalance_every > 0 and world == 1
+    rebal_stats = {"s": 0.0, "n": 0, "moves": 0}
 
     def maybe_rebalance(i):
         if do_rebalance and i % args.rebalance_every == 0:
-            server.raw.rebalance_spill(args.rebalance_moves)
+            t = time.perf_counter()
+            rebal_stats["moves"] += server.raw.rebalance_spill(args.rebalance_moves)
+            rebal_stats["s"] += time.perf_counter() - t
+            rebal_stats["n"] += 1
 
     is_cuda = server.rt.device.type == "cuda"
     for i in range(args.warmup):
@@ -92,6 +96,10 @@ def main():
         el = worker.allreduce(el, op="max")
 
     st = server.stats()
+    if rank == 0 and rebal_stats["n"]:
+        print(f"rebalance: {rebal_stats['n']} calls, {rebal_stats['s']:.3f}s total, "
+              f"{1000*rebal_stats['s']/rebal_stats['n']:.1f} ms/call, {rebal_stats['moves']} moves",
+              file=sys.stderr, flush=True)
     if rank == 0 and os.environ.get("ADAPM_CPP_TIMING"):
         print("cpp timing:", {k: v for k, v in st.items() if k.startswith("t_")},
               file=sys.stderr, flush=True)
