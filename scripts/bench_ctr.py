@@ -35,6 +35,9 @@ def main():
                     help=">0: call rebalance_spill every N steps (tiered store "
                          "keeps the hot set HBM-resident; world==1 only)")
     ap.add_argument("--rebalance-moves", type=int, default=131072)
+    ap.add_argument("--rebalance-warmup-only", action="store_true",
+                    help="stop rebalancing when the timed region starts (isolates "
+                         "steady-state layout effect from rebalance call cost)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -65,6 +68,8 @@ def main():
     rebal_stats = {"s": 0.0, "n": 0, "moves": 0}
 
     def maybe_rebalance(i):
+        if args.rebalance_warmup_only and i >= args.warmup:
+            return
         if do_rebalance and i % args.rebalance_every == 0:
             t = time.perf_counter()
             rebal_stats["moves"] += server.raw.rebalance_spill(args.rebalance_moves)
