@@ -45,6 +45,7 @@ _SETUP = {}
 
 def setup(num_keys: int, num_threads: int, use_techniques: str = "", num_channels: int = -1,
           device: Optional[str] = None, max_sync_per_sec: float = 1000.0,
+          sync_threshold: float = 0.0,
           time_intent_actions: bool = True, capacity_factor: float = 2.0,
           location_caches: bool = True, locality_stats: bool = False,
           trace_keys=None, stats_out: Optional[str] = None,
@@ -63,6 +64,7 @@ def setup(num_keys: int, num_threads: int, use_techniques: str = "", num_channel
     _SETUP.update(dict(num_keys=num_keys, num_threads=num_threads, techniques=tech,
                        num_channels=num_channels, device=device,
                        max_sync_per_sec=max_sync_per_sec,
+                       sync_threshold=sync_threshold,
                        time_intent_actions=time_intent_actions,
                        capacity_factor=capacity_factor,
                        location_caches=location_caches, locality_stats=locality_stats,
@@ -112,6 +114,7 @@ class Server:
             techniques=cfg["techniques"], location_caches=cfg["location_caches"],
             device_cap_floats=int(cfg.get("device_cap_gb", 0) * (1 << 30) / 4),
             host_spill_floats=int(cfg.get("host_spill_gb", 0) * (1 << 30) / 4),
+            sync_threshold=cfg.get("sync_threshold", 0.0),
         )
         if cfg.get("locality_stats"):
             self._s.enable_locality_stats()

@@ -265,13 +265,15 @@ class Server {
  public:
   Server(int64_t num_keys, torch::Tensor lens, int rank, int world, int num_channels,
          int num_workers, std::string device, double capacity_factor, int techniques,
-         bool location_caches, int64_t device_cap_floats = 0, int64_t host_spill_floats = 0)
+         bool location_caches, int64_t device_cap_floats = 0, int64_t host_spill_floats = 0,
+         double sync_threshold = 0.0)
       : num_keys_(num_keys),
         rank_(rank),
         world_(world),
         nch_(num_channels),
         techniques_(techniques),
         use_loc_cache_(location_caches),
+        sync_threshold_(sync_threshold),
         dev_(device) {
     TORCH_CHECK((nch_ & (nch_ - 1)) == 0, "num_channels must be a power of 2");
     log2ch_ = 0;
@@ -440,6 +442,18 @@ class Server {
     auto d = to_dev(hb);
     auto in_c = in.is_contiguous() ? in : in.contiguous();
     ops_scatter_rmw_gpu(slab_.bases(), d.b, in_c.data_ptr<float>(), current_stream(dev_));
+  }
+  void run_delta_sqnorm(const HostBatch& hb, const std::vector<int64_t>& sync_off,
+                        torch::Tensor out) {
+    if (hb.size() == 0) return;
+    auto d = to_dev(hb, &sync_off);
+    if (dev_.is_cuda()) {
+      ops_delta_sqnorm_gpu(slab_.bases(), d.b, d.aux_t.data_ptr<int64_t>(),
+                           out.data_ptr<float>(), current_stream(dev_));
+    } else {
+      std::lock_guard<std::mutex> g(cpu_val_mu_);
+      ops_delta_sqnorm_cpu(slab_.bases(), d.b, sync_off.data(), out.data_ptr<float>());
+    }
   }
   void run_extract(const HostBatch& hb, const std::vector<int64_t>& sync_off, torch::Tensor out) {
     if (hb.size() == 0) return;
@@ -1069,6 +1083,47 @@ class Server {
       replica_snapshot.assign(C.replicas.begin(), C.replicas.end());
     }
 
+    // 2b. sync threshold (reference --sys.sync.threshold,
+    // readAndPotentiallyDropReplica sync_manager.h:601-662): updated
+    // replicas whose pending-delta L2 norm is below the threshold skip
+    // this round — UPDATED stays set and sync_state untouched, so the
+    // delta keeps accumulating and ships later (or with the drop, which
+    // always carries a payload). The norm is computed on-device; the
+    // readback is the round's only sync point and only exists when a
+    // threshold is configured.
+    std::unordered_set<Key> below_threshold;
+    if (sync_threshold_ > 0.0) {
+      HostBatch nb;
+      std::vector<int64_t> nsync;
+      std::vector<Key> cand;
+      for (Key k : replica_snapshot) {
+        bool has_intent;
+        {
+          std::lock_guard<std::mutex> g(C.mu);
+          has_intent = C.intents.count(k) > 0;
+        }
+        if (!has_intent) continue;  // would drop: always ships
+        std::lock_guard<std::mutex> lk(stripe(k));
+        uint8_t f = flags_[k];
+        if (!(f & F_PRESENT) || (f & F_OWNER)) continue;
+        if (!(f & F_UPDATED) || (f & F_STUB)) continue;
+        nb.add(loc_[k].load(), (int64_t)cand.size(), len_of(k));
+        nsync.push_back(sync_loc_[k]);
+        cand.push_back(k);
+      }
+      if (!cand.empty()) {
+        quiesce();  // norms must follow in-flight push kernels
+        auto norms = torch::empty({(int64_t)cand.size()},
+                                  torch::TensorOptions().dtype(torch::kFloat32).device(dev_));
+        run_delta_sqnorm(nb, nsync, norms);
+        auto h = norms.cpu();
+        const float* hp = h.data_ptr<float>();
+        float thr2 = (float)(sync_threshold_ * sync_threshold_);
+        for (size_t i = 0; i < cand.size(); ++i)
+          if (hp[i] < thr2) below_threshold.insert(cand[i]);
+      }
+    }
+
     // 3. per replica: extract delta / drop (reference readAndPotentiallyDropReplica)
     struct DeltaRec {
       Key k;
@@ -1091,7 +1146,7 @@ class Server {
         if (!(f & F_PRESENT) || (f & F_OWNER)) {
           erase_from_replicas = true;  // became owner via relocation
         } else {
-          bool updated = f & F_UPDATED;
+          bool updated = (f & F_UPDATED) && !below_threshold.count(k);
           bool is_new = f & F_STUB;
           bool drop = !has_intent && !is_new;
           // dropped replicas ALWAYS carry a payload: a concurrent local
@@ -2309,6 +2364,7 @@ class Server {
     d["slab_capacity"] = slab_.capacity;
     d["host_spill_in_use"] = slab_.host_in_use.load();
     d["spill_rebalance_moves"] = stat_spill_moves_.load();
+    d["sync_threshold"] = sync_threshold_;
     d["host_spill_capacity"] = slab_.host_capacity;
     int64_t rounds = 0;
     for (auto& c : channels_) rounds += c.rounds.load();
@@ -2376,6 +2432,7 @@ class Server {
 
   std::atomic<int> inflight_{0};
   std::atomic<int> migrating_{0};                       // spill-rebalance stop-the-world gate
+  double sync_threshold_ = 0.0;                         // --sys.sync.threshold equivalent
   std::unique_ptr<std::atomic<uint32_t>[]> heat_;       // per-key access heat (spill stores)
   std::mutex spill_mu_;
   std::vector<Key> spill_touched_;                      // spilled keys accessed since last rebalance
@@ -2527,12 +2584,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mf_update_step", &mf_update_step, py::call_guard<py::gil_scoped_release>());
   py::class_<Server>(m, "Server")
       .def(py::init<int64_t, torch::Tensor, int, int, int, int, std::string, double, int, bool,
-                    int64_t, int64_t>(),
+                    int64_t, int64_t, double>(),
            py::arg("num_keys"), py::arg("value_lengths"), py::arg("rank"), py::arg("world"),
            py::arg("num_channels"), py::arg("num_workers"), py::arg("device"),
            py::arg("capacity_factor") = 2.0, py::arg("techniques") = 0,
            py::arg("location_caches") = true, py::arg("device_cap_floats") = 0,
-           py::arg("host_spill_floats") = 0)
+           py::arg("host_spill_floats") = 0, py::arg("sync_threshold") = 0.0)
       .def("pull", &Server::pull, py::call_guard<py::gil_scoped_release>())
       .def("push", &Server::push, py::call_guard<py::gil_scoped_release>())
       .def("pull_if_local", &Server::pull_if_local, py::call_guard<py::gil_scoped_release>())

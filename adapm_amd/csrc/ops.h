@@ -45,6 +45,15 @@ void ops_scatter_cpu(const SlabBases& slab, const OpsBatch& b, const float* in, 
 // round-trip per dword (measured 144x slower than the zero-copy read),
 // while a read+add+posted-write streams at PCIe bandwidth.
 void ops_scatter_rmw_gpu(const SlabBases& slab, const OpsBatch& b, const float* in, void* stream);
+// per-key squared L2 norm of the pending delta (val - sync_state):
+// out[dst_off[i]] = ||slab[src_off[i]..] - slab[sync_off[i]..]||^2.
+// Drives the reference's --sys.sync.threshold ("send only deltas whose
+// norm clears the threshold"; sync_manager.h:601-662) without reading
+// the values back to the host.
+void ops_delta_sqnorm_gpu(const SlabBases& slab, const OpsBatch& b, const int64_t* sync_off,
+                          float* out, void* stream);
+void ops_delta_sqnorm_cpu(const SlabBases& slab, const OpsBatch& b, const int64_t* sync_off,
+                          float* out);
 
 // replica delta extraction (src_off = val offsets, dst_off = out buffer
 // offsets, aux_off = sync_state offsets):

@@ -105,3 +105,20 @@ void ops_zero_cpu(const SlabBases& sb, const OpsBatch& b) {
 }
 
 }  // namespace adapm
+
+namespace adapm {
+void ops_delta_sqnorm_cpu(const SlabBases& slab, const OpsBatch& b, const int64_t* sync_off,
+                          float* out) {
+  for (int i = 0; i < b.n; ++i) {
+    int64_t v = b.src_off[i], sy = sync_off[i];
+    const float* vb = sel_base_c(slab, v);
+    const float* sb = sel_base_c(slab, sy);
+    float acc = 0.f;
+    for (int e = 0; e < b.lens[i]; ++e) {
+      float d = vb[v + e] - sb[sy + e];
+      acc += d * d;
+    }
+    out[b.dst_off[i]] = acc;
+  }
+}
+}  // namespace adapm

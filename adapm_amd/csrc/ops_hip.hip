@@ -226,6 +226,42 @@ void ops_scatter_rmw_gpu(const SlabBases& slab, const OpsBatch& b, const float* 
   hipLaunchKernelGGL(k_scatter_rmw, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
                      slab.dev, slab.host, b.src_off, b.dst_off, b.lens, b.n, in);
 }
+__global__ void k_delta_sqnorm(float* __restrict__ sdev, float* __restrict__ shost,
+                               const int64_t* __restrict__ src_off,
+                               const int64_t* __restrict__ dst_off,
+                               const int32_t* __restrict__ lens, int n,
+                               const int64_t* __restrict__ sync_off, float* __restrict__ out) {
+  __shared__ float red[THREADS / 64];
+  for (int i = blockIdx.x; i < n; i += gridDim.x) {
+    int64_t v = src_off[i], sy = sync_off[i];
+    const float* vb = sel_base(sdev, shost, v);
+    const float* sb = sel_base(sdev, shost, sy);
+    int32_t len = lens[i];
+    float acc = 0.f;
+    for (int e = threadIdx.x; e < len; e += THREADS) {
+      float d = vb[v + e] - sb[sy + e];
+      acc += d * d;
+    }
+    // wave64 reduce then cross-wave via LDS
+    for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = acc;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      float t = 0.f;
+      for (int w = 0; w < THREADS / 64; ++w) t += red[w];
+      out[dst_off[i]] = t;
+    }
+    __syncthreads();
+  }
+}
+
+void ops_delta_sqnorm_gpu(const SlabBases& slab, const OpsBatch& b, const int64_t* sync_off,
+                          float* out, void* stream) {
+  if (b.n == 0) return;
+  hipLaunchKernelGGL(k_delta_sqnorm, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,
+                     slab.dev, slab.host, b.src_off, b.dst_off, b.lens, b.n, sync_off, out);
+}
+
 void ops_extract_gpu(const SlabBases& slab, const OpsBatch& b, const int64_t* sync_off, float* out, void* stream) {
   if (b.n == 0) return;
   hipLaunchKernelGGL(k_extract, dim3(grid_for(b.n)), dim3(THREADS), 0, (hipStream_t)stream,

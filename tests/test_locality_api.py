@@ -4,6 +4,7 @@ to the intender; under conflicting intent it is replicated; replicas are
 dropped when the intent window passes."""
 import time
 
+import numpy as np
 import torch
 
 from dist_helper import run_dist
@@ -126,3 +127,53 @@ def _pull_if_local_semantics(rank, world):
 
 def test_pull_if_local_ws2():
     run_dist(2, _pull_if_local_semantics, timeout=180)
+
+
+def _threshold_dist(rank, world):
+    """--sys.sync.threshold parity (reference sync_manager.h:601-662):
+    sub-threshold replica deltas are NOT sent (they keep accumulating in
+    sync_state-relative form); the drop at intent expiry always ships the
+    accumulated payload, so nothing is ever lost."""
+    import adapm_amd
+
+    adapm_amd.setup(num_keys=8, num_threads=1, device="cpu",
+                    max_sync_per_sec=2000.0, sync_threshold=0.5)
+    s = adapm_amd.Server(4)
+    w = adapm_amd.Worker(0, s)
+    w.barrier()
+    key = np.array([0], dtype=np.int64)  # owner = rank 0
+    # conflicting intent -> REPLICATION (a sole remote intent would
+    # relocate instead, and thresholds only govern replica sync)
+    w.intent(key, 1, 20)
+    time.sleep(0.3)
+    if rank == 1:
+        assert w.is_local(key[0])
+        # two pushes: the ACCUMULATED delta (what the threshold tests,
+        # like the reference's val - sync_state) reaches L2 norm 0.4 < 0.5
+        for _ in range(2):
+            w.push(key, np.full((1, 4), 0.1, dtype=np.float32))
+            w.wait_sync()
+    w.barrier()
+    out = np.zeros((1, 4), dtype=np.float32)
+    if rank == 0:
+        # owner must NOT have seen the sub-threshold deltas
+        w.pull(key, out)
+        assert np.allclose(out, 0.0), f"sub-threshold delta leaked: {out}"
+    w.barrier()
+    # expire the intent -> replica drop ships the accumulated payload
+    for _ in range(25):
+        w.advance_clock()
+    if rank == 1:
+        w.wait_sync()
+        w.wait_sync()
+    w.barrier()
+    w.wait_sync()
+    w.pull(key, out)
+    assert np.allclose(out, 0.2), f"accumulated delta lost on drop: {out}"
+    w.barrier()
+    w.finalize()
+    s.shutdown()
+
+
+def test_sync_threshold_ws2():
+    run_dist(2, _threshold_dist, timeout=240)
