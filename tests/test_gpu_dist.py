@@ -89,3 +89,53 @@ def _gpu_kge_dist(rank, world):
 
 def test_gpu_kge_distributed_ws2():
     run_dist(2, _gpu_kge_dist, timeout=300)
+
+
+def _gpu_threshold(rank, world):
+    os.environ["ADAPM_FORCE_GLOO"] = "1"
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    import adapm_amd
+
+    adapm_amd._SETUP.clear()
+    adapm_amd.runtime._RUNTIME = None
+    adapm_amd.setup(num_keys=8, num_threads=1, device="cuda:0",
+                    max_sync_per_sec=2000.0, sync_threshold=0.5)
+    s = adapm_amd.Server(4)
+    w = adapm_amd.Worker(0, s)
+    w.barrier()
+    key = np.array([0], dtype=np.int64)
+    w.intent(key, 1, 20)  # conflicting intent -> replication
+    import time
+
+    time.sleep(0.3)
+    if rank == 1:
+        for _ in range(2):
+            w.push(key, torch.full((1, 4), 0.1, device="cuda"))
+            w.wait_sync()
+    w.barrier()
+    out = torch.zeros(1, 4, device="cuda")
+    if rank == 0:
+        w.pull(key, out)
+        torch.cuda.synchronize()
+        assert torch.allclose(out, torch.zeros_like(out)), out
+    w.barrier()
+    for _ in range(25):
+        w.advance_clock()
+    if rank == 1:
+        w.wait_sync()
+        w.wait_sync()
+    w.barrier()
+    w.wait_sync()
+    w.pull(key, out)
+    torch.cuda.synchronize()
+    assert torch.allclose(out, torch.full_like(out, 0.2)), out
+    w.barrier()
+    w.finalize()
+    s.shutdown()
+
+
+def test_gpu_sync_threshold_ws2():
+    """k_delta_sqnorm on the HBM store: sub-threshold deltas held back,
+    accumulated payload ships with the drop."""
+    run_dist(2, _gpu_threshold, timeout=300)
