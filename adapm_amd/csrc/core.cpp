@@ -607,7 +607,7 @@ class Server {
             cum += l;
           }
         }
-        constexpr int64_t G = 8192;
+        constexpr int64_t G = 2048;
         int64_t nchunks = (n + G - 1) / G;
         struct Part {
           HostBatch local;
@@ -756,7 +756,7 @@ class Server {
             cum += len_of(kp[i]);
           }
         }
-        constexpr int64_t G = 8192;
+        constexpr int64_t G = 2048;
         int64_t nchunks = (n + G - 1) / G;
         struct Part {
           HostBatch merge, assign, merge_spill;
