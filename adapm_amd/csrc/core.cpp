@@ -387,6 +387,91 @@ class Server {
     size_t size() const { return src.size(); }
   };
 
+  // Persistent thread pool for the per-key metadata passes. ATen's
+  // at::parallel_for is a header template: compiled WITHOUT -fopenmp
+  // (as this extension is) its omp pragma is a no-op and the "parallel"
+  // pass runs serial; compiling WITH -fopenmp loads ROCm's libomp next
+  // to torch's libgomp and everything slows 2-4x (two spinning
+  // runtimes). A tiny dedicated pool avoids both. The pass is memory-
+  // latency-bound (~85 ns/key serial: two random cache misses over
+  // GB-scale metadata arrays), so threads scale it well.
+  struct PassPool {
+    struct Task {
+      const std::function<void(int64_t, int64_t)>* fn = nullptr;
+      int64_t n = 0;
+      std::atomic<int64_t>* next = nullptr;
+      std::atomic<int>* done = nullptr;
+      uint64_t gen = 0;
+    };
+    std::vector<std::thread> threads;
+    std::mutex mu;
+    std::condition_variable cv;
+    Task task;
+    uint64_t gen = 0;
+    bool stop = false;
+
+    ~PassPool() {
+      {
+        std::lock_guard<std::mutex> g(mu);
+        stop = true;
+      }
+      cv.notify_all();
+      for (auto& t : threads) t.join();
+    }
+
+    void ensure_started() {
+      if (!threads.empty()) return;
+      unsigned hw = std::thread::hardware_concurrency();
+      int nt = (int)std::min<unsigned>(hw > 2 ? hw - 2 : 1, 15);
+      for (int i = 0; i < nt; ++i)
+        threads.emplace_back([this] {
+          uint64_t seen = 0;
+          for (;;) {
+            Task t;
+            {
+              std::unique_lock<std::mutex> lk(mu);
+              cv.wait(lk, [&] { return stop || gen != seen; });
+              if (stop) return;
+              seen = gen;
+              t = task;
+            }
+            work(t);
+          }
+        });
+    }
+
+    static void work(const Task& t) {
+      constexpr int64_t STEP = 4;
+      for (;;) {
+        int64_t b = t.next->fetch_add(STEP, std::memory_order_relaxed);
+        if (b >= t.n) break;
+        (*t.fn)(b, std::min(t.n, b + STEP));
+      }
+      t.done->fetch_add(1, std::memory_order_acq_rel);
+    }
+
+    // run fn over [0, n) chunks; caller participates. fn must not throw
+    // on worker threads — callers validate inputs before entering.
+    void run(int64_t n, const std::function<void(int64_t, int64_t)>& fn) {
+      if (n <= 1) {
+        if (n == 1) fn(0, 1);
+        return;
+      }
+      ensure_started();
+      std::atomic<int64_t> next{0};
+      std::atomic<int> done{0};
+      {
+        std::lock_guard<std::mutex> g(mu);
+        gen++;
+        task = Task{&fn, n, &next, &done, gen};
+      }
+      cv.notify_all();
+      work(task);  // caller participates
+      int want = (int)threads.size() + 1;
+      while (done.load(std::memory_order_acquire) < want) std::this_thread::yield();
+    }
+  };
+
   struct DevBatch {
     torch::Tensor src_t, dst_t, len_t, aux_t;
     OpsBatch b;
@@ -616,7 +701,7 @@ class Server {
           int64_t n_repl = 0;
         };
         std::vector<Part> parts(nchunks);
-        at::parallel_for(0, nchunks, 1, [&](int64_t c0, int64_t c1) {
+        pass_pool_.run(nchunks, [&](int64_t c0, int64_t c1) {
           for (int64_t c = c0; c < c1; ++c) {
             Part& P = parts[c];
             int64_t e = std::min(n, (c + 1) * G);
@@ -765,7 +850,7 @@ class Server {
           int64_t n_repl = 0;
         };
         std::vector<Part> parts(nchunks);
-        at::parallel_for(0, nchunks, 1, [&](int64_t c0, int64_t c1) {
+        pass_pool_.run(nchunks, [&](int64_t c0, int64_t c1) {
           for (int64_t c = c0; c < c1; ++c) {
             Part& P = parts[c];
             int64_t e = std::min(n, (c + 1) * G);
@@ -2259,7 +2344,10 @@ class Server {
     const int64_t* cp = c.data_ptr<int64_t>();
     int64_t* op = out.data_ptr<int64_t>();
     std::atomic<int64_t> checks{0};
-    at::parallel_for(0, n, 4096, [&](int64_t b, int64_t e) {
+    constexpr int64_t SG = 4096;
+    int64_t snchunks = (n + SG - 1) / SG;
+    pass_pool_.run(snchunks, [&](int64_t c0, int64_t c1) {
+      int64_t b = c0 * SG, e = std::min(n, c1 * SG);
       int64_t local_checks = 0;
       for (int64_t i = b; i < e; ++i) {
         Key k = cp[i];
@@ -2431,6 +2519,7 @@ class Server {
   std::atomic<bool> layout_identity_{true};
 
   std::atomic<int> inflight_{0};
+  PassPool pass_pool_;
   std::atomic<int> migrating_{0};                       // spill-rebalance stop-the-world gate
   double sync_threshold_ = 0.0;                         // --sys.sync.threshold equivalent
   std::unique_ptr<std::atomic<uint32_t>[]> heat_;       // per-key access heat (spill stores)
