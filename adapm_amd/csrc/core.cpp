@@ -276,6 +276,15 @@ class Server {
         sync_threshold_(sync_threshold),
         dev_(device) {
     TORCH_CHECK((nch_ & (nch_ - 1)) == 0, "num_channels must be a power of 2");
+    {
+      // metadata-pass pool: share the node's cores across the co-located
+      // ranks (the driver runs world_ ranks per node), leave headroom
+      // for the sync threads and torch's own pools
+      int hw = (int)std::thread::hardware_concurrency();
+      int nt = std::max(1, std::min(hw / std::max(1, world_) - 2, 31));
+      if (const char* e = getenv("ADAPM_PASS_THREADS")) nt = std::max(1, atoi(e));
+      pass_pool_.want_threads = nt;
+    }
     log2ch_ = 0;
     while ((1 << log2ch_) < nch_) log2ch_++;
 
@@ -419,10 +428,15 @@ class Server {
       for (auto& t : threads) t.join();
     }
 
+    int want_threads = 0;  // set by Server ctor (cores / world, capped)
+
     void ensure_started() {
       if (!threads.empty()) return;
-      unsigned hw = std::thread::hardware_concurrency();
-      int nt = (int)std::min<unsigned>(hw > 2 ? hw - 2 : 1, 15);
+      int nt = want_threads;
+      if (nt <= 0) {
+        unsigned hw = std::thread::hardware_concurrency();
+        nt = (int)std::min<unsigned>(hw > 2 ? hw - 2 : 1, 15);
+      }
       for (int i = 0; i < nt; ++i)
         threads.emplace_back([this] {
           uint64_t seen = 0;
