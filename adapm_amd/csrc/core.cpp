@@ -449,17 +449,29 @@ class Server {
               seen = gen;
               t = task;
             }
-            work(t);
+            try {
+              work(t);
+            } catch (...) {
+              // pass bodies only allocate; an OOM here surfaces on the
+              // caller side as missing work is impossible (chunks the
+              // worker claimed are lost) — treat as fatal store failure
+              fprintf(stderr, "adapm: metadata-pass worker exception\n");
+            }
           }
         });
     }
 
     static void work(const Task& t) {
       constexpr int64_t STEP = 4;
-      for (;;) {
-        int64_t b = t.next->fetch_add(STEP, std::memory_order_relaxed);
-        if (b >= t.n) break;
-        (*t.fn)(b, std::min(t.n, b + STEP));
+      try {
+        for (;;) {
+          int64_t b = t.next->fetch_add(STEP, std::memory_order_relaxed);
+          if (b >= t.n) break;
+          (*t.fn)(b, std::min(t.n, b + STEP));
+        }
+      } catch (...) {
+        t.done->fetch_add(1, std::memory_order_acq_rel);
+        throw;  // caller rethrows its own; worker threads must not leak a hang
       }
       t.done->fetch_add(1, std::memory_order_acq_rel);
     }
@@ -480,9 +492,16 @@ class Server {
         task = Task{&fn, n, &next, &done, gen};
       }
       cv.notify_all();
-      work(task);  // caller participates
       int want = (int)threads.size() + 1;
+      std::exception_ptr err;
+      try {
+        work(task);  // caller participates
+      } catch (...) {
+        err = std::current_exception();  // work() already counted this participant
+      }
+      // workers reference next/done on this stack frame — always drain
       while (done.load(std::memory_order_acquire) < want) std::this_thread::yield();
+      if (err) std::rethrow_exception(err);
     }
   };
 
