@@ -413,6 +413,7 @@ class Server {
       uint64_t gen = 0;
     };
     std::vector<std::thread> threads;
+    std::mutex run_mu_;  // one run() at a time: task is a single slot
     std::mutex mu;
     std::condition_variable cv;
     Task task;
@@ -483,6 +484,9 @@ class Server {
         if (n == 1) fn(0, 1);
         return;
       }
+      // concurrent worker threads each bring their own pass; serialize
+      // (they would otherwise overwrite the single task slot and hang)
+      std::lock_guard<std::mutex> rg(run_mu_);
       ensure_started();
       std::atomic<int64_t> next{0};
       std::atomic<int> done{0};

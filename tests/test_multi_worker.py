@@ -91,3 +91,43 @@ def _multi_worker_dist(rank, world):
 
 def test_multi_worker_ws2():
     run_dist(2, _multi_worker_dist, timeout=300)
+
+
+def test_concurrent_slow_pass_threads():
+    """Concurrent worker threads on a NON-identity store (spill forces the
+    slow metadata pass) — the PassPool must serialize its task slot or a
+    caller hangs."""
+    import threading
+
+    import adapm_amd
+
+    adapm_amd._SETUP.clear()
+    adapm_amd.runtime._RUNTIME = None
+    N = 50_000
+    adapm_amd.setup(num_keys=N, num_threads=4, device="cpu",
+                    device_cap_gb=N * 32 * 4 * 0.5 / 2**30, host_spill_gb=0.5)
+    s = adapm_amd.Server(32)
+    workers = [adapm_amd.Worker(i, s) for i in range(4)]
+    vals = torch.ones(2000, 32)
+    errs = []
+
+    def hammer(w, seed):
+        try:
+            rng = np.random.default_rng(seed)
+            for _ in range(25):
+                ks = rng.choice(N, 2000, replace=False).astype(np.int64)
+                w.push(ks, vals)
+                out = torch.zeros(2000, 32)
+                w.pull(ks, out)
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+
+    ts = [threading.Thread(target=hammer, args=(w, i)) for i, w in enumerate(workers)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join(90)
+    assert not errs, errs
+    assert not any(t.is_alive() for t in ts), "a worker hung in the slow pass"
+    s.shutdown()
+    adapm_amd._SETUP.clear()
