@@ -2249,6 +2249,9 @@ class Server {
     TORCH_CHECK(world_ == 1, "rebalance_spill currently requires world==1");
     TORCH_CHECK(uniform_len_ >= 0, "rebalance_spill requires a uniform-length store");
     if (!heat_ || slab_.host_capacity == 0 || max_moves <= 0) return 0;
+    // one rebalance at a time (concurrent callers would interleave their
+    // stop-the-world windows and double-apply swaps)
+    std::lock_guard<std::mutex> rg(rebalance_mu_);
     const int32_t l = uniform_len_;
     rebalance_calls_++;
 
@@ -2561,6 +2564,7 @@ class Server {
   double sync_threshold_ = 0.0;                         // --sys.sync.threshold equivalent
   std::unique_ptr<std::atomic<uint32_t>[]> heat_;       // per-key access heat (spill stores)
   std::mutex spill_mu_;
+  std::mutex rebalance_mu_;
   std::vector<Key> spill_touched_;                      // spilled keys accessed since last rebalance
   int64_t rebalance_calls_ = 0;
   std::atomic<int64_t> stat_spill_moves_{0};
