@@ -3,11 +3,14 @@ ranks x multiple worker threads, torch + numpy tensors, intent, sampling.
 
 Run:  python -m adapm_amd.launch -n 4 examples/example.py
 """
+import os
+import sys
 import threading
 
 import numpy as np
 import torch
 
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import adapm_amd
 
 NUM_KEYS = 1000
