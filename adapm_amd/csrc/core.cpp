@@ -1991,8 +1991,6 @@ class Server {
     torch::Tensor out;
     HostBatch hb;
     int delivered = 0;
-    bool complete = false;
-    torch::Tensor caller_out;
     {
       std::lock_guard<std::mutex> g(tickets_mu_);
       auto it = tickets_.find(ts);
@@ -2006,11 +2004,12 @@ class Server {
         t.out_off[oi] = -1 - t.out_off[oi];
         delivered++;
       }
-      t.received += delivered;
-      if (t.received >= t.expected) {
-        complete = true;
-        caller_out = t.caller_out;
-      }
+      // received is NOT bumped here: a multi-channel pull gets one bulk
+      // response per channel, applied by different sync threads. If this
+      // response's rows counted before its copy ran, a racing response
+      // could reach expected, erase the ticket and wake the caller while
+      // our rows are still unwritten (observed as zero-pulls under heavy
+      // oversubscription).
     }
     if (hb.size()) {
       SlabBases pb{payload.data_ptr<float>(), nullptr};
@@ -2025,11 +2024,14 @@ class Server {
     if (use_loc_cache_) {
       for (int64_t i = 0; i < nk; ++i) loc_cache_[keys[i]] = src;
     }
-    if (complete) {
+    if (delivered) {
       std::lock_guard<std::mutex> g(tickets_mu_);
       auto it = tickets_.find(ts);
-      if (it != tickets_.end()) {
-        if (caller_out.defined()) caller_out.view({-1}).copy_(it->second->out.view({-1}));
+      if (it == tickets_.end()) return;
+      Ticket& t = *it->second;
+      t.received += delivered;
+      if (t.received >= t.expected) {
+        if (t.caller_out.defined()) t.caller_out.view({-1}).copy_(t.out.view({-1}));
         tickets_.erase(it);
         tickets_cv_.notify_all();
       }
@@ -2427,6 +2429,15 @@ class Server {
     }
   }
 
+  // debug/observability: raw metadata snapshot for one key
+  // (flags, slab offset, version, believed owner from the directory)
+  std::tuple<int, int64_t, int64_t, int> debug_key_state(int64_t k) {
+    TORCH_CHECK((uint64_t)k < (uint64_t)num_keys_, "key out of range");
+    int owner = -1;
+    if ((Key)(k % world_) == (Key)rank_) owner = owner_of_[k / world_];
+    return {(int)flags_[k].load(), loc_[k].load(), (int64_t)version_[k].load(), owner};
+  }
+
   void enable_key_trace(torch::Tensor keys) {
     std::lock_guard<std::mutex> g(trace_mu_);
     if (keys.numel() == 1 && keys.data_ptr<int64_t>()[0] == -1) {
@@ -2756,6 +2767,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("owner_hint", &Server::owner_hint)
       .def("enable_locality_stats", &Server::enable_locality_stats)
       .def("enable_key_trace", &Server::enable_key_trace)
+      .def("debug_key_state", &Server::debug_key_state)
       .def("dump_locality_stats", &Server::dump_locality_stats)
       .def("dump_traces", &Server::dump_traces)
       .def("stats", &Server::stats)
