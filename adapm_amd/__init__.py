@@ -227,6 +227,7 @@ class Worker:
         self.server = server
         self.wid = customer_id
         self._s = server._s
+        self._ulen = self._s.uniform_len()
 
     # ---- data ops (torch / numpy overloads, like bindings.cc:160-290)
 
@@ -242,9 +243,14 @@ class Worker:
         return vals, None
 
     def _check_len(self, kt, vt):
-        need = sum(self._s.get_len(int(k)) for k in kt) if kt.numel() < 1024 else None
-        if need is not None and need != vt.numel():
-            raise ValueError(f"value array has {vt.numel()} floats, keys need {need}")
+        """Always-on size validation (reference bindings.cc:174-186). The
+        uniform-length check is O(1) here and raises ValueError; per-key
+        length stores are validated by the C++ core on every call."""
+        if self._ulen >= 0:
+            need = kt.numel() * self._ulen
+            if need != vt.numel():
+                raise ValueError(f"value array has {vt.numel()} floats, "
+                                 f"{kt.numel()} key(s) need {need}")
 
     def pull(self, keys, vals, async_: bool = False, **kw):
         async_ = kw.get("async", async_)

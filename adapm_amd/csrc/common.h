@@ -52,6 +52,8 @@ enum MsgCode : int64_t {
   M_PULL_REQ_BULK = 4,   // hdr(code, nkeys, origin, req_id, hops) + keys + out_idx
   M_PUSH_REQ_BULK = 5,   // hdr(code, nkeys, origin, req_id, set<<32|hops) + keys. payload: nkeys rows
   M_PULL_RESP_BULK = 14, // hdr(code, nkeys, req_id, 0, 0) + keys + out_idx. payload: nkeys rows
+  M_NACK = 15,           // hop-limit give-up: f0=req_id, f1=count. The origin's
+                         // ticket fails loudly instead of hanging/acking a drop.
 };
 
 enum DeltaFlags : int64_t {

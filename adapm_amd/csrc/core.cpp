@@ -209,6 +209,7 @@ struct Ticket {
   torch::Tensor caller_out;      // caller's tensor if different device
   std::vector<int64_t> out_off;  // flat offsets per original key index
   std::vector<int32_t> out_len;
+  std::string fail;              // non-empty: some records were NACKed; wait() throws
 };
 
 // pending outgoing record (remote op or forward), per channel
@@ -257,6 +258,11 @@ struct ChannelState {
   std::unordered_map<Key, int64_t> reloc_round;  // round a key relocated IN (sync thread
                                                  // only): relocation cooldown, see below
   std::atomic<int64_t> rounds{0};
+  // strong-WaitSync bookkeeping: a round is "globally idle" when NO rank
+  // sent anything on this channel. idle2_events counts rounds that were
+  // the >=2nd consecutive globally-idle round (sync thread only writes).
+  int64_t idle_streak = 0;
+  std::atomic<int64_t> idle2_events{0};
 };
 
 // ---------------------------------------------------------------- server
@@ -668,9 +674,11 @@ class Server {
   // the caller's stream (reference coloc_kv_worker.h:253-318 contract).
   int64_t pull(int wid, torch::Tensor keys, torch::Tensor vals) {
     (void)wid;
+    check_not_failed();
     check_keys(keys);
     TORCH_CHECK(vals.is_contiguous() && vals.scalar_type() == torch::kFloat32,
                 "vals must be contiguous float32");
+    check_val_size(keys, vals.numel(), "pull");
     torch::Tensor vals_dev =
         vals.device() == dev_ ? vals
                               : torch::empty({vals.numel()},
@@ -826,8 +834,10 @@ class Server {
   // preserving delta semantics.
   int64_t push(int wid, torch::Tensor keys, torch::Tensor vals, bool set_mode) {
     (void)wid;
+    check_not_failed();
     check_keys(keys);
     TORCH_CHECK(vals.scalar_type() == torch::kFloat32, "vals must be float32");
+    check_val_size(keys, vals.numel(), "push");
     auto tick = [&]() { return cpp_timing_ ? std::chrono::steady_clock::now().time_since_epoch().count() : 0; };
     int64_t tp0 = tick();
     torch::Tensor vals_dev = vals.device() == dev_ ? vals : vals.to(dev_);
@@ -1004,6 +1014,7 @@ class Server {
   // PullIfLocal: all-or-nothing local pull (reference coloc_kv_worker.h)
   bool pull_if_local(torch::Tensor keys, torch::Tensor vals) {
     check_keys(keys);
+    check_val_size(keys, vals.numel(), "pull_if_local");
     int64_t n = keys.numel();
     const int64_t* kp = keys.data_ptr<int64_t>();
     torch::Tensor vals_dev =
@@ -1074,15 +1085,44 @@ class Server {
     return tickets_.find(ts) == tickets_.end();
   }
 
+  // ops issued after a transport failure raise instead of creating
+  // tickets that can never complete (the sync threads are dead)
+  inline void check_not_failed() {
+    if (failed_flag_.load(std::memory_order_relaxed)) {
+      std::lock_guard<std::mutex> g(tickets_mu_);
+      TORCH_CHECK(false, "adapm server failed: ", failed_reason_);
+    }
+  }
+
   void wait(int64_t ts) {
     if (ts < 0) return;
     std::unique_lock<std::mutex> g(tickets_mu_);
-    tickets_cv_.wait(g, [&] { return tickets_.find(ts) == tickets_.end(); });
+    tickets_cv_.wait(g, [&] {
+      return tickets_.find(ts) == tickets_.end() || failed_flag_.load(std::memory_order_relaxed);
+    });
+    TORCH_CHECK(!failed_flag_.load(std::memory_order_relaxed),
+                "adapm server failed while waiting: ", failed_reason_);
+    auto it = failed_tickets_.find(ts);
+    if (it != failed_tickets_.end()) {
+      std::string why = std::move(it->second);
+      failed_tickets_.erase(it);
+      TORCH_CHECK(false, "adapm op ", ts, " failed: ", why);
+    }
   }
 
   void wait_all() {
     std::unique_lock<std::mutex> g(tickets_mu_);
-    tickets_cv_.wait(g, [&] { return tickets_.empty(); });
+    tickets_cv_.wait(g, [&] {
+      return tickets_.empty() || failed_flag_.load(std::memory_order_relaxed);
+    });
+    TORCH_CHECK(!failed_flag_.load(std::memory_order_relaxed),
+                "adapm server failed while waiting: ", failed_reason_);
+    if (!failed_tickets_.empty()) {
+      std::string why = failed_tickets_.begin()->second;
+      size_t n = failed_tickets_.size();
+      failed_tickets_.clear();
+      TORCH_CHECK(false, "adapm: ", n, " async op(s) failed, first: ", why);
+    }
   }
 
   // WaitSync support (reference coloc_kv_worker.h:517-550): callers grab
@@ -1095,10 +1135,40 @@ class Server {
   void wait_rounds(std::vector<int64_t> targets) {
     std::unique_lock<std::mutex> g(rounds_mu_);
     rounds_cv_.wait(g, [&] {
+      if (failed_flag_.load(std::memory_order_relaxed)) return true;
       for (int c = 0; c < nch_; ++c)
         if (channels_[c].rounds.load() < targets[c]) return false;
       return true;
     });
+    if (failed_flag_.load(std::memory_order_relaxed)) {
+      std::lock_guard<std::mutex> tg(tickets_mu_);
+      TORCH_CHECK(false, "adapm server failed: ", failed_reason_);
+    }
+  }
+
+  // Strong WaitSync (globally-idle rounds): wait until each channel has
+  // completed a round that was the 2nd consecutive round in which NO rank
+  // had outbound traffic — at that point every in-flight delta, forward
+  // and refresh has drained, so the caller observes all prior pushes
+  // (stronger than the reference's fixed 2-round window, which forwarded
+  // deltas can escape under relocation churn).
+  std::vector<int64_t> idle_counts() {
+    std::vector<int64_t> v;
+    for (auto& c : channels_) v.push_back(c.idle2_events.load());
+    return v;
+  }
+  void wait_idle(std::vector<int64_t> targets) {
+    std::unique_lock<std::mutex> g(rounds_mu_);
+    rounds_cv_.wait(g, [&] {
+      if (failed_flag_.load(std::memory_order_relaxed)) return true;
+      for (int c = 0; c < nch_; ++c)
+        if (channels_[c].idle2_events.load() < targets[c]) return false;
+      return true;
+    });
+    if (failed_flag_.load(std::memory_order_relaxed)) {
+      std::lock_guard<std::mutex> tg(tickets_mu_);
+      TORCH_CHECK(false, "adapm server failed: ", failed_reason_);
+    }
   }
 
   // ------------------------------------------------ sync round: phase A out
@@ -1467,18 +1537,34 @@ class Server {
     }
   }
 
-  // re-enqueue a record whose believed destination was wrong; bounded hops
+  // re-enqueue a record whose believed destination was wrong; bounded hops.
+  // At the hop cap (64 — a directory pathology, not normal churn):
+  //  - replica DELTAS are never dropped (that would silently lose pushed
+  //    updates; reference never drops — addressbook routing converges at
+  //    the manager). They keep retrying toward the manager, counted in
+  //    stats as overhops.
+  //  - ticketed Push/Set/Pull requests fail LOUDLY: the origin's ticket is
+  //    marked failed (wait(ts) throws) instead of acking a dropped op.
   void requeue_bounded(int ch, const OutRec& r, int hops_field) {
     OutRec nr = r;
     int64_t* hf = hops_field == 1 ? &nr.f1 : &nr.f2;
     int hops = (int)(*hf >> 32);
-    if (hops >= 64) {
-      // give up: complete tickets so callers don't hang; value lost is
-      // impossible here (ownership exists somewhere; 16 hops means a
-      // directory pathologie) — log via stats.
+    if (hops >= max_hops_) {
+      if (r.code == M_DELTA) {
+        stat_delta_overhops_ += 1;  // keep chasing; hops stays at the cap
+        nr.dest = manager_of(nr.key);
+        enqueue_out(ch, std::move(nr));
+        return;
+      }
       stat_dropped_records_ += 1;
-      if ((int)r.f0 == rank_ && (r.code == M_PUSH_REQ || r.code == M_SET_REQ))
-        complete_ticket(r.f1, 1);
+      static const char* why = "remote op dropped at hop limit (directory pathology)";
+      if ((int)r.f0 == rank_) {
+        fail_ticket(r.f1, 1, why);
+      } else {
+        std::lock_guard<std::mutex> g(channels_[ch].mu);
+        channels_[ch].responses.push_back(
+            RespRec{(int)r.f0, M_NACK, r.key, r.f1, 1, 0, -1, 0, false, {}});
+      }
       return;
     }
     *hf = (*hf & 0xffffffff) | ((int64_t)(hops + 1) << 32);
@@ -1517,8 +1603,15 @@ class Server {
       C.responses.push_back(std::move(resp));
     }
     for (auto& [d, g] : fwd) {
-      if (hops >= 64) {
+      if (hops >= max_hops_) {
         stat_dropped_records_ += (int64_t)g.first.size();
+        if (origin == rank_) {
+          fail_ticket(req_id, (int)g.first.size(), "remote pull dropped at hop limit");
+        } else {
+          std::lock_guard<std::mutex> gm(C.mu);
+          C.responses.push_back(RespRec{origin, M_NACK, 0, req_id,
+                                        (int64_t)g.first.size(), 0, -1, 0, false, {}});
+        }
         continue;
       }
       stat_forwards_ += (int64_t)g.first.size();
@@ -1566,9 +1659,15 @@ class Server {
     }
     auto rows = rows_flat.view({nk, (int64_t)l});
     for (auto& [d, g] : fwd) {
-      if (hops >= 64) {
+      if (hops >= max_hops_) {
         stat_dropped_records_ += (int64_t)g.first.size();
-        if (origin == rank_) complete_ticket(req_id, (int)g.first.size());
+        if (origin == rank_) {
+          fail_ticket(req_id, (int)g.first.size(), "remote push dropped at hop limit");
+        } else {
+          std::lock_guard<std::mutex> gm(C.mu);
+          C.responses.push_back(RespRec{origin, M_NACK, 0, req_id,
+                                        (int64_t)g.first.size(), 0, -1, 0, false, {}});
+        }
         continue;
       }
       stat_forwards_ += (int64_t)g.first.size();
@@ -1939,6 +2038,10 @@ class Server {
           complete_ticket(f0, (int)f1);
           break;
         }
+        case M_NACK: {
+          fail_ticket(f0, (int)f1, "remote op dropped at hop limit (directory pathology)");
+          break;
+        }
         case M_RESIDENCE: {
           apply_residence(k, (int)f0, (uint32_t)f1);
           break;
@@ -2032,6 +2135,7 @@ class Server {
       t.received += delivered;
       if (t.received >= t.expected) {
         if (t.caller_out.defined()) t.caller_out.view({-1}).copy_(t.out.view({-1}));
+        if (!t.fail.empty()) failed_tickets_[ts] = std::move(t.fail);
         tickets_.erase(it);
         tickets_cv_.notify_all();
       }
@@ -2051,6 +2155,7 @@ class Server {
     t.received++;
     if (t.received >= t.expected) {
       if (t.caller_out.defined()) t.caller_out.view({-1}).copy_(t.out.view({-1}));
+      if (!t.fail.empty()) failed_tickets_[ts] = std::move(t.fail);
       tickets_.erase(it);
       tickets_cv_.notify_all();
     }
@@ -2062,6 +2167,25 @@ class Server {
     if (it == tickets_.end()) return;
     it->second->received += cnt;
     if (it->second->received >= it->second->expected) {
+      if (!it->second->fail.empty()) failed_tickets_[ts] = std::move(it->second->fail);
+      tickets_.erase(it);
+      tickets_cv_.notify_all();
+    }
+  }
+
+  // hop-limit give-up: count the records toward the ticket (so it
+  // completes) but mark it failed — wait(ts) throws instead of
+  // pretending the dropped op succeeded (reference never drops:
+  // addressbook.h routing always converges at the manager).
+  void fail_ticket(int64_t ts, int cnt, const char* why) {
+    std::lock_guard<std::mutex> g(tickets_mu_);
+    auto it = tickets_.find(ts);
+    if (it == tickets_.end()) return;
+    Ticket& t = *it->second;
+    t.received += cnt;
+    if (t.fail.empty()) t.fail = why;
+    if (t.received >= t.expected) {
+      failed_tickets_[ts] = std::move(t.fail);
       tickets_.erase(it);
       tickets_cv_.notify_all();
     }
@@ -2075,6 +2199,7 @@ class Server {
     {
       std::lock_guard<std::mutex> g(tickets_mu_);
       failed_reason_ = std::move(reason);
+      failed_flag_.store(true, std::memory_order_relaxed);
       tickets_.clear();
     }
     tickets_cv_.notify_all();
@@ -2085,10 +2210,16 @@ class Server {
     return failed_reason_;
   }
 
-  void sync_finish(int ch) {
+  void sync_finish(int ch, bool globally_idle = false) {
     {
       std::lock_guard<std::mutex> g(rounds_mu_);
-      channels_[ch].rounds++;
+      ChannelState& C = channels_[ch];
+      C.rounds++;
+      if (globally_idle) {
+        if (++C.idle_streak >= 2) C.idle2_events++;
+      } else {
+        C.idle_streak = 0;
+      }
     }
     rounds_cv_.notify_all();
   }
@@ -2413,6 +2544,7 @@ class Server {
   // ------------------------------------------------ info / stats
 
   int64_t get_len(Key k) { return len_of(k); }
+  int64_t uniform_len() const { return uniform_len_; }  // -1 if per-key lengths
   int64_t num_keys() const { return num_keys_; }
   int rank() const { return rank_; }
   int world() const { return world_; }
@@ -2427,6 +2559,13 @@ class Server {
       key_accesses_[i].store(0);
       key_local_[i].store(0);
     }
+  }
+
+  // test hook: poison the location cache so a request routes to a wrong
+  // destination and exercises the forward/NACK machinery deterministically
+  void debug_set_loc_cache(int64_t k, int r) {
+    TORCH_CHECK(use_loc_cache_ && (uint64_t)k < (uint64_t)num_keys_);
+    loc_cache_[k] = r;
   }
 
   // debug/observability: raw metadata snapshot for one key
@@ -2496,6 +2635,7 @@ class Server {
     d["replica_drops"] = stat_drops_.load();
     d["forwards"] = stat_forwards_.load();
     d["dropped_records"] = stat_dropped_records_.load();
+    d["delta_overhops"] = stat_delta_overhops_.load();
     d["bytes_sent"] = stat_bytes_sent_.load();
     d["bytes_recv"] = stat_bytes_recv_.load();
     d["sampling_checks"] = stat_sampling_checks_.load();
@@ -2527,6 +2667,26 @@ class Server {
   void check_keys(const torch::Tensor& keys) {
     TORCH_CHECK(keys.device().is_cpu() && keys.scalar_type() == torch::kInt64 && keys.is_contiguous(),
                 "keys must be a contiguous CPU int64 tensor");
+  }
+
+  // always-on value-size validation (reference bindings.cc:174-186
+  // validates every call; a mis-sized vals tensor must be a Python
+  // error, never kernel UB)
+  void check_val_size(const torch::Tensor& keys, int64_t have, const char* op) {
+    int64_t need;
+    int64_t n = keys.numel();
+    if (uniform_len_ >= 0) {
+      need = n * (int64_t)uniform_len_;
+    } else {
+      need = 0;
+      const int64_t* kp = keys.data_ptr<int64_t>();
+      for (int64_t i = 0; i < n; ++i) {
+        TORCH_CHECK((uint64_t)kp[i] < (uint64_t)num_keys_, "key out of range: ", kp[i]);
+        need += lens_[kp[i]];
+      }
+    }
+    TORCH_CHECK(have == need, op, ": value tensor has ", have, " floats but ", n,
+                " key(s) need ", need);
   }
 
   int64_t num_keys_;
@@ -2570,6 +2730,8 @@ class Server {
   std::atomic<bool> layout_identity_{true};
 
   std::atomic<int> inflight_{0};
+  // forward-hop cap before a request NACKs (test hook: ADAPM_MAX_HOPS)
+  int max_hops_ = getenv("ADAPM_MAX_HOPS") ? atoi(getenv("ADAPM_MAX_HOPS")) : 64;
   PassPool pass_pool_;
   std::atomic<int> migrating_{0};                       // spill-rebalance stop-the-world gate
   double sync_threshold_ = 0.0;                         // --sys.sync.threshold equivalent
@@ -2588,7 +2750,9 @@ class Server {
   std::mutex tickets_mu_;
   std::condition_variable tickets_cv_;
   std::unordered_map<int64_t, std::unique_ptr<Ticket>> tickets_;
+  std::unordered_map<int64_t, std::string> failed_tickets_;  // completed-but-failed: wait() throws
   std::string failed_reason_;
+  std::atomic<bool> failed_flag_{false};
   std::mutex rounds_mu_;
   std::condition_variable rounds_cv_;
 
@@ -2607,7 +2771,8 @@ class Server {
       stat_pull_local_{0}, stat_push_local_{0}, stat_pull_replica_{0}, stat_push_replica_{0},
       stat_remote_pulls_served_{0}, stat_remote_pushes_served_{0}, stat_relocations_{0},
       stat_relocated_in_{0}, stat_replications_{0}, stat_drops_{0}, stat_forwards_{0},
-      stat_dropped_records_{0}, stat_bytes_sent_{0}, stat_bytes_recv_{0}, stat_sampling_checks_{0};
+      stat_dropped_records_{0}, stat_delta_overhops_{0}, stat_bytes_sent_{0},
+      stat_bytes_recv_{0}, stat_sampling_checks_{0};
 };
 
 // ------------------------------------------------------- app kernel wrappers
@@ -2744,12 +2909,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("is_finished", &Server::is_finished)
       .def("round_counts", &Server::round_counts)
       .def("wait_rounds", &Server::wait_rounds, py::call_guard<py::gil_scoped_release>())
+      .def("idle_counts", &Server::idle_counts)
+      .def("wait_idle", &Server::wait_idle, py::call_guard<py::gil_scoped_release>())
       .def("set_intent_ahead", &Server::set_intent_ahead)
       .def("sync_collect", &Server::sync_collect, py::call_guard<py::gil_scoped_release>())
       .def("sync_process", &Server::sync_process, py::call_guard<py::gil_scoped_release>())
       .def("sync_respond", &Server::sync_respond, py::call_guard<py::gil_scoped_release>())
       .def("sync_apply", &Server::sync_apply, py::call_guard<py::gil_scoped_release>())
-      .def("sync_finish", &Server::sync_finish, py::call_guard<py::gil_scoped_release>())
+      .def("sync_finish", &Server::sync_finish, py::arg("ch"),
+           py::arg("globally_idle") = false, py::call_guard<py::gil_scoped_release>())
       .def("fail", &Server::fail, py::call_guard<py::gil_scoped_release>())
       .def("failed_reason", &Server::failed_reason)
       .def("scan_local", &Server::scan_local, py::call_guard<py::gil_scoped_release>())
@@ -2760,6 +2928,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
            py::call_guard<py::gil_scoped_release>())
       .def("key_tier", &Server::key_tier)
       .def("get_len", &Server::get_len)
+      .def("uniform_len", &Server::uniform_len)
       .def("num_keys", &Server::num_keys)
       .def("rank", &Server::rank)
       .def("world", &Server::world)
@@ -2768,6 +2937,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("enable_locality_stats", &Server::enable_locality_stats)
       .def("enable_key_trace", &Server::enable_key_trace)
       .def("debug_key_state", &Server::debug_key_state)
+      .def("debug_set_loc_cache", &Server::debug_set_loc_cache)
       .def("dump_locality_stats", &Server::dump_locality_stats)
       .def("dump_traces", &Server::dump_traces)
       .def("stats", &Server::stats)
