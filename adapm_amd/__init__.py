@@ -187,13 +187,32 @@ class Server:
     def stats(self) -> dict:
         return dict(self._s.stats())
 
-    def wait_sync(self):
+    def wait_sync(self, strong: bool = False):
         """Block until 2 more sync rounds completed on every channel
-        (reference WaitSync, coloc_kv_worker.h:517-550)."""
+        (reference WaitSync, coloc_kv_worker.h:517-550).
+
+        strong=True waits instead for a *globally idle* point: on every
+        channel, 2 consecutive rounds in which no rank sent or
+        self-handled anything. Forwarded deltas can escape the fixed
+        2-round window under relocation churn (the reference has the
+        same hazard); after a globally idle point every in-flight
+        delta/forward/refresh has drained, so all prior pushes are
+        visible. Unbounded if other ranks push continuously — use after
+        a barrier. (With sync_threshold > 0, deltas below the threshold
+        are withheld by design and not covered.)"""
         if self.rt.world <= 1:
             return
-        counts = self._s.round_counts()
-        self._s.wait_rounds([c + 2 for c in counts])
+        if strong:
+            # +2, not +1: the round already in flight at call time was
+            # collected before our ops landed and may still complete as
+            # "idle"; only one round can be mid-flight (single comm
+            # thread), so the second idle event is necessarily observed
+            # by a round that saw our work.
+            counts = self._s.idle_counts()
+            self._s.wait_idle([c + 2 for c in counts])
+        else:
+            counts = self._s.round_counts()
+            self._s.wait_rounds([c + 2 for c in counts])
 
     @property
     def raw(self):
@@ -347,8 +366,8 @@ class Worker:
     def is_finished(self, ts: int) -> bool:
         return self._s.is_finished(ts)
 
-    def wait_sync(self):
-        self.server.wait_sync()
+    def wait_sync(self, strong: bool = False):
+        self.server.wait_sync(strong=strong)
 
     def wait_replica_sync(self):  # deprecated alias (reference bindings.cc:355)
         import warnings
@@ -399,7 +418,10 @@ class Worker:
 
     def allreduce(self, value, op: str = "sum"):
         """Reduce a scalar or tensor across ranks (loss/eval aggregation;
-        replaces reference utils.h ps_allreduce). op: sum | max | min."""
+        replaces reference utils.h ps_allreduce). op: sum | max | min.
+        Rides the gloo control-plane group (host TCP): worker threads may
+        call this concurrently with sync rounds, and NCCL traffic from
+        two threads on one device can deadlock — gloo cannot."""
         rt = self.server.rt
         scalar = not isinstance(value, torch.Tensor)
         t = torch.tensor([float(value)]) if scalar else value
@@ -408,8 +430,7 @@ class Worker:
 
             ops = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX,
                    "min": dist.ReduceOp.MIN}
-            dev = rt.device if rt.backend == "nccl" else torch.device("cpu")
-            td = t.to(dev)
+            td = t.to("cpu")
             with rt.worker_group_lock:
                 dist.all_reduce(td, op=ops[op], group=rt.worker_group)
             t = td.to(t.device)

@@ -2211,6 +2211,13 @@ class Server {
   }
 
   void sync_finish(int ch, bool globally_idle = false) {
+    if (globally_idle) {
+      // self-handled records can requeue work invisibly to peers; a
+      // round with anything still queued locally is not idle
+      std::lock_guard<std::mutex> g(channels_[ch].mu);
+      if (!channels_[ch].out_queue.empty() || !channels_[ch].responses.empty())
+        globally_idle = false;
+    }
     {
       std::lock_guard<std::mutex> g(rounds_mu_);
       ChannelState& C = channels_[ch];
@@ -2222,6 +2229,26 @@ class Server {
       }
     }
     rounds_cv_.notify_all();
+  }
+
+  // watchdog/observability: counts of everything still in flight
+  py::dict debug_pending() {
+    py::dict d;
+    {
+      std::lock_guard<std::mutex> g(tickets_mu_);
+      d["tickets"] = tickets_.size();
+    }
+    py::list outq, respq, rounds;
+    for (auto& C : channels_) {
+      std::lock_guard<std::mutex> g(C.mu);
+      outq.append(C.out_queue.size());
+      respq.append(C.responses.size());
+      rounds.append(C.rounds.load());
+    }
+    d["out_queues"] = outq;
+    d["responses"] = respq;
+    d["rounds"] = rounds;
+    return d;
   }
 
   // ------------------------------------------------ fused app operator
@@ -2938,6 +2965,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("enable_key_trace", &Server::enable_key_trace)
       .def("debug_key_state", &Server::debug_key_state)
       .def("debug_set_loc_cache", &Server::debug_set_loc_cache)
+      .def("debug_pending", &Server::debug_pending)
       .def("dump_locality_stats", &Server::dump_locality_stats)
       .def("dump_traces", &Server::dump_traces)
       .def("stats", &Server::stats)

@@ -36,7 +36,16 @@ class Runtime:
     device: torch.device
     backend: str
     num_channels: int
-    channel_groups: List[object] = field(default_factory=list)
+    # ONE process group for the whole sync engine, driven by ONE comm
+    # thread per rank. With NCCL, concurrent collectives on multiple
+    # communicators sharing a device can deadlock unless every rank
+    # issues them in the same global order; a single group + single
+    # issuing thread makes the order deterministic by construction.
+    sync_group: Optional[object] = None
+    # Worker-side barrier/allreduce ride a SEPARATE gloo group (host TCP):
+    # tiny control-plane collectives that must never interleave with the
+    # sync engine's NCCL traffic (they are called from worker threads,
+    # concurrently with sync rounds).
     worker_group: Optional[object] = None
     worker_group_lock: threading.Lock = field(default_factory=threading.Lock)
 
@@ -79,14 +88,14 @@ def init_runtime(num_channels: int = 2, device: str | None = None,
                 backend=backend, rank=rank, world_size=world,
                 timeout=datetime.timedelta(seconds=timeout_s),
             )
-        ch_groups = [dist.new_group(backend=backend) for _ in range(num_channels)]
-        worker_group = dist.new_group(backend=backend)
+        sync_group = dist.new_group(backend=backend)
+        worker_group = dist.new_group(backend="gloo")
     else:
-        ch_groups = [None] * num_channels
+        sync_group = None
         worker_group = None
 
     _RUNTIME = Runtime(rank=rank, world=world, device=dev, backend=backend,
-                       num_channels=num_channels, channel_groups=ch_groups,
+                       num_channels=num_channels, sync_group=sync_group,
                        worker_group=worker_group)
     return _RUNTIME
 

@@ -1,39 +1,52 @@
-"""Per-channel sync loop: the RCCL-over-xGMI replacement of the reference
+"""The sync engine: RCCL-over-xGMI replacement of the reference
 SyncManager (reference include/ps/sync_manager.h).
 
-Every channel runs one thread on every rank, in lockstep rounds:
+ONE comm thread per rank drives ALL channels in lockstep "superrounds":
 
-  phase A: server.sync_collect(ch)  -> per-dest {replica deltas, replica
-           requests, remote Pull/Push/Set requests, forwards}
-           size all-gather  +  batched P2P all-to-all-v
-           server.sync_process(ch, src, ...) for each incoming message
-  phase B: server.sync_respond(ch)  -> per-dest {refreshes, relocations,
-           pull responses, push acks, residence updates}
-           size all-gather  +  batched P2P all-to-all-v
-           server.sync_apply(ch, src, ...)
-  server.sync_finish(ch)
+  phase A: for every channel, server.sync_collect(ch) -> per-dest
+           {replica deltas, replica requests, remote Pull/Push/Set
+           requests, forwards}; ONE size all-gather covering all
+           channels + ONE batched P2P all-to-all-v; then
+           server.sync_process(ch, src, ...) per incoming message.
+  phase B: same with server.sync_respond(ch) -> {refreshes, relocations,
+           pull responses, push acks, NACKs, residence updates} and
+           server.sync_apply(ch, src, ...).
+  server.sync_finish(ch, globally_idle) per channel.
 
-On GPU the P2P exchange is dist.batch_isend_irecv = grouped ncclSend/Recv
-over the xGMI point-to-point links (per-destination batching, exactly the
-per-destination message builds of the reference, sync_manager.h:305,361 —
-but without rings, since xGMI is per-link P2P). On CPU it is gloo over
-loopback, which is the multi-process no-GPU test tier.
+Why one thread and one process group (round-1 ran a thread per channel on
+its own group): with NCCL, concurrent collectives on multiple
+communicators sharing one device deadlock unless every rank issues them
+in the same global order. A single comm thread iterating channels in a
+fixed order on a single communicator makes the issuance order identical
+on every rank by construction — and merging the channels' size
+all-gathers into one cuts per-round collective count. Worker-side
+barrier/allreduce ride a separate *gloo* group (host TCP), so they can
+never interleave with the engine's NCCL traffic. Channels remain the
+unit of key partitioning and per-channel protocol state (reference
+--sys.channels), they just share the transport round.
 
-Rounds are collective, so they run in lockstep on every rank; shutdown is
-coordinated through a flag column in the size matrix.
+On GPU the P2P exchange is dist.batch_isend_irecv = grouped
+ncclSend/Recv over the xGMI point-to-point links (per-destination
+batching, matching the reference's per-destination message builds,
+sync_manager.h:305,361 — no rings: xGMI is per-link P2P). Staging
+copies and the P2P ops run on a dedicated comm stream, event-ordered
+against the default stream where the store kernels run, so transfers
+overlap worker compute. On CPU it is gloo over loopback (the
+multi-process no-GPU test tier).
 """
 from __future__ import annotations
 
 import math
 import os
+import sys
 import threading
 import time
+from collections import defaultdict
 
 import torch
 import torch.distributed as dist
 
 _TRACE_KEY = int(os.environ.get("ADAPM_TRACE_KEY", "-1"))
-
 
 _BULK_CODES = {4: 2, 5: 1, 14: 2}  # code -> extra words per key
 
@@ -85,7 +98,7 @@ class ActionTimer:
 
 
 class SyncManager:
-    """Drives the sync loop threads for all channels of one rank."""
+    """Drives the single comm thread (all channels) of one rank."""
 
     def __init__(self, server, runtime, max_per_sec: float = 1000.0,
                  time_intent_actions: bool = True):
@@ -96,29 +109,31 @@ class SyncManager:
         self.stop_requested = threading.Event()
         self.kick_event = threading.Event()
         self.failed = False
-        from collections import defaultdict
-
         self.phase_totals = defaultdict(float)
         self.threads = []
         self.timer = ActionTimer()
+        self.watchdog_s = float(os.environ.get("ADAPM_WATCHDOG_S", "120"))
+        self._progress = (0, time.monotonic())  # (total rounds, when it last moved)
         if not time_intent_actions:
             server.set_intent_ahead(1 << 40)
 
     def start(self):
         if self.rt.world <= 1:
             return  # single rank: nothing to sync (reference sync_manager.h:454-457)
-        for ch in range(self.rt.num_channels):
-            t = threading.Thread(target=self._loop, args=(ch,), daemon=True,
-                                 name=f"adapm-sync-ch{ch}")
-            t.start()
-            self.threads.append(t)
+        t = threading.Thread(target=self._run, daemon=True, name="adapm-sync")
+        t.start()
+        self.threads.append(t)
+        if self.watchdog_s > 0:
+            w = threading.Thread(target=self._watchdog, daemon=True, name="adapm-watchdog")
+            w.start()
+            # not joined: daemon; it exits when stop is requested
 
     def request_stop(self):
         self.stop_requested.set()
         self.kick_event.set()
 
     def kick(self):
-        """Wake the sync loops early: a worker enqueued remote ops and is
+        """Wake the comm thread early: a worker enqueued remote ops and is
         (or will be) waiting on the round. All ranks under symmetric load
         kick at similar times, so the collective rounds speed up together;
         an early kicker just reaches the size all-gather sooner."""
@@ -128,9 +143,6 @@ class SyncManager:
         for t in self.threads:
             t.join()
         self.threads = []
-        import os
-        import sys
-
         if os.environ.get("ADAPM_VERBOSE", "0") != "0" and self.phase_totals.get("rounds"):
             pt = dict(self.phase_totals)
             n = pt.pop("rounds")
@@ -138,44 +150,78 @@ class SyncManager:
                   ", ".join(f"{k}={1000*v/n:.2f}" for k, v in sorted(pt.items())),
                   file=sys.stderr, flush=True)
 
+    # ------------------------------------------------------------ watchdog
+
+    def _watchdog(self):
+        """Dump per-channel protocol state if rounds stop advancing while
+        work is pending — a stalled collective should leave evidence, not
+        a silently hung lease."""
+        interval = min(10.0, self.watchdog_s / 4)
+        while not self.stop_requested.wait(timeout=interval):
+            if self.failed or not self.threads:
+                return
+            try:
+                pend = self.server.debug_pending()
+            except Exception:
+                return
+            total = sum(pend["rounds"])
+            last_total, last_t = self._progress
+            if total != last_total:
+                self._progress = (total, time.monotonic())
+                continue
+            stalled_for = time.monotonic() - last_t
+            busy = pend["tickets"] or any(pend["out_queues"]) or any(pend["responses"])
+            if stalled_for > self.watchdog_s and busy:
+                print(f"[adapm WATCHDOG r{self.rt.rank}] sync rounds stalled "
+                      f"{stalled_for:.0f}s with work pending: rounds={pend['rounds']} "
+                      f"tickets={pend['tickets']} out={pend['out_queues']} "
+                      f"resp={pend['responses']}", file=sys.stderr, flush=True)
+                self._progress = (total, time.monotonic())  # rate-limit the dump
+
     # ---------------------------------------------------------------- loop
 
-    def _loop(self, ch: int):
+    def _run(self):
         try:
-            self._loop_inner(ch)
+            self._run_inner()
         except Exception as e:  # transport failure: a peer likely died
-            import sys
-
-            print(f"[adapm] rank {self.rt.rank}: sync channel {ch} failed: "
+            print(f"[adapm] rank {self.rt.rank}: sync engine failed: "
                   f"{type(e).__name__}: {e}", file=sys.stderr, flush=True)
             self.failed = True
-            self.server.fail(f"sync channel {ch}: {e}")
+            self.server.fail(f"sync engine: {e}")
 
-    def _loop_inner(self, ch: int):
+    def _run_inner(self):
         rt = self.rt
-        group = rt.channel_groups[ch]
-        world, rank = rt.world, rt.rank
-        dev = rt.device if rt.backend == "nccl" else torch.device("cpu")
+        nch = rt.num_channels
+        # comm tensors live on the backend's device: GPU for NCCL (xGMI
+        # P2P), CPU for gloo
+        comm_dev = rt.device if rt.backend == "nccl" else torch.device("cpu")
+        use_streams = rt.backend == "nccl" and rt.is_cuda
         if rt.is_cuda:
             torch.cuda.set_device(rt.device)
+        if use_streams:
+            self._comm_stream = torch.cuda.Stream(rt.device)
+            self._ev_fwd = torch.cuda.Event()
+            self._ev_bwd = torch.cuda.Event()
+        else:
+            self._comm_stream = None
 
-        verbose = __import__("os").environ.get("ADAPM_VERBOSE", "0") != "0"
+        verbose = os.environ.get("ADAPM_VERBOSE", "0") != "0"
         last_report = time.monotonic()
         rounds_at_report = 0
         n_rounds = 0
         while True:
             t0 = time.monotonic()
-            if verbose and ch == 0 and t0 - last_report > 10.0:
+            if verbose and t0 - last_report > 10.0:
                 clocks = self.server.worker_clocks()
-                print(f"[adapm sync r{rank} ch{ch}] {(n_rounds - rounds_at_report) / (t0 - last_report):.0f} rounds/s, "
+                print(f"[adapm sync r{rt.rank}] "
+                      f"{(n_rounds - rounds_at_report) / (t0 - last_report):.0f} rounds/s, "
                       f"worker clocks {clocks}", flush=True)
                 last_report, rounds_at_report = t0, n_rounds
-            if ch == 0 and self.time_intent_actions:
+            if self.time_intent_actions:
                 self.server.set_intent_ahead(self.timer.update(self.server.worker_clocks()))
 
             stop = self.stop_requested.is_set()
-            all_stopped, any_work = self._round(ch, group, world, rank, dev, stop)
-            self.server.sync_finish(ch)
+            all_stopped, any_work = self._superround(nch, comm_dev, stop)
             n_rounds += 1
             if all_stopped:
                 return
@@ -187,84 +233,146 @@ class SyncManager:
                 self.kick_event.wait(timeout=period - dt)
                 self.kick_event.clear()
 
-    def _round(self, ch, group, world, rank, dev, stop_flag):
+    def _superround(self, nch, comm_dev, stop_flag):
+        s = self.server
         t0 = time.perf_counter()
-        out_a = self.server.sync_collect(ch)
+        outs_a = [s.sync_collect(ch) for ch in range(nch)]
         t1 = time.perf_counter()
-        all_stopped, work_a = self._exchange(ch, group, world, rank, dev, out_a,
-                                             self.server.sync_process, stop_flag)
+        all_stopped, st_a = self._exchange(nch, comm_dev, outs_a, s.sync_process, stop_flag)
         t2 = time.perf_counter()
-        out_b = self.server.sync_respond(ch)
+        outs_b = [s.sync_respond(ch) for ch in range(nch)]
         t3 = time.perf_counter()
-        _, work_b = self._exchange(ch, group, world, rank, dev, out_b,
-                                   self.server.sync_apply, stop_flag)
+        _, st_b = self._exchange(nch, comm_dev, outs_b, s.sync_apply, stop_flag)
         t4 = time.perf_counter()
+        any_work = False
+        for ch in range(nch):
+            a_meta, a_pay, a_local = st_a[ch]
+            b_meta, b_pay, b_local = st_b[ch]
+            # "Globally idle" for strong WaitSync = no DATA moved on this
+            # channel anywhere: phase A carried no payload (payload-less
+            # per-replica poll records are steady-state — every replica
+            # announces its version each round, like the reference's
+            # per-replica sync messages — and move no data) and phase B
+            # carried nothing at all (any refresh/ack/response resets the
+            # streak, so an in-flight request's answer keeps the round
+            # busy). Every data-carrying record contributes payload on
+            # every hop it travels, so an in-flight update always marks
+            # its round non-idle.
+            ch_idle = a_pay == 0 and a_local == 0 and b_meta == 0 and b_local == 0
+            any_work = any_work or a_meta or a_pay or a_local or b_meta or b_local
+            s.sync_finish(ch, ch_idle)
         pt = self.phase_totals
         pt["collect"] += t1 - t0
         pt["exchange_a"] += t2 - t1
         pt["respond"] += t3 - t2
         pt["exchange_b"] += t4 - t3
         pt["rounds"] += 1
-        return all_stopped, (work_a or work_b)
+        return all_stopped, bool(any_work)
 
-    def _exchange(self, ch, group, world, rank, dev, outgoing, handler, stop_flag) -> bool:
-        # size matrix: row = this rank's (n_meta_i64, n_payload_f32) per dest
-        # + one stop flag in the last column of dest 0's slot
-        sizes = torch.zeros(world, 3, dtype=torch.int64)
+    def _exchange(self, nch, comm_dev, outs_per_ch, handler, stop_flag):
+        """One phase for ALL channels: a single size all-gather + a single
+        grouped P2P all-to-all-v, then handlers in deterministic order.
+        Returns (all_stopped, per-channel (meta, payload, local) totals
+        summed over all rank pairs — the superround derives idleness)."""
+        rt = self.rt
+        world, rank = rt.world, rt.rank
+        group = rt.sync_group
+        use_streams = self._comm_stream is not None
+
+        # size matrix row (per rank): nch x (n_meta, n_payload, local_work)
+        # + one stop flag at the end
+        sizes = torch.zeros(world * nch * 3 + 1, dtype=torch.int64)
         msgs = {}
-        for dest, meta, payload in outgoing:
-            _trace(rank, ch, "out", dest, meta)
-            if dest == rank:
-                handler(ch, rank, meta, payload)
-                continue
-            sizes[dest, 0] = meta.numel()
-            sizes[dest, 1] = payload.numel()
-            msgs[dest] = (meta, payload)
-        sizes[:, 2] = 1 if stop_flag else 0
+        local = []  # (ch, meta, payload) handled without transport
+        for ch, outs in enumerate(outs_per_ch):
+            for dest, meta, payload in outs:
+                _trace(rank, ch, "out", dest, meta)
+                if dest == rank:
+                    local.append((ch, meta, payload))
+                    sizes[(rank * nch + ch) * 3 + 2] = 1  # self-traffic: not idle
+                    continue
+                base = (dest * nch + ch) * 3
+                sizes[base + 0] = meta.numel()
+                sizes[base + 1] = payload.numel()
+                msgs[(dest, ch)] = (meta, payload)
+        sizes[-1] = 1 if stop_flag else 0
 
-        sizes_d = sizes.to(dev, non_blocking=False)
-        gathered = [torch.zeros_like(sizes_d) for _ in range(world)]
-        dist.all_gather(gathered, sizes_d, group=group)
-        gathered = [g.cpu() for g in gathered]
-        all_stopped = all(int(g[0, 2]) == 1 for g in gathered)
-        any_work = any(int(g[:, :2].sum()) > 0 for g in gathered)
+        if use_streams:
+            # comm stream must see the extract/gather kernels' writes to
+            # the outgoing payloads (enqueued on the default stream)
+            self._ev_fwd.record()
+        stream_ctx = torch.cuda.stream(self._comm_stream) if use_streams else _nullctx()
+        with stream_ctx:
+            if use_streams:
+                self._comm_stream.wait_event(self._ev_fwd)
+            sizes_d = sizes.to(comm_dev, non_blocking=False)
+            gathered = [torch.zeros_like(sizes_d) for _ in range(world)]
+            dist.all_gather(gathered, sizes_d, group=group)
+            gathered = [g.cpu() for g in gathered]
+            all_stopped = all(int(g[-1]) == 1 for g in gathered)
 
-        # post sends/recvs (meta then payload per peer; order pairs them)
-        p2p = []
-        recv_bufs = {}
-        for peer in range(world):
-            if peer == rank:
-                continue
-            n_meta = int(gathered[peer][rank, 0])
-            n_pay = int(gathered[peer][rank, 1])
-            if n_meta > 0 or n_pay > 0:
-                rm = torch.empty(n_meta, dtype=torch.int64, device=dev)
-                rp = torch.empty(n_pay, dtype=torch.float32, device=dev)
-                recv_bufs[peer] = (rm, rp)
-                if n_meta:
-                    p2p.append(dist.P2POp(dist.irecv, rm, peer, group))
-                if n_pay:
-                    p2p.append(dist.P2POp(dist.irecv, rp, peer, group))
-            if peer in msgs:
-                meta, payload = msgs[peer]
-                sm = meta.reshape(-1).to(dev)
-                sp = payload.to(dev) if payload.device != dev else payload
-                if sm.numel():
-                    p2p.append(dist.P2POp(dist.isend, sm, peer, group))
-                if sp.numel():
-                    p2p.append(dist.P2POp(dist.isend, sp, peer, group))
-        if p2p:
-            reqs = dist.batch_isend_irecv(p2p)
-            for r in reqs:
-                r.wait()
-        # handle incoming in fixed rank order for determinism
-        store_dev = self.rt.device
+            # post sends/recvs in identical (peer, channel) order on every
+            # rank; meta-then-payload order pairs the tensors
+            p2p = []
+            recv_bufs = {}
+            for peer in range(world):
+                if peer == rank:
+                    continue
+                for ch in range(nch):
+                    base = (rank * nch + ch) * 3
+                    n_meta = int(gathered[peer][base + 0])
+                    n_pay = int(gathered[peer][base + 1])
+                    if n_meta > 0 or n_pay > 0:
+                        rm = torch.empty(n_meta, dtype=torch.int64, device=comm_dev)
+                        rp = torch.empty(n_pay, dtype=torch.float32, device=comm_dev)
+                        recv_bufs[(peer, ch)] = (rm, rp)
+                        if n_meta:
+                            p2p.append(dist.P2POp(dist.irecv, rm, peer, group))
+                        if n_pay:
+                            p2p.append(dist.P2POp(dist.irecv, rp, peer, group))
+                    if (peer, ch) in msgs:
+                        meta, payload = msgs[(peer, ch)]
+                        sm = meta.reshape(-1).to(comm_dev)
+                        sp = payload.to(comm_dev) if payload.device != comm_dev else payload
+                        if sm.numel():
+                            p2p.append(dist.P2POp(dist.isend, sm, peer, group))
+                        if sp.numel():
+                            p2p.append(dist.P2POp(dist.isend, sp, peer, group))
+            if p2p:
+                reqs = dist.batch_isend_irecv(p2p)
+                for r in reqs:
+                    r.wait()
+            if use_streams:
+                self._ev_bwd.record(self._comm_stream)
+        if use_streams:
+            # store kernels (handlers run on the default stream) must see
+            # the received payloads
+            torch.cuda.default_stream(rt.device).wait_event(self._ev_bwd)
+
+        # per-channel traffic totals over all (sender, dest) pairs
+        gm = torch.stack([g[:-1] for g in gathered]).view(world, world, nch, 3)
+        tot = gm.sum(dim=(0, 1))  # [nch, 3] = (meta, payload, local)
+        stats = [(int(tot[ch, 0]), int(tot[ch, 1]), int(tot[ch, 2])) for ch in range(nch)]
+
+        # handle local (self-targeted) messages first, then incoming in
+        # fixed (peer, channel) order for determinism
+        store_dev = rt.device
         th0 = time.perf_counter()
-        for peer in sorted(recv_bufs):
-            rm, rp = recv_bufs[peer]
+        for ch, meta, payload in local:
+            handler(ch, rank, meta, payload)
+        for (peer, ch) in sorted(recv_bufs):
+            rm, rp = recv_bufs[(peer, ch)]
             meta = rm.cpu()
             _trace(rank, ch, "in ", peer, meta)
             payload = rp if rp.device == store_dev else rp.to(store_dev)
             handler(ch, peer, meta, payload)
         self.phase_totals["handlers"] += time.perf_counter() - th0
-        return all_stopped, any_work
+        return all_stopped, stats
+
+
+class _nullctx:
+    def __enter__(self):
+        return None
+
+    def __exit__(self, *a):
+        return False

@@ -63,9 +63,10 @@ def _worker_loop(rank, world, techniques):
         if not (out >= floor - 1e-3).all():
             # a delta forwarded through a mid-flight relocation takes one
             # extra round per hop, so 2-round WaitSync visibility is
-            # best-effort (bounded staleness) — it must hold after one
-            # more sync window
-            w.wait_sync()
+            # best-effort (bounded staleness). The STRONG WaitSync waits
+            # for a globally-idle point instead, after which every
+            # in-flight delta has drained — one strong wait must fix it.
+            w.wait_sync(strong=True)
             w.pull(pull_keys, out)
         assert (out >= floor - 1e-3).all(), \
             f"rank {rank} it {it}: pulled {out} < floor {floor} (keys {pull_keys})"
@@ -86,19 +87,15 @@ def _worker_loop(rank, world, techniques):
     w.wait_sync()
     w.barrier()
 
-    # ---- phase 3: eventual consistency, exact aggregate everywhere.
-    # Bounded retry: forwarded deltas (relocation racing the final sync)
-    # may land a round or two after WaitSync returns; the aggregate must
-    # be exact within a few more sync windows. A genuinely lost or
-    # duplicated update stays exact-failing no matter how many retries.
+    # ---- phase 3: eventual consistency, exact aggregate everywhere —
+    # NO retries: after a strong WaitSync (2 globally-idle rounds per
+    # channel) every in-flight delta/forward/refresh has drained, so one
+    # pull must be exact. A lost or duplicated update fails here.
     total = w.allreduce(torch.from_numpy(my_pushes)).numpy()
     exp = init + total
     out = np.zeros((NUM_KEYS, LEN), dtype=np.float32)
-    for attempt in range(5):
-        w.pull(np.arange(NUM_KEYS, dtype=np.int64), out)
-        if np.allclose(out, exp, atol=1e-2):
-            break
-        w.wait_sync()
+    w.wait_sync(strong=True)
+    w.pull(np.arange(NUM_KEYS, dtype=np.int64), out)
     assert np.allclose(out, exp, atol=1e-2), \
         f"rank {rank} final mismatch at keys {np.where(np.abs(out - exp) > 1e-2)[0]}"
 
