@@ -2303,6 +2303,272 @@ class Server {
     return loss;
   }
 
+  // ---- fused slab-direct steps at world>1 / relocated layouts --------
+  //
+  // The identity-layout fused step above requires world==1. At world>1
+  // the multi-GPU hot path would otherwise fall off the fused cliff onto
+  // the classic pull/kernel/push path for EVERY sample, even though
+  // intent-driven relocation makes >95% of keys local. This general
+  // variant runs a host metadata pass (the same per-key cost the classic
+  // path pays anyway) that resolves each key's slab offset, compacts the
+  // samples whose keys are ALL local into an offsets-mode fused kernel
+  // launch (row_at world==0), and returns the indices of the samples
+  // with any remote/spilled/stub key — the caller routes those through
+  // the classic path. Owned keys get their version bumped and replica
+  // keys their UPDATED flag set (same bookkeeping as push()), so the
+  // sync protocol ships the fused updates like any other.
+  struct FusedResolve {
+    std::vector<std::vector<int64_t>> offs;  // per input array, compacted
+    std::vector<int64_t> missed;             // sample indices (ascending)
+    int64_t b_hit = 0;
+  };
+
+  FusedResolve resolve_fused(const std::vector<std::pair<const int64_t*, int>>& arrs,
+                             int64_t B) {
+    FusedResolve out;
+    const int na = (int)arrs.size();
+    constexpr int64_t G = 1024;
+    int64_t nchunks = (B + G - 1) / G;
+    struct Part {
+      std::vector<std::vector<int64_t>> offs;
+      std::vector<int64_t> missed;
+    };
+    std::vector<Part> parts(nchunks);
+    pass_pool_.run(nchunks, [&](int64_t c0, int64_t c1) {
+      for (int64_t c = c0; c < c1; ++c) {
+        Part& P = parts[c];
+        P.offs.resize(na);
+        int64_t e = std::min(B, (c + 1) * G);
+        std::vector<int64_t> tmp;
+        for (int64_t b = c * G; b < e; ++b) {
+          tmp.clear();
+          bool ok = true;
+          for (int a = 0; a < na && ok; ++a) {
+            const int64_t* kp = arrs[a].first;
+            int cnt = arrs[a].second;
+            for (int i = 0; i < cnt; ++i) {
+              Key k = kp[b * cnt + i];
+              uint8_t f = flags_[k].load(std::memory_order_acquire);
+              if (!(f & F_PRESENT) || (f & F_STUB)) { ok = false; break; }
+              int64_t off = loc_[k].load(std::memory_order_acquire);
+              if (off & SPILL_BIT) { ok = false; break; }  // classic path handles spill
+              tmp.push_back(off);
+            }
+          }
+          if (!ok) {
+            P.missed.push_back(b);
+            continue;
+          }
+          // bookkeeping so the sync protocol ships these updates
+          size_t t = 0;
+          for (int a = 0; a < na; ++a) {
+            const int64_t* kp = arrs[a].first;
+            int cnt = arrs[a].second;
+            for (int i = 0; i < cnt; ++i) {
+              Key k = kp[b * cnt + i];
+              uint8_t f = flags_[k].load(std::memory_order_acquire);
+              if (f & F_OWNER)
+                version_[k].fetch_add(1, std::memory_order_relaxed);
+              else
+                flags_[k].fetch_or(F_UPDATED);
+              P.offs[a].push_back(tmp[t++]);
+            }
+          }
+        }
+      }
+    });
+    out.offs.resize(na);
+    for (auto& P : parts) {
+      for (int a = 0; a < na; ++a)
+        out.offs[a].insert(out.offs[a].end(), P.offs[a].begin(), P.offs[a].end());
+      out.missed.insert(out.missed.end(), P.missed.begin(), P.missed.end());
+    }
+    out.b_hit = na ? (int64_t)out.offs[0].size() / arrs[0].second : 0;
+    return out;
+  }
+
+  static torch::Tensor i64vec_to_tensor(const std::vector<int64_t>& v) {
+    auto t = torch::empty({(int64_t)v.size()}, torch::TensorOptions().dtype(torch::kInt64));
+    if (!v.empty()) std::memcpy(t.data_ptr<int64_t>(), v.data(), v.size() * sizeof(int64_t));
+    return t;
+  }
+
+  torch::Tensor offs_to_dev(const std::vector<int64_t>& v) {
+    auto t = torch::from_blob((void*)v.data(), {(int64_t)v.size()},
+                              torch::TensorOptions().dtype(torch::kInt64))
+                 .clone();
+    return dev_.is_cuda() ? t.to(dev_, /*non_blocking=*/true) : t;
+  }
+
+  std::tuple<torch::Tensor, torch::Tensor> kge_step_fused_general(
+      torch::Tensor keys_s, torch::Tensor keys_r, torch::Tensor keys_o, torch::Tensor keys_neg,
+      int64_t N, int64_t D, double lr, double eps) {
+    check_not_failed();
+    TORCH_CHECK(uniform_len_ == 2 * D, "kge_step_fused: store rows must be [emb|accum] = 2D");
+    for (auto* t : {&keys_s, &keys_r, &keys_o, &keys_neg}) {
+      check_keys(*t);
+      const int64_t* kp = t->data_ptr<int64_t>();
+      for (int64_t i = 0; i < t->numel(); ++i)
+        TORCH_CHECK((uint64_t)kp[i] < (uint64_t)num_keys_, "key out of range: ", kp[i]);
+    }
+    int64_t B = keys_s.numel();
+    TORCH_CHECK(keys_r.numel() == B && keys_o.numel() == B && keys_neg.numel() == B * N);
+
+    torch::Tensor loss, missed_t;
+    {
+      InflightGuard g(this);
+      auto res = resolve_fused({{keys_s.data_ptr<int64_t>(), 1},
+                                {keys_r.data_ptr<int64_t>(), 1},
+                                {keys_o.data_ptr<int64_t>(), 1},
+                                {keys_neg.data_ptr<int64_t>(), (int)N}},
+                               B);
+      int64_t Bh = res.b_hit;
+      loss = torch::empty({Bh}, torch::TensorOptions().dtype(torch::kFloat32).device(dev_));
+      if (Bh > 0) {
+        if (dev_.is_cuda()) {
+          auto os = offs_to_dev(res.offs[0]);
+          auto orr = offs_to_dev(res.offs[1]);
+          auto oo = offs_to_dev(res.offs[2]);
+          auto on = offs_to_dev(res.offs[3]);
+          kge_complex_step_fused_gpu(slab_.data, os.data_ptr<int64_t>(), orr.data_ptr<int64_t>(),
+                                     oo.data_ptr<int64_t>(), on.data_ptr<int64_t>(),
+                                     loss.data_ptr<float>(), (int)Bh, (int)N, (int)D,
+                                     Slab::padded(uniform_len_), /*world=offsets-mode*/ 0, rank_,
+                                     (float)lr, (float)eps, current_stream(dev_));
+        } else {
+          std::lock_guard<std::mutex> vg(cpu_val_mu_);
+          kge_complex_step_fused_offs_cpu(slab_.data, res.offs[0].data(), res.offs[1].data(),
+                                          res.offs[2].data(), res.offs[3].data(),
+                                          loss.data_ptr<float>(), (int)Bh, (int)N, (int)D,
+                                          (float)lr, (float)eps);
+        }
+      }
+      missed_t = i64vec_to_tensor(res.missed);
+      int64_t total = (3 + N) * Bh;
+      stat_pull_keys_ += total;
+      stat_pull_local_ += total;
+      stat_push_keys_ += total;
+      stat_push_local_ += total;
+      stat_pulls_ += 1;
+      stat_pushes_ += 1;
+    }
+    return {loss, missed_t};
+  }
+
+  std::tuple<torch::Tensor, torch::Tensor> w2v_step_fused_general(
+      torch::Tensor keys_ctr, torch::Tensor keys_ctx, torch::Tensor keys_neg, int64_t N,
+      int64_t D, double lr, double eps) {
+    check_not_failed();
+    TORCH_CHECK(uniform_len_ == 2 * D, "w2v_step_fused: store rows must be [emb|accum] = 2D");
+    for (auto* t : {&keys_ctr, &keys_ctx, &keys_neg}) {
+      check_keys(*t);
+      const int64_t* kp = t->data_ptr<int64_t>();
+      for (int64_t i = 0; i < t->numel(); ++i)
+        TORCH_CHECK((uint64_t)kp[i] < (uint64_t)num_keys_, "key out of range: ", kp[i]);
+    }
+    int64_t B = keys_ctr.numel();
+    TORCH_CHECK(keys_ctx.numel() == B && keys_neg.numel() == B * N);
+    torch::Tensor loss, missed_t;
+    {
+      InflightGuard g(this);
+      auto res = resolve_fused({{keys_ctr.data_ptr<int64_t>(), 1},
+                                {keys_ctx.data_ptr<int64_t>(), 1},
+                                {keys_neg.data_ptr<int64_t>(), (int)N}},
+                               B);
+      int64_t Bh = res.b_hit;
+      loss = torch::empty({Bh}, torch::TensorOptions().dtype(torch::kFloat32).device(dev_));
+      if (Bh > 0) {
+        if (dev_.is_cuda()) {
+          auto oc = offs_to_dev(res.offs[0]);
+          auto ox = offs_to_dev(res.offs[1]);
+          auto on = offs_to_dev(res.offs[2]);
+          w2v_sgns_step_fused_gpu(slab_.data, oc.data_ptr<int64_t>(), ox.data_ptr<int64_t>(),
+                                  on.data_ptr<int64_t>(), loss.data_ptr<float>(), (int)Bh,
+                                  (int)N, (int)D, Slab::padded(uniform_len_), 0, (float)lr,
+                                  (float)eps, current_stream(dev_));
+        } else {
+          std::lock_guard<std::mutex> vg(cpu_val_mu_);
+          w2v_sgns_step_fused_offs_cpu(slab_.data, res.offs[0].data(), res.offs[1].data(),
+                                       res.offs[2].data(), loss.data_ptr<float>(), (int)Bh,
+                                       (int)N, (int)D, (float)lr, (float)eps);
+        }
+      }
+      missed_t = i64vec_to_tensor(res.missed);
+      int64_t total = (2 + N) * Bh;
+      stat_pull_keys_ += total;
+      stat_pull_local_ += total;
+      stat_push_keys_ += total;
+      stat_push_local_ += total;
+      stat_pulls_ += 1;
+      stat_pushes_ += 1;
+    }
+    return {loss, missed_t};
+  }
+
+  std::tuple<torch::Tensor, torch::Tensor> mf_step_fused_general(torch::Tensor keys_w,
+                                                                 torch::Tensor keys_h,
+                                                                 torch::Tensor x, int64_t R,
+                                                                 double lr, double lambda,
+                                                                 double eps) {
+    check_not_failed();
+    TORCH_CHECK(uniform_len_ == 2 * R, "mf_step_fused: store rows must be [emb|accum] = 2R");
+    TORCH_CHECK(x.scalar_type() == torch::kFloat32, "ratings must be float32");
+    for (auto* t : {&keys_w, &keys_h}) {
+      check_keys(*t);
+      const int64_t* kp = t->data_ptr<int64_t>();
+      for (int64_t i = 0; i < t->numel(); ++i)
+        TORCH_CHECK((uint64_t)kp[i] < (uint64_t)num_keys_, "key out of range: ", kp[i]);
+    }
+    int64_t B = keys_w.numel();
+    TORCH_CHECK(keys_h.numel() == B && x.numel() == B);
+    torch::Tensor loss, missed_t;
+    {
+      InflightGuard g(this);
+      auto res = resolve_fused(
+          {{keys_w.data_ptr<int64_t>(), 1}, {keys_h.data_ptr<int64_t>(), 1}}, B);
+      int64_t Bh = res.b_hit;
+      loss = torch::empty({Bh}, torch::TensorOptions().dtype(torch::kFloat32).device(dev_));
+      if (Bh > 0) {
+        // compact the ratings to the hit samples
+        auto xc = x.contiguous();
+        torch::Tensor xh = torch::empty({Bh}, torch::TensorOptions().dtype(torch::kFloat32));
+        {
+          const float* xp = xc.data_ptr<float>();
+          float* xo = xh.data_ptr<float>();
+          size_t mi = 0;
+          int64_t w = 0;
+          for (int64_t b = 0; b < B; ++b) {
+            if (mi < res.missed.size() && res.missed[mi] == b) { mi++; continue; }
+            xo[w++] = xp[b];
+          }
+        }
+        if (dev_.is_cuda()) {
+          auto ow = offs_to_dev(res.offs[0]);
+          auto oh = offs_to_dev(res.offs[1]);
+          auto xd = xh.to(dev_, true);
+          mf_update_step_fused_gpu(slab_.data, ow.data_ptr<int64_t>(), oh.data_ptr<int64_t>(),
+                                   xd.data_ptr<float>(), loss.data_ptr<float>(), (int)Bh,
+                                   (int)R, Slab::padded(uniform_len_), 0, (float)lr,
+                                   (float)lambda, (float)eps, current_stream(dev_));
+        } else {
+          std::lock_guard<std::mutex> vg(cpu_val_mu_);
+          mf_update_step_fused_offs_cpu(slab_.data, res.offs[0].data(), res.offs[1].data(),
+                                        xh.data_ptr<float>(), loss.data_ptr<float>(), (int)Bh,
+                                        (int)R, (float)lr, (float)lambda, (float)eps);
+        }
+      }
+      missed_t = i64vec_to_tensor(res.missed);
+      int64_t total = 2 * Bh;
+      stat_pull_keys_ += total;
+      stat_pull_local_ += total;
+      stat_push_keys_ += total;
+      stat_push_local_ += total;
+      stat_pulls_ += 1;
+      stat_pushes_ += 1;
+    }
+    return {loss, missed_t};
+  }
+
   torch::Tensor w2v_step_fused(torch::Tensor keys_ctr, torch::Tensor keys_ctx,
                                torch::Tensor keys_neg, int64_t N, int64_t D, double lr,
                                double eps) {
@@ -2572,6 +2838,7 @@ class Server {
 
   int64_t get_len(Key k) { return len_of(k); }
   int64_t uniform_len() const { return uniform_len_; }  // -1 if per-key lengths
+  bool layout_identity() const { return layout_identity_.load(std::memory_order_acquire); }
   int64_t num_keys() const { return num_keys_; }
   int rank() const { return rank_; }
   int world() const { return world_; }
@@ -2951,11 +3218,18 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("kge_step_fused", &Server::kge_step_fused, py::call_guard<py::gil_scoped_release>())
       .def("w2v_step_fused", &Server::w2v_step_fused, py::call_guard<py::gil_scoped_release>())
       .def("mf_step_fused", &Server::mf_step_fused, py::call_guard<py::gil_scoped_release>())
+      .def("kge_step_fused_general", &Server::kge_step_fused_general,
+           py::call_guard<py::gil_scoped_release>())
+      .def("w2v_step_fused_general", &Server::w2v_step_fused_general,
+           py::call_guard<py::gil_scoped_release>())
+      .def("mf_step_fused_general", &Server::mf_step_fused_general,
+           py::call_guard<py::gil_scoped_release>())
       .def("rebalance_spill", &Server::rebalance_spill, py::arg("max_moves") = 4096,
            py::call_guard<py::gil_scoped_release>())
       .def("key_tier", &Server::key_tier)
       .def("get_len", &Server::get_len)
       .def("uniform_len", &Server::uniform_len)
+      .def("layout_identity", &Server::layout_identity)
       .def("num_keys", &Server::num_keys)
       .def("rank", &Server::rank)
       .def("world", &Server::world)

@@ -28,17 +28,32 @@ void kge_complex_step_cpu(const float* s, const float* r, const float* o, const 
                           int B, int N, int D, float lr, float eps);
 
 // FUSED ComplEx step: reads entity/relation rows DIRECTLY from the slab
-// (identity layout: row k at (k/world)*plen) and atomically accumulates
-// the AdaGrad-transformed deltas back — no intermediate pull/push
-// buffers (the classic path pays a gather write + kernel read + delta
-// write + scatter read; this saves all four). Duplicate keys within the
-// batch see hogwild-style concurrent updates (async-PS semantics).
-// keys_* are DEVICE int64 pointers. Keys not owned here are skipped
-// (callers must ensure all-local, e.g. world==1).
+// and atomically accumulates the AdaGrad-transformed deltas back — no
+// intermediate pull/push buffers (the classic path pays a gather write +
+// kernel read + delta write + scatter read; this saves all four).
+// Duplicate keys within the batch see hogwild-style concurrent updates
+// (async-PS semantics). keys_* are DEVICE int64 pointers.
+// Addressing modes (row_at): world >= 1 — identity layout, key k at
+// (k/world)*plen (single-rank fast path, zero host work); world == 0 —
+// keys_* carry precomputed float OFFSETS into the slab (the world>1 /
+// relocated-layout path: the host pass resolves offsets and compacts
+// all-local samples, remote ones take the classic path).
 void kge_complex_step_fused_gpu(float* slab, const int64_t* keys_s, const int64_t* keys_r,
                                 const int64_t* keys_o, const int64_t* keys_neg, float* loss,
                                 int B, int N, int D, int32_t plen, int world, int rank,
                                 float lr, float eps, void* stream);
+
+// CPU tier of the offsets-mode fused steps (exercises the world>1 fused
+// protocol path in the gloo multi-process tests; serial, in-place).
+void kge_complex_step_fused_offs_cpu(float* slab, const int64_t* offs_s, const int64_t* offs_r,
+                                     const int64_t* offs_o, const int64_t* offs_neg, float* loss,
+                                     int B, int N, int D, float lr, float eps);
+void w2v_sgns_step_fused_offs_cpu(float* slab, const int64_t* offs_ctr, const int64_t* offs_ctx,
+                                  const int64_t* offs_neg, float* loss, int B, int N, int D,
+                                  float lr, float eps);
+void mf_update_step_fused_offs_cpu(float* slab, const int64_t* offs_w, const int64_t* offs_h,
+                                   const float* x, float* loss, int B, int R, float lr,
+                                   float lambda, float eps);
 
 // FUSED SGNS step (same contract as kge_complex_step_fused_gpu: slab-
 // direct reads + atomic AdaGrad writes, hogwild on duplicates).

@@ -139,6 +139,124 @@ void mf_update_step_cpu(const float* w, const float* h, const float* x, float* d
   }
 }
 
+// Fused slab-direct steps, offsets mode (CPU tier of the world>1 fused
+// path: offs_* carry float offsets into the slab, resolved by the host
+// metadata pass; math mirrors the GPU fused kernels — object rows are
+// updated immediately per negative, s/r accumulated then applied).
+void kge_complex_step_fused_offs_cpu(float* slab, const int64_t* offs_s, const int64_t* offs_r,
+                                     const int64_t* offs_o, const int64_t* offs_neg, float* loss,
+                                     int B, int N, int D, float lr, float eps) {
+  const int dc = D >> 1;
+  std::vector<float> a_sre(dc), a_sim(dc), a_rre(dc), a_rim(dc), u_re(dc), u_im(dc);
+  for (int b = 0; b < B; ++b) {
+    float* sb = slab + offs_s[b];
+    float* rb = slab + offs_r[b];
+    for (int k = 0; k < dc; ++k) {
+      u_re[k] = sb[k] * rb[k] - sb[dc + k] * rb[dc + k];
+      u_im[k] = sb[dc + k] * rb[k] + sb[k] * rb[dc + k];
+      a_sre[k] = a_sim[k] = a_rre[k] = a_rim[k] = 0.f;
+    }
+    float lsum = 0.f;
+    for (int j = 0; j <= N; ++j) {
+      float* ob = slab + ((j == 0) ? offs_o[b] : offs_neg[(int64_t)b * N + (j - 1)]);
+      float y = (j == 0) ? 1.f : -1.f;
+      float psi = 0.f;
+      for (int k = 0; k < dc; ++k) psi += u_re[k] * ob[k] + u_im[k] * ob[dc + k];
+      float c = -y * sigmoidf_(-y * psi);
+      lsum += softplusf_(-y * psi);
+      for (int k = 0; k < dc; ++k) {
+        float o_re = ob[k], o_im = ob[dc + k];
+        a_sre[k] += c * (rb[k] * o_re + rb[dc + k] * o_im);
+        a_sim[k] += c * (rb[k] * o_im - rb[dc + k] * o_re);
+        a_rre[k] += c * (sb[k] * o_re + sb[dc + k] * o_im);
+        a_rim[k] += c * (sb[k] * o_im - sb[dc + k] * o_re);
+        float g_re = c * u_re[k], g_im = c * u_im[k];
+        float G_re = ob[D + k] + g_re * g_re;
+        float G_im = ob[D + dc + k] + g_im * g_im;
+        ob[k] += -lr * g_re / std::sqrt(G_re + eps);
+        ob[dc + k] += -lr * g_im / std::sqrt(G_im + eps);
+        ob[D + k] += g_re * g_re;
+        ob[D + dc + k] += g_im * g_im;
+      }
+    }
+    for (int k = 0; k < dc; ++k) {
+      float Gsr = sb[D + k] + a_sre[k] * a_sre[k];
+      float Gsi = sb[D + dc + k] + a_sim[k] * a_sim[k];
+      sb[k] += -lr * a_sre[k] / std::sqrt(Gsr + eps);
+      sb[dc + k] += -lr * a_sim[k] / std::sqrt(Gsi + eps);
+      sb[D + k] += a_sre[k] * a_sre[k];
+      sb[D + dc + k] += a_sim[k] * a_sim[k];
+      float Grr = rb[D + k] + a_rre[k] * a_rre[k];
+      float Gri = rb[D + dc + k] + a_rim[k] * a_rim[k];
+      rb[k] += -lr * a_rre[k] / std::sqrt(Grr + eps);
+      rb[dc + k] += -lr * a_rim[k] / std::sqrt(Gri + eps);
+      rb[D + k] += a_rre[k] * a_rre[k];
+      rb[D + dc + k] += a_rim[k] * a_rim[k];
+    }
+    loss[b] = lsum;
+  }
+}
+
+void w2v_sgns_step_fused_offs_cpu(float* slab, const int64_t* offs_ctr, const int64_t* offs_ctx,
+                                  const int64_t* offs_neg, float* loss, int B, int N, int D,
+                                  float lr, float eps) {
+  std::vector<float> a_c(D), c_emb(D);
+  for (int b = 0; b < B; ++b) {
+    float* cb = slab + offs_ctr[b];
+    for (int k = 0; k < D; ++k) {
+      c_emb[k] = cb[k];
+      a_c[k] = 0.f;
+    }
+    float lsum = 0.f;
+    for (int j = 0; j <= N; ++j) {
+      float* xb = slab + ((j == 0) ? offs_ctx[b] : offs_neg[(int64_t)b * N + (j - 1)]);
+      float y = (j == 0) ? 1.f : -1.f;
+      float dot = 0.f;
+      for (int k = 0; k < D; ++k) dot += c_emb[k] * xb[k];
+      float g = -y * sigmoidf_(-y * dot);
+      lsum += softplusf_(-y * dot);
+      for (int k = 0; k < D; ++k) {
+        float xv = xb[k];
+        a_c[k] += g * xv;
+        float gx = g * c_emb[k];
+        float G = xb[D + k] + gx * gx;
+        xb[k] += -lr * gx / std::sqrt(G + eps);
+        xb[D + k] += gx * gx;
+      }
+    }
+    for (int k = 0; k < D; ++k) {
+      float G = cb[D + k] + a_c[k] * a_c[k];
+      cb[k] += -lr * a_c[k] / std::sqrt(G + eps);
+      cb[D + k] += a_c[k] * a_c[k];
+    }
+    loss[b] = lsum;
+  }
+}
+
+void mf_update_step_fused_offs_cpu(float* slab, const int64_t* offs_w, const int64_t* offs_h,
+                                   const float* x, float* loss, int B, int R, float lr,
+                                   float lambda, float eps) {
+  for (int b = 0; b < B; ++b) {
+    float* wb = slab + offs_w[b];
+    float* hb = slab + offs_h[b];
+    float pred = 0.f;
+    for (int k = 0; k < R; ++k) pred += wb[k] * hb[k];
+    float e = x[b] - pred;
+    loss[b] = e * e;
+    for (int k = 0; k < R; ++k) {
+      float wv = wb[k], hv = hb[k];
+      float gw = -2.f * e * hv + 2.f * lambda * wv;
+      float gh = -2.f * e * wv + 2.f * lambda * hv;
+      float Gw = wb[R + k] + gw * gw;
+      float Gh = hb[R + k] + gh * gh;
+      wb[k] += -lr * gw / std::sqrt(Gw + eps);
+      wb[R + k] += gw * gw;
+      hb[k] += -lr * gh / std::sqrt(Gh + eps);
+      hb[R + k] += gh * gh;
+    }
+  }
+}
+
 static inline uint64_t pcg_hash64_c(uint64_t x) {
   x ^= x >> 33; x *= 0xff51afd7ed558ccdULL;
   x ^= x >> 33; x *= 0xc4ceb9fe1a85ec53ULL;

@@ -427,6 +427,16 @@ void rescal_step_gpu(const float* s, const float* r, const float* o, const float
 
 namespace adapm {
 
+// Row resolution for the fused slab-direct kernels. world >= 1: identity
+// layout, key k lives at (k/world)*plen (the world==1 single-rank fast
+// path with zero host work). world == 0: the "keys" arrays carry
+// precomputed float OFFSETS into the slab (the world>1 / relocated-layout
+// path: the host metadata pass resolves each key's slot and compacts
+// all-local samples; remote ones go through the classic pull/push path).
+__device__ inline float* row_at(float* slab, int64_t v, int32_t plen, int world) {
+  return world ? slab + (v / world) * (int64_t)plen : slab + v;
+}
+
 // Fused ComplEx step — see kernels.h. Structure mirrors k_kge_step but
 // rows come from / return to the slab itself.
 template <int KPT>
@@ -439,8 +449,8 @@ __global__ void k_kge_step_fused(float* __restrict__ slab, const int64_t* __rest
   __shared__ float lds[KT / 64];
   const int dc = D >> 1;
   for (int b = blockIdx.x; b < B; b += gridDim.x) {
-    const float* sb = slab + (keys_s[b] / world) * (int64_t)plen;
-    const float* rb = slab + (keys_r[b] / world) * (int64_t)plen;
+    const float* sb = row_at(slab, keys_s[b], plen, world);
+    const float* rb = row_at(slab, keys_r[b], plen, world);
 
     float s_re[KPT], s_im[KPT], r_re[KPT], r_im[KPT];
     float a_sre[KPT], a_sim[KPT], a_rre[KPT], a_rim[KPT];
@@ -464,7 +474,7 @@ __global__ void k_kge_step_fused(float* __restrict__ slab, const int64_t* __rest
     float lsum = 0.f;
     for (int j = 0; j <= N; ++j) {
       int64_t ok = (j == 0) ? keys_o[b] : keys_neg[(int64_t)b * N + (j - 1)];
-      float* ob = slab + (ok / world) * (int64_t)plen;
+      float* ob = row_at(slab, ok, plen, world);
       float y = (j == 0) ? 1.f : -1.f;
       float part = 0.f;
       float o_re[KPT], o_im[KPT];
@@ -551,7 +561,7 @@ __global__ void k_w2v_step_fused(float* __restrict__ slab, const int64_t* __rest
                                  float eps) {
   __shared__ float lds[KT / 64];
   for (int b = blockIdx.x; b < B; b += gridDim.x) {
-    float* cb = slab + (keys_ctr[b] / world) * (int64_t)plen;
+    float* cb = row_at(slab, keys_ctr[b], plen, world);
     float c_emb[KPT], a_c[KPT];
 #pragma unroll
     for (int i = 0; i < KPT; ++i) {
@@ -562,7 +572,7 @@ __global__ void k_w2v_step_fused(float* __restrict__ slab, const int64_t* __rest
     float lsum = 0.f;
     for (int j = 0; j <= N; ++j) {
       int64_t xk = (j == 0) ? keys_ctx[b] : keys_neg[(int64_t)b * N + (j - 1)];
-      float* xb = slab + (xk / world) * (int64_t)plen;
+      float* xb = row_at(slab, xk, plen, world);
       float y = (j == 0) ? 1.f : -1.f;
       float part = 0.f;
       float x_emb[KPT];
@@ -607,8 +617,8 @@ __global__ void k_mf_step_fused(float* __restrict__ slab, const int64_t* __restr
                                 float lr, float lambda, float eps) {
   __shared__ float lds[KT / 64];
   for (int b = blockIdx.x; b < B; b += gridDim.x) {
-    float* wb = slab + (keys_w[b] / world) * (int64_t)plen;
-    float* hb = slab + (keys_h[b] / world) * (int64_t)plen;
+    float* wb = row_at(slab, keys_w[b], plen, world);
+    float* hb = row_at(slab, keys_h[b], plen, world);
     float part = 0.f;
     for (int k = threadIdx.x; k < R; k += KT) part += wb[k] * hb[k];
     float pred = block_reduce_sum(part, lds);

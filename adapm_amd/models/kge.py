@@ -103,16 +103,9 @@ class ComplEx:
         s, r, o = self.keys_of(triples)
         self.worker.intent(np.concatenate([s, r, o]), start, end)
 
-    def train_batch(self, triples: np.ndarray, async_push: bool = True,
-                    sync_loss: bool = True):
-        """One training step over B positive triples. Returns mean loss
-        (float if sync_loss else a device tensor, no host sync)."""
+    def _draw_negatives(self, B):
         cfg = self.cfg
         w = self.worker
-        B = len(triples)
-        t0 = time.perf_counter() if _PHASE_TIMING else 0.0
-        s_keys, r_keys, o_keys = self.keys_of(triples)
-
         # negatives via the sampling manager (reference PrepareSample path)
         if self.server.sampling is not None:
             sid = w.prepare_sample(B * cfg.neg_samples, w.current_clock(),
@@ -122,6 +115,19 @@ class ComplEx:
         else:
             neg_keys = self.rng.integers(0, cfg.num_entities, size=B * cfg.neg_samples,
                                          dtype=np.int64)
+        return neg_keys
+
+    def train_batch(self, triples: np.ndarray, async_push: bool = True,
+                    sync_loss: bool = True, neg_keys: np.ndarray = None):
+        """One training step over B positive triples. Returns mean loss
+        (float if sync_loss else a device tensor, no host sync)."""
+        cfg = self.cfg
+        w = self.worker
+        B = len(triples)
+        t0 = time.perf_counter() if _PHASE_TIMING else 0.0
+        s_keys, r_keys, o_keys = self.keys_of(triples)
+        if neg_keys is None:
+            neg_keys = self._draw_negatives(B)
         t0 = self._ph("sample", t0)
 
         # one fused pull of all rows: [s | r | o | neg]
@@ -164,30 +170,44 @@ class ComplEx:
             return out
         return loss
 
-    def train_batch_fused(self, triples: np.ndarray, sync_loss: bool = False):
-        """Single-rank GPU fast path: one fused kernel reads the rows
-        straight from the HBM slab and accumulates the AdaGrad deltas
-        back (Server.kge_step_fused) — no pull/push buffers. Falls back
-        to train_batch when the preconditions don't hold."""
+    def train_batch_fused(self, triples: np.ndarray, sync_loss: bool = False,
+                          force_general: bool = False):
+        """Fused slab-direct step: kernels read rows straight from the
+        HBM slab and accumulate the AdaGrad deltas back — no pull/push
+        buffers. world==1 with the identity layout uses the zero-host-
+        work path (Server.kge_step_fused); otherwise the general path
+        (kge_step_fused_general) resolves slab offsets in a host pass,
+        runs the fused kernel on the samples whose keys are all local
+        (>95% after intent-driven relocation) and routes the rest
+        through the classic pull/kernel/push path. force_general also
+        enables the general path on the CPU store (the gloo test tier)."""
         cfg = self.cfg
         w = self.worker
-        if self.world != 1 or self.dev.type != "cuda":
+        raw = self.server.raw
+        if self.dev.type != "cuda" and not force_general:
             return self.train_batch(triples, sync_loss=sync_loss)
         B = len(triples)
         s_keys, r_keys, o_keys = self.keys_of(triples)
-        if self.server.sampling is not None:
-            sid = w.prepare_sample(B * cfg.neg_samples, w.current_clock(),
-                                   w.current_clock() + 2)
-            neg_keys = self.server.sampling.pull(w, sid, B * cfg.neg_samples)
-            w.finish_sample(sid)
-        else:
-            neg_keys = self.rng.integers(0, cfg.num_entities, size=B * cfg.neg_samples,
-                                         dtype=np.int64)
-        loss = self.server.raw.kge_step_fused(
+        neg_keys = np.ascontiguousarray(self._draw_negatives(B), dtype=np.int64)
+
+        if self.world == 1 and not force_general and raw.layout_identity():
+            loss = raw.kge_step_fused(
+                torch.from_numpy(s_keys), torch.from_numpy(r_keys), torch.from_numpy(o_keys),
+                torch.from_numpy(neg_keys), cfg.neg_samples, cfg.dim, cfg.lr, cfg.eps)
+            return float(loss.mean().item()) if sync_loss else loss
+
+        loss, missed = raw.kge_step_fused_general(
             torch.from_numpy(s_keys), torch.from_numpy(r_keys), torch.from_numpy(o_keys),
-            torch.from_numpy(np.ascontiguousarray(neg_keys, dtype=np.int64)),
-            cfg.neg_samples, cfg.dim, cfg.lr, cfg.eps)
-        return float(loss.mean().item()) if sync_loss else loss
+            torch.from_numpy(neg_keys), cfg.neg_samples, cfg.dim, cfg.lr, cfg.eps)
+        if missed.numel() == 0:
+            return float(loss.mean().item()) if sync_loss else loss
+        midx = missed.numpy()
+        sub_negs = neg_keys.reshape(B, cfg.neg_samples)[midx].reshape(-1)
+        mloss = self.train_batch(triples[midx], sync_loss=False, neg_keys=sub_negs)
+        if not torch.is_tensor(mloss):
+            mloss = torch.tensor([mloss])
+        full = torch.cat([loss, mloss.to(loss.device)])
+        return float(full.mean().item()) if sync_loss else full
 
     def drain(self):
         for t in self._pending:

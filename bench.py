@@ -103,7 +103,10 @@ def main():
 
     handles = deque()
 
-    use_fused = (args.pipeline == 0) and (not args.no_fused) and world == 1 and is_cuda
+    # world==1: identity-layout fused kernel (zero host work); world>1:
+    # general fused path (host offset pass + fused kernel on all-local
+    # samples, classic path on the remote remainder)
+    use_fused = (args.pipeline == 0) and (not args.no_fused) and is_cuda
 
     def issue(i):
         if not args.no_intent:
@@ -178,7 +181,7 @@ def main():
     # slab-direct step is the world==1 hot path; the classic
     # pull/kernel/push path is what runs at world>1, so report both
     classic_ops_per_s = None
-    if use_fused:
+    if use_fused and world == 1:
         for _ in range(2):
             model.train_batch(get_batch(step_i))
             worker.advance_clock()
