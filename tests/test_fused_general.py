@@ -42,11 +42,15 @@ def _expected_after_classic(init_rows, s, r, o, negs, lr, eps):
     return exp, loss
 
 
-def _fused_general_worker(rank, world, use_intent):
+def _fused_general_worker(rank, world, use_intent, device="cpu"):
+    import os
+
+    if device != "cpu":
+        os.environ["ADAPM_FORCE_GLOO"] = "1"  # 2 ranks share one GPU
     import adapm_amd
     from adapm_amd.models.kge import ComplEx, ComplExConfig
 
-    adapm_amd.setup(num_keys=ENT + REL, num_threads=1, device="cpu",
+    adapm_amd.setup(num_keys=ENT + REL, num_threads=1, device=device,
                     max_sync_per_sec=4000.0)
     server = adapm_amd.Server(2 * DIM)
     worker = adapm_amd.Worker(0, server)
@@ -121,3 +125,9 @@ def _fused_general_worker(rank, world, use_intent):
 @pytest.mark.parametrize("use_intent", [False, True])
 def test_fused_general_world2(use_intent):
     run_dist(2, _fused_general_worker, use_intent, timeout=180)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("use_intent", [False, True])
+def test_fused_general_world2_gpu(use_intent):
+    run_dist(2, _fused_general_worker, use_intent, "cuda:0", timeout=180)
