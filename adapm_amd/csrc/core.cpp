@@ -516,29 +516,38 @@ class Server {
   };
 
   struct DevBatch {
-    torch::Tensor src_t, dst_t, len_t, aux_t;
+    torch::Tensor src_t;  // the packed staging blob (device or pinned host)
+    const int64_t* aux_dev = nullptr;
     OpsBatch b;
   };
 
+  // Stage a host batch for the device kernels with ONE allocation + ONE
+  // H2D copy: [src i64 | dst i64 | aux i64 | len i32] packed into a
+  // pinned blob (torch's caching host allocator recycles it), instead of
+  // 3-4 separate pageable clones + copies. This was the largest C++ term
+  // on the CTR spill config (t_todev, docs/NOTES_NEXT_ROUND.md lever c).
   DevBatch to_dev(const HostBatch& hb, const std::vector<int64_t>* aux = nullptr) {
     int64_t t0 = cpp_timing_ ? std::chrono::steady_clock::now().time_since_epoch().count() : 0;
     DevBatch d;
-    auto i64 = torch::TensorOptions().dtype(torch::kInt64);
-    auto i32 = torch::TensorOptions().dtype(torch::kInt32);
-    d.src_t = torch::from_blob((void*)hb.src.data(), {(int64_t)hb.size()}, i64).clone();
-    d.dst_t = torch::from_blob((void*)hb.dst.data(), {(int64_t)hb.size()}, i64).clone();
-    d.len_t = torch::from_blob((void*)hb.len.data(), {(int64_t)hb.size()}, i32).clone();
-    if (aux) d.aux_t = torch::from_blob((void*)aux->data(), {(int64_t)aux->size()}, i64).clone();
-    if (dev_.is_cuda()) {
-      d.src_t = d.src_t.to(dev_, /*non_blocking=*/true);
-      d.dst_t = d.dst_t.to(dev_, true);
-      d.len_t = d.len_t.to(dev_, true);
-      if (aux) d.aux_t = d.aux_t.to(dev_, true);
-    }
-    d.b.src_off = d.src_t.data_ptr<int64_t>();
-    d.b.dst_off = d.dst_t.data_ptr<int64_t>();
-    d.b.lens = d.len_t.data_ptr<int32_t>();
-    d.b.n = (int)hb.size();
+    const int64_t n = (int64_t)hb.size();
+    const int64_t na = aux ? (int64_t)aux->size() : 0;
+    const int64_t bytes = n * 8 * 2 + na * 8 + n * 4;
+    auto blob = torch::empty({bytes}, torch::TensorOptions()
+                                          .dtype(torch::kUInt8)
+                                          .pinned_memory(dev_.is_cuda()));
+    uint8_t* p = blob.data_ptr<uint8_t>();
+    std::memcpy(p, hb.src.data(), n * 8);
+    std::memcpy(p + n * 8, hb.dst.data(), n * 8);
+    if (na) std::memcpy(p + n * 16, aux->data(), na * 8);
+    std::memcpy(p + n * 16 + na * 8, hb.len.data(), n * 4);
+    torch::Tensor dblob = dev_.is_cuda() ? blob.to(dev_, /*non_blocking=*/true) : blob;
+    uint8_t* dp = dblob.data_ptr<uint8_t>();
+    d.src_t = dblob;  // keeps the device blob alive
+    d.b.src_off = (const int64_t*)dp;
+    d.b.dst_off = (const int64_t*)(dp + n * 8);
+    if (na) d.aux_dev = (const int64_t*)(dp + n * 16);
+    d.b.lens = (const int32_t*)(dp + n * 16 + na * 8);
+    d.b.n = (int)n;
     if (cpp_timing_)
       t_todev_ += std::chrono::steady_clock::now().time_since_epoch().count() - t0;
     return d;
@@ -576,7 +585,7 @@ class Server {
     if (hb.size() == 0) return;
     auto d = to_dev(hb, &sync_off);
     if (dev_.is_cuda()) {
-      ops_delta_sqnorm_gpu(slab_.bases(), d.b, d.aux_t.data_ptr<int64_t>(),
+      ops_delta_sqnorm_gpu(slab_.bases(), d.b, d.aux_dev,
                            out.data_ptr<float>(), current_stream(dev_));
     } else {
       std::lock_guard<std::mutex> g(cpu_val_mu_);
@@ -587,7 +596,7 @@ class Server {
     if (hb.size() == 0) return;
     auto d = to_dev(hb, &sync_off);
     if (dev_.is_cuda()) {
-      ops_extract_gpu(slab_.bases(), d.b, d.aux_t.data_ptr<int64_t>(), out.data_ptr<float>(),
+      ops_extract_gpu(slab_.bases(), d.b, d.aux_dev, out.data_ptr<float>(),
                       current_stream(dev_));
     } else {
       std::lock_guard<std::mutex> g(cpu_val_mu_);
@@ -598,7 +607,7 @@ class Server {
     if (hb.size() == 0) return;
     auto d = to_dev(hb, &sync_off);
     if (dev_.is_cuda()) {
-      ops_refresh_gpu(slab_.bases(), d.b, d.aux_t.data_ptr<int64_t>(), in.data_ptr<float>(),
+      ops_refresh_gpu(slab_.bases(), d.b, d.aux_dev, in.data_ptr<float>(),
                       current_stream(dev_));
     } else {
       std::lock_guard<std::mutex> g(cpu_val_mu_);

@@ -24,6 +24,12 @@ def _entry(rank: int, world: int, port: int, fn, args, q):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     os.environ.setdefault("OMP_NUM_THREADS", "2")
+    # hang forensics: dump every thread's stack if the worker is still
+    # alive this long (run_dist timeouts then show WHERE it hung)
+    import faulthandler
+
+    faulthandler.dump_traceback_later(
+        float(os.environ.get("ADAPM_TEST_DUMP_S", "240")), exit=False)
     try:
         fn(rank, world, *args)
         q.put((rank, None))
