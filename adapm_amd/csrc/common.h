@@ -16,12 +16,16 @@ using Key = int64_t;
 using Clock = int64_t;
 
 // per-key metadata flags (host-side; cf. reference Parameter struct,
-// coloc_kv_server_handle.h:121-152)
+// coloc_kv_server_handle.h:121-152). Packed with the slab offset into
+// ONE atomic int64 per key (see Server::meta_), so the worker metadata
+// pass costs one cache miss per key, not two.
 enum KeyFlags : uint8_t {
-  F_PRESENT = 1,  // value allocated in local slab
-  F_OWNER = 2,    // we hold the main copy
-  F_STUB = 4,     // replica placeholder, no data received yet (version "-1")
-  F_UPDATED = 8,  // replica has local (unsynced) updates
+  F_PRESENT = 1,   // value allocated in local slab
+  F_OWNER = 2,     // we hold the main copy
+  F_STUB = 4,      // replica placeholder, no data received yet (version "-1")
+  F_UPDATED = 8,   // replica has local (unsynced) updates
+  F_HASREP = 32,   // owned key with >=1 granted replica: version bumps are
+                   // observable (unreplicated keys skip the version_ touch)
 };
 
 // management techniques (reference postoffice.h management_techniques)

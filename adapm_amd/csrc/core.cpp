@@ -303,12 +303,10 @@ class Server {
       lens_.assign(lens.data_ptr<int32_t>(), lens.data_ptr<int32_t>() + num_keys_);
     }
 
-    flags_ = std::vector<std::atomic<uint8_t>>(num_keys_);
-    loc_ = std::vector<std::atomic<int64_t>>(num_keys_);
+    meta_ = std::vector<std::atomic<int64_t>>(num_keys_);
     version_ = std::vector<std::atomic<uint32_t>>(num_keys_);
     for (int64_t i = 0; i < num_keys_; ++i) {
-      flags_[i].store(0, std::memory_order_relaxed);
-      loc_[i].store(-1, std::memory_order_relaxed);
+      meta_[i].store(0, std::memory_order_relaxed);
       version_[i].store(0, std::memory_order_relaxed);
     }
     sync_loc_.assign(num_keys_, -1);
@@ -336,8 +334,7 @@ class Server {
     if (owned_floats > cap) layout_identity_.store(false);
     for (Key k = rank_; k < num_keys_; k += world_) {
       int32_t l = len_of(k);
-      loc_[k] = slab_.alloc(l);
-      flags_[k] = F_PRESENT | F_OWNER;
+      meta_[k] = mpack(slab_.alloc(l), F_PRESENT | F_OWNER);
     }
     // slab is zero-initialized by torch::zeros; bump-fresh slots stay zero
 
@@ -349,6 +346,23 @@ class Server {
 
   // ------------------------------------------------ helpers
 
+  // ---- packed per-key metadata: one atomic int64 = flag bits (0-3),
+  // spill bit (4), has-replica bit (5), slab offset (bits 6+). A
+  // flags+loc pair updates ATOMICALLY, which retires round-1's
+  // loc-before-flags write-ordering rules; readers pay ONE cache miss
+  // per key. loc values keep their external encoding (SPILL_BIT = bit
+  // 62) at the pack/unpack boundary.
+  static constexpr int64_t MSPILL = 16;
+  static constexpr int64_t MFLAGS = F_PRESENT | F_OWNER | F_STUB | F_UPDATED | F_HASREP;
+  static inline int64_t mpack(int64_t loc, int64_t flags) {
+    if (loc < 0) return flags;
+    return ((loc & ~SPILL_BIT) << 6) | ((loc & SPILL_BIT) ? MSPILL : 0) | flags;
+  }
+  static inline uint8_t mflags(int64_t m) { return (uint8_t)(m & MFLAGS); }
+  static inline int64_t mloc(int64_t m) {
+    return (int64_t)((uint64_t)m >> 6) | ((m & MSPILL) ? SPILL_BIT : 0);
+  }
+
   inline int32_t len_of(Key k) const { return uniform_len_ >= 0 ? uniform_len_ : lens_[k]; }
   inline int manager_of(Key k) const { return (int)(k % world_); }
   inline int channel_of(Key k) const {
@@ -359,7 +373,7 @@ class Server {
 
   // believed current location of a key (reference addressbook.h:50-70)
   int directions(Key k) {
-    if (flags_[k] & F_OWNER) return rank_;
+    if (meta_[k].load(std::memory_order_acquire) & F_OWNER) return rank_;
     if (manager_of(k) == rank_) return owner_of_[k / world_];
     if (use_loc_cache_) {
       int c = loc_cache_[k];
@@ -761,9 +775,10 @@ class Server {
             int64_t e = std::min(n, (c + 1) * G);
             for (int64_t i = c * G; i < e; ++i) {
               Key k = kp[i];
-              uint8_t f = flags_[k].load(std::memory_order_acquire);
+              int64_t m = meta_[k].load(std::memory_order_acquire);  // ONE miss: flags+loc
+              uint8_t f = mflags(m);
               if ((f & F_PRESENT) && !(f & F_STUB)) {
-                int64_t off = loc_[k].load(std::memory_order_acquire);
+                int64_t off = mloc(m);
                 P.local.add(off, out_off[i], out_len[i]);
                 if (!(f & F_OWNER)) P.n_repl++;
                 if (heat_) {
@@ -913,9 +928,10 @@ class Server {
             for (int64_t i = c * G; i < e; ++i) {
               Key k = kp[i];
               int32_t l = len_of(k);
-              uint8_t f = flags_[k].load(std::memory_order_acquire);
+              int64_t m = meta_[k].load(std::memory_order_acquire);  // ONE miss: flags+loc
+              uint8_t f = mflags(m);
               if ((f & F_PRESENT) && (f & F_OWNER)) {
-                int64_t off = loc_[k].load(std::memory_order_acquire);
+                int64_t off = mloc(m);
                 // spilled merges go through the non-atomic RMW kernel
                 // (PCIe atomics are ~144x slower; dups re-routed below)
                 bool spill_merge = heat_ && (off & SPILL_BIT) && !set_mode && dev_.is_cuda();
@@ -923,15 +939,17 @@ class Server {
                   P.merge_spill.add(off, offs[i], l);
                 else
                   (set_mode ? P.assign : P.merge).add(off, offs[i], l);
-                version_[k].fetch_add(1, std::memory_order_relaxed);
+                // version bumps are observable only once a replica was
+                // granted (F_HASREP) — unreplicated keys skip the touch
+                if (f & F_HASREP) version_[k].fetch_add(1, std::memory_order_relaxed);
                 if (heat_) {
                   heat_[k].fetch_add(1, std::memory_order_relaxed);
                   if (off & SPILL_BIT) P.spilled.push_back(k);
                 }
               } else if ((f & F_PRESENT) && !set_mode) {
                 // replica/stub: merge locally, flush at next sync round
-                P.merge.add(loc_[k].load(std::memory_order_acquire), offs[i], l);
-                flags_[k].fetch_or(F_UPDATED);
+                P.merge.add(mloc(m), offs[i], l);
+                meta_[k].fetch_or((int64_t)F_UPDATED);
                 P.n_repl++;
               } else {
                 P.remote.push_back({k, offs[i], l});
@@ -1038,9 +1056,10 @@ class Server {
         Key k = kp[i];
         int32_t l = len_of(k);
         {
-            uint8_t f = flags_[k].load(std::memory_order_acquire);
+          int64_t m = meta_[k].load(std::memory_order_acquire);
+          uint8_t f = mflags(m);
           if (!((f & F_PRESENT) && !(f & F_STUB))) return false;
-          local.add(loc_[k].load(std::memory_order_acquire), cum, l);
+          local.add(mloc(m), cum, l);
         }
         cum += l;
       }
@@ -1051,7 +1070,7 @@ class Server {
   }
 
   bool is_local(Key k) {
-    uint8_t f = flags_[k].load(std::memory_order_acquire);
+    uint8_t f = mflags(meta_[k].load(std::memory_order_acquire));
     return (f & F_PRESENT) && !(f & F_STUB);
   }
 
@@ -1228,7 +1247,7 @@ class Server {
         continue;
       }
       for (Key k : req.keys) {
-        if (!(flags_[k].load(std::memory_order_acquire) & F_PRESENT)) {
+        if (!(meta_[k].load(std::memory_order_acquire) & F_PRESENT)) {
           if (!new_stub_keys.insert(k).second) continue;  // dedup within round
           int32_t l = len_of(k);
           layout_identity_.store(false, std::memory_order_release);
@@ -1245,10 +1264,9 @@ class Server {
                            // later worker merge kernel on these slots
     for (auto& ns : new_stubs) {
       std::lock_guard<std::mutex> lk(stripe(ns.k));
-      loc_[ns.k] = ns.v_off;      // loc before flags (meta rules)
       sync_loc_[ns.k] = ns.s_off;
       version_[ns.k] = 0;
-      flags_[ns.k] = F_PRESENT | F_STUB;
+      meta_[ns.k].store(mpack(ns.v_off, F_PRESENT | F_STUB), std::memory_order_release);
       trace_event(ns.k, "REPLICA_SETUP");
     }
     {
@@ -1305,10 +1323,11 @@ class Server {
         }
         if (!has_intent) continue;  // would drop: always ships
         std::lock_guard<std::mutex> lk(stripe(k));
-        uint8_t f = flags_[k];
+        int64_t m = meta_[k].load();
+        uint8_t f = mflags(m);
         if (!(f & F_PRESENT) || (f & F_OWNER)) continue;
         if (!(f & F_UPDATED) || (f & F_STUB)) continue;
-        nb.add(loc_[k].load(), (int64_t)cand.size(), len_of(k));
+        nb.add(mloc(m), (int64_t)cand.size(), len_of(k));
         nsync.push_back(sync_loc_[k]);
         cand.push_back(k);
       }
@@ -1343,7 +1362,8 @@ class Server {
       bool erase_from_replicas = false;
       {
         std::lock_guard<std::mutex> lk(stripe(k));
-        uint8_t f = flags_[k];
+        int64_t m = meta_[k].load();
+        uint8_t f = mflags(m);
         if (!(f & F_PRESENT) || (f & F_OWNER)) {
           erase_from_replicas = true;  // became owner via relocation
         } else {
@@ -1358,13 +1378,13 @@ class Server {
           int64_t fl = (payload ? D_HAS_PAYLOAD : 0) | (drop ? D_DROPPING : 0) |
                        (has_intent ? D_WANT_REFRESH : 0) | (is_new ? D_NEW : 0);
           deltas.push_back(DeltaRec{k, (int64_t)version_[k].load(), fl,
-                                    payload ? loc_[k].load() : -1, sync_loc_[k], len_of(k)});
-          if (updated && !drop) flags_[k].fetch_and((uint8_t)~F_UPDATED);
+                                    payload ? mloc(m) : -1, sync_loc_[k], len_of(k)});
+          if (updated && !drop) meta_[k].fetch_and(~(int64_t)F_UPDATED);
           if (drop) {
             layout_identity_.store(false, std::memory_order_release);
-            frees.push_back({loc_[k], len_of(k)});
+            frees.push_back({mloc(m), len_of(k)});
             frees.push_back({sync_loc_[k], len_of(k)});
-            flags_[k] = 0;  // loc field left stale on purpose (see meta rules)
+            meta_[k].store(0);
             sync_loc_[k] = -1;
             erase_from_replicas = true;
             stat_drops_ += 1;
@@ -1474,10 +1494,11 @@ class Server {
         HostBatch b;
         {
           std::lock_guard<std::mutex> lk(stripe(r.key));
-          if (flags_[r.key] & F_OWNER) {
+          int64_t m = meta_[r.key].load();
+          if (m & F_OWNER) {
             owner = true;
-            b.add(loc_[r.key], 0, len_of(r.key));
-            version_[r.key]++;
+            b.add(mloc(m), 0, len_of(r.key));
+            if (m & F_HASREP) version_[r.key]++;
           }
         }
         if (owner) {
@@ -1499,9 +1520,10 @@ class Server {
         int32_t l = len_of(r.key);
         {
           std::lock_guard<std::mutex> lk(stripe(r.key));
-          if (flags_[r.key] & F_OWNER) {
+          int64_t m = meta_[r.key].load();
+          if (m & F_OWNER) {
             owner = true;
-            voff = loc_[r.key];
+            voff = mloc(m);
           }
         }
         if (owner) {
@@ -1527,10 +1549,11 @@ class Server {
         HostBatch b;
         {
           std::lock_guard<std::mutex> lk(stripe(r.key));
-          owner = flags_[r.key] & F_OWNER;
+          int64_t m = meta_[r.key].load();
+          owner = m & F_OWNER;
           if (owner && r.payload.defined()) {
-            b.add(loc_[r.key], 0, len_of(r.key));
-            version_[r.key]++;
+            b.add(mloc(m), 0, len_of(r.key));
+            if (m & F_HASREP) version_[r.key]++;
           }
         }
         if (owner) {
@@ -1594,11 +1617,11 @@ class Server {
     std::map<int, std::pair<std::vector<Key>, std::vector<int64_t>>> fwd;
     for (int64_t i = 0; i < nk; ++i) {
       Key k = keys[i];
-      uint8_t f = flags_[k].load(std::memory_order_acquire);
-      if (f & F_OWNER) {
+      int64_t m = meta_[k].load(std::memory_order_acquire);
+      if (m & F_OWNER) {
         resp.keys.push_back(k);
         resp.aux.push_back(oidx[i]);
-        resp.slab_offs.push_back(loc_[k].load(std::memory_order_acquire));
+        resp.slab_offs.push_back(mloc(m));
       } else {
         auto& g = fwd[directions(k)];
         g.first.push_back(k);
@@ -1645,10 +1668,10 @@ class Server {
     int64_t applied = 0;
     for (int64_t i = 0; i < nk; ++i) {
       Key k = keys[i];
-      uint8_t f = flags_[k].load(std::memory_order_acquire);
-      if (f & F_OWNER) {
-        apply.add(loc_[k].load(std::memory_order_acquire), i * (int64_t)l, l);
-        version_[k].fetch_add(1, std::memory_order_relaxed);
+      int64_t m = meta_[k].load(std::memory_order_acquire);
+      if (m & F_OWNER) {
+        apply.add(mloc(m), i * (int64_t)l, l);
+        if (m & F_HASREP) version_[k].fetch_add(1, std::memory_order_relaxed);
         applied++;
       } else {
         auto& g = fwd[directions(k)];
@@ -1735,10 +1758,11 @@ class Server {
           bool owner;
           {
             std::lock_guard<std::mutex> lk(stripe(k));
-            owner = flags_[k] & F_OWNER;
+            int64_t m = meta_[k].load();
+            owner = m & F_OWNER;
             if (owner && has_payload) {
-              merges.add(loc_[k], poff, l);
-              version_[k]++;
+              merges.add(mloc(m), poff, l);
+              if (m & F_HASREP) version_[k]++;
             }
           }
           if (!owner) {
@@ -1757,10 +1781,11 @@ class Server {
           bool owner;
           {
             std::lock_guard<std::mutex> lk(stripe(k));
-            owner = flags_[k] & F_OWNER;
+            int64_t m = meta_[k].load();
+            owner = m & F_OWNER;
             if (owner) {
-              (code == M_SET_REQ ? assigns : merges).add(loc_[k], poff, l);
-              version_[k]++;
+              (code == M_SET_REQ ? assigns : merges).add(mloc(m), poff, l);
+              if (m & F_HASREP) version_[k]++;
             }
           }
           if (owner) {
@@ -1780,8 +1805,9 @@ class Server {
           int64_t voff = -1;
           {
             std::lock_guard<std::mutex> lk(stripe(k));
-            owner = flags_[k] & F_OWNER;
-            if (owner) voff = loc_[k];
+            int64_t m = meta_[k].load();
+            owner = m & F_OWNER;
+            if (owner) voff = mloc(m);
           }
           if (owner) {
             std::lock_guard<std::mutex> g(C.mu);
@@ -1811,12 +1837,20 @@ class Server {
                           int64_t dflags) {
     (void)ch;
     if (dflags & D_DROPPING) {
-      std::lock_guard<std::mutex> g(C.mu);
-      auto it = C.holders.find(k);
-      if (it != C.holders.end()) {
-        it->second &= ~(1ULL << origin_rank);
-        if (it->second == 0) C.holders.erase(it);
+      bool none_left = false;
+      {
+        std::lock_guard<std::mutex> g(C.mu);
+        auto it = C.holders.find(k);
+        if (it != C.holders.end()) {
+          it->second &= ~(1ULL << origin_rank);
+          if (it->second == 0) {
+            C.holders.erase(it);
+            none_left = true;
+          }
+        }
       }
+      // last replica gone: owner pushes stop paying the version touch
+      if (none_left) meta_[k].fetch_and(~(int64_t)F_HASREP);
       return;
     }
     bool local_intent;
@@ -1848,9 +1882,10 @@ class Server {
       int64_t new_ver;
       {
         std::lock_guard<std::mutex> lk(stripe(k));
-        if (!(flags_[k] & F_OWNER)) return;  // raced
-        voff = loc_[k];
-        flags_[k] = 0;  // absent: new local ops route remotely (loc left stale)
+        int64_t m = meta_[k].load();
+        if (!(m & F_OWNER)) return;  // raced
+        voff = mloc(m);
+        meta_[k].store(0);  // absent: new local ops route remotely
         new_ver = version_[k].fetch_add(1) + 1;
       }
       quiesce();  // no worker op may still hold the old offset
@@ -1878,8 +1913,12 @@ class Server {
     } else {
       bool is_new = dflags & D_NEW;
       // a replica of one of our keys now exists: versions become
-      // observable, so the fast path (which skips version bumps) ends.
+      // observable, so the fast paths (identity layout, and the
+      // per-key F_HASREP version-bump skip) end. HASREP must be set
+      // BEFORE reading cur_ver: any push after this read then bumps,
+      // so the replica can never miss a refresh.
       layout_identity_.store(false, std::memory_order_release);
+      meta_[k].fetch_or((int64_t)F_HASREP);
       {
         std::lock_guard<std::mutex> g(C.mu);
         C.holders[k] |= 1ULL << origin_rank;
@@ -1888,8 +1927,9 @@ class Server {
       int32_t l = len_of(k);
       {
         std::lock_guard<std::mutex> lk(stripe(k));
+        int64_t m = meta_[k].load();
         cur_ver = version_[k];
-        voff = loc_[k];
+        voff = mloc(m);
       }
       if (is_new || cur_ver != reported_ver) {
         std::lock_guard<std::mutex> g(C.mu);
@@ -2014,10 +2054,11 @@ class Server {
           bool handled = false;
           {
             std::lock_guard<std::mutex> lk(stripe(k));
-            uint8_t f = flags_[k];
+            int64_t m = meta_[k].load();
+            uint8_t f = mflags(m);
             if ((f & F_PRESENT) && !(f & F_OWNER)) {
               // delta-form apply: val += state - sync; sync = state
-              refreshes.add(loc_[k], poff, l);
+              refreshes.add(mloc(m), poff, l);
               refresh_sync.push_back(sync_loc_[k]);
               posts.push_back({k, f0, relocate, -1, ctr});
               handled = true;
@@ -2068,15 +2109,18 @@ class Server {
         std::lock_guard<std::mutex> lk(stripe(p.k));
         version_[p.k] = (uint32_t)p.new_ver;
         if (p.acquire_off >= 0) {
-          loc_[p.k] = p.acquire_off;
           sync_loc_[p.k] = -1;
-          flags_[p.k] = F_PRESENT | F_OWNER;
+          meta_[p.k].store(mpack(p.acquire_off, F_PRESENT | F_OWNER));
         } else if (p.relocate) {
           local_frees.push_back({sync_loc_[p.k], l});
           sync_loc_[p.k] = -1;
-          flags_[p.k] = F_PRESENT | F_OWNER;
+          // keep the value slot, become owner. A worker-push fetch_or
+          // of F_UPDATED racing this store can only lose the UPDATED
+          // bit, which is meaningless on an owned key (the merge kernel
+          // itself already targeted this same slot).
+          meta_[p.k].store(mpack(mloc(meta_[p.k].load()), F_PRESENT | F_OWNER));
         } else {
-          flags_[p.k].fetch_and((uint8_t)~F_STUB);  // preserves a concurrent UPDATED
+          meta_[p.k].fetch_and(~(int64_t)F_STUB);  // preserves a concurrent UPDATED
         }
       }
       if (p.relocate) {
@@ -2357,11 +2401,11 @@ class Server {
             int cnt = arrs[a].second;
             for (int i = 0; i < cnt; ++i) {
               Key k = kp[b * cnt + i];
-              uint8_t f = flags_[k].load(std::memory_order_acquire);
+              int64_t m = meta_[k].load(std::memory_order_acquire);
+              uint8_t f = mflags(m);
               if (!(f & F_PRESENT) || (f & F_STUB)) { ok = false; break; }
-              int64_t off = loc_[k].load(std::memory_order_acquire);
-              if (off & SPILL_BIT) { ok = false; break; }  // classic path handles spill
-              tmp.push_back(off);
+              if (m & MSPILL) { ok = false; break; }  // classic path handles spill
+              tmp.push_back(mloc(m));
             }
           }
           if (!ok) {
@@ -2375,11 +2419,12 @@ class Server {
             int cnt = arrs[a].second;
             for (int i = 0; i < cnt; ++i) {
               Key k = kp[b * cnt + i];
-              uint8_t f = flags_[k].load(std::memory_order_acquire);
-              if (f & F_OWNER)
-                version_[k].fetch_add(1, std::memory_order_relaxed);
-              else
-                flags_[k].fetch_or(F_UPDATED);
+              int64_t m = meta_[k].load(std::memory_order_acquire);
+              if (m & F_OWNER) {
+                if (m & F_HASREP) version_[k].fetch_add(1, std::memory_order_relaxed);
+              } else {
+                meta_[k].fetch_or((int64_t)F_UPDATED);
+              }
               P.offs[a].push_back(tmp[t++]);
             }
           }
@@ -2661,9 +2706,10 @@ class Server {
   // -1 = not locally present.
   int key_tier(int64_t k) {
     TORCH_CHECK((uint64_t)k < (uint64_t)num_keys_, "key out of range: ", k);
-    uint8_t f = flags_[k].load(std::memory_order_acquire);
+    int64_t m = meta_[k].load(std::memory_order_acquire);
+    uint8_t f = mflags(m);
     if (!(f & F_PRESENT) || (f & F_STUB)) return -1;
-    return (loc_[k].load(std::memory_order_acquire) & SPILL_BIT) ? 1 : 0;
+    return (m & MSPILL) ? 1 : 0;
   }
 
   // Promote the hottest host-spilled rows into HBM ("HBM as a cache
@@ -2721,9 +2767,10 @@ class Server {
       size_t before = cold.size();
       for (int64_t tries = 0; tries < 8 * CHUNK && cold.size() < before + CHUNK; ++tries) {
         Key k = (Key)(rng() % (uint64_t)num_keys_);
-        uint8_t f = flags_[k].load(std::memory_order_acquire);
+        int64_t m = meta_[k].load(std::memory_order_acquire);
+        uint8_t f = mflags(m);
         if (!(f & F_PRESENT) || (f & F_STUB)) continue;
-        if (loc_[k].load(std::memory_order_acquire) & SPILL_BIT) continue;
+        if (m & MSPILL) continue;
         cold.push_back({k, heat_[k].load(std::memory_order_relaxed)});
       }
       std::sort(cold.begin() + before, cold.end(),
@@ -2741,16 +2788,17 @@ class Server {
     for (auto& hc : hot) {
       if (moves >= max_moves) break;
       Key ks = hc.k;
-      uint8_t fs = flags_[ks].load(std::memory_order_acquire);
+      int64_t ms = meta_[ks].load(std::memory_order_acquire);
+      uint8_t fs = mflags(ms);
       if (!(fs & F_PRESENT) || (fs & F_STUB)) continue;
-      int64_t off_s = loc_[ks].load(std::memory_order_acquire);
+      int64_t off_s = mloc(ms);
       if (!(off_s & SPILL_BIT)) continue;
       // free HBM headroom? plain move, no eviction
       int64_t off_new = slab_.try_alloc_device(l);
       if (off_new >= 0) {
         from.add(off_s, pos, l);
         to.add(off_new, pos, l);
-        loc_[ks].store(off_new, std::memory_order_release);
+        meta_[ks].store(mpack(off_new, fs), std::memory_order_release);
         frees.push_back({off_s, l});
         pos += l;
         moves++;
@@ -2764,23 +2812,23 @@ class Server {
       // skip cold entries invalidated since sampling
       while (ci < cold.size()) {
         Key kd = cold[ci].k;
-        uint8_t fd = flags_[kd].load(std::memory_order_acquire);
-        if ((fd & F_PRESENT) && !(fd & F_STUB) &&
-            !(loc_[kd].load(std::memory_order_acquire) & SPILL_BIT))
-          break;
+        int64_t md = meta_[kd].load(std::memory_order_acquire);
+        uint8_t fd = mflags(md);
+        if ((fd & F_PRESENT) && !(fd & F_STUB) && !(md & MSPILL)) break;
         ci++;
       }
       if (ci >= cold.size()) continue;
       // hysteresis: a swap must be clearly profitable or keys ping-pong
       if ((int64_t)hc.h <= 2 * (int64_t)cold[ci].h + 1) break;
       Key kd = cold[ci].k;
-      int64_t off_d = loc_[kd].load(std::memory_order_acquire);
+      int64_t md = meta_[kd].load(std::memory_order_acquire);
+      int64_t off_d = mloc(md);
       from.add(off_s, pos, l);
       from.add(off_d, pos + l, l);
       to.add(off_d, pos, l);
       to.add(off_s, pos + l, l);
-      loc_[ks].store(off_d, std::memory_order_release);
-      loc_[kd].store(off_s, std::memory_order_release);
+      meta_[ks].store(mpack(off_d, fs), std::memory_order_release);
+      meta_[kd].store(mpack(off_s, mflags(md)), std::memory_order_release);
       pos += 2 * (int64_t)l;
       ci++;
       moves++;
@@ -2829,7 +2877,7 @@ class Server {
         Key start = k;
         while (true) {
           local_checks++;
-          uint8_t f = flags_[k].load(std::memory_order_relaxed);
+          uint8_t f = mflags(meta_[k].load(std::memory_order_relaxed));
           if ((f & F_PRESENT) && !(f & F_STUB)) break;
           k++;
           if (k >= hi) k = lo;
@@ -2877,7 +2925,8 @@ class Server {
     TORCH_CHECK((uint64_t)k < (uint64_t)num_keys_, "key out of range");
     int owner = -1;
     if ((Key)(k % world_) == (Key)rank_) owner = owner_of_[k / world_];
-    return {(int)flags_[k].load(), loc_[k].load(), (int64_t)version_[k].load(), owner};
+    int64_t m = meta_[k].load();
+    return {(int)mflags(m), (m & F_PRESENT) ? mloc(m) : -1, (int64_t)version_[k].load(), owner};
   }
 
   void enable_key_trace(torch::Tensor keys) {
@@ -2962,7 +3011,7 @@ class Server {
   torch::Tensor debug_flags() {
     auto t = torch::empty({(int64_t)num_keys_}, torch::TensorOptions().dtype(torch::kUInt8));
     uint8_t* p = t.data_ptr<uint8_t>();
-    for (int64_t i = 0; i < num_keys_; ++i) p[i] = flags_[i].load();
+    for (int64_t i = 0; i < num_keys_; ++i) p[i] = mflags(meta_[i].load());
     return t;
   }
 
@@ -3000,17 +3049,17 @@ class Server {
   std::vector<int32_t> lens_;
 
   Slab slab_;
-  // Per-key metadata. flags_/loc_/version_ are atomics so the worker
-  // metadata pass reads them LOCK-FREE (and in parallel); all WRITERS
-  // still serialize on the stripe mutexes. Consistency of the (flags,
-  // loc) pair without a common lock relies on three rules:
-  //  1. transitions to available write loc BEFORE flags,
-  //  2. transitions to absent write flags only — the stale loc field is
-  //     harmless because slab-slot reuse is stream-ordered and the sync
-  //     thread quiesces in-flight ops before structural changes,
-  //  3. bit updates racing with reads use fetch_or/fetch_and.
-  std::vector<std::atomic<uint8_t>> flags_;
-  std::vector<std::atomic<int64_t>> loc_;
+  // Per-key metadata: ONE packed atomic int64 per key (mpack/mflags/
+  // mloc) holding flags + spill bit + slab offset, so the lock-free
+  // worker metadata pass pays one cache miss per key and every
+  // (flags, loc) transition is atomic — no write-ordering rules needed.
+  // WRITERS of structural transitions still serialize on the stripe
+  // mutexes; racing bit updates use fetch_or/fetch_and (a full store
+  // under the stripe lock may only ever clobber a concurrent
+  // F_UPDATED fetch_or, and only on transitions where UPDATED becomes
+  // meaningless — replica→owner upgrades). version_ stays separate: it
+  // is only touched for keys with a granted replica (F_HASREP).
+  std::vector<std::atomic<int64_t>> meta_;
   std::vector<int64_t> sync_loc_;
   std::vector<std::atomic<uint32_t>> version_;
   std::vector<int32_t> loc_cache_;

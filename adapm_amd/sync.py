@@ -114,6 +114,7 @@ class SyncManager:
         self.timer = ActionTimer()
         self.watchdog_s = float(os.environ.get("ADAPM_WATCHDOG_S", "120"))
         self._progress = (0, time.monotonic())  # (total rounds, when it last moved)
+        self._state = "init"  # coarse comm-thread position, for the watchdog dump
         if not time_intent_actions:
             server.set_intent_ahead(1 << 40)
 
@@ -170,12 +171,13 @@ class SyncManager:
                 self._progress = (total, time.monotonic())
                 continue
             stalled_for = time.monotonic() - last_t
-            busy = pend["tickets"] or any(pend["out_queues"]) or any(pend["responses"])
-            if stalled_for > self.watchdog_s and busy:
+            if stalled_for > self.watchdog_s:
+                busy = pend["tickets"] or any(pend["out_queues"]) or any(pend["responses"])
                 print(f"[adapm WATCHDOG r{self.rt.rank}] sync rounds stalled "
-                      f"{stalled_for:.0f}s with work pending: rounds={pend['rounds']} "
-                      f"tickets={pend['tickets']} out={pend['out_queues']} "
-                      f"resp={pend['responses']}", file=sys.stderr, flush=True)
+                      f"{stalled_for:.0f}s ({'work pending' if busy else 'no queued work'}): "
+                      f"rounds={pend['rounds']} tickets={pend['tickets']} "
+                      f"out={pend['out_queues']} resp={pend['responses']} "
+                      f"state={self._state}", file=sys.stderr, flush=True)
                 self._progress = (total, time.monotonic())  # rate-limit the dump
 
     # ---------------------------------------------------------------- loop
@@ -236,14 +238,19 @@ class SyncManager:
     def _superround(self, nch, comm_dev, stop_flag):
         s = self.server
         t0 = time.perf_counter()
+        self._state = "collect"
         outs_a = [s.sync_collect(ch) for ch in range(nch)]
         t1 = time.perf_counter()
+        self._state = "exchange_a"
         all_stopped, st_a = self._exchange(nch, comm_dev, outs_a, s.sync_process, stop_flag)
         t2 = time.perf_counter()
+        self._state = "respond"
         outs_b = [s.sync_respond(ch) for ch in range(nch)]
         t3 = time.perf_counter()
+        self._state = "exchange_b"
         _, st_b = self._exchange(nch, comm_dev, outs_b, s.sync_apply, stop_flag)
         t4 = time.perf_counter()
+        self._state = "finish"
         any_work = False
         for ch in range(nch):
             a_meta, a_pay, a_local = st_a[ch]
@@ -307,18 +314,31 @@ class SyncManager:
                 self._comm_stream.wait_event(self._ev_fwd)
             sizes_d = sizes.to(comm_dev, non_blocking=False)
             gathered = [torch.zeros_like(sizes_d) for _ in range(world)]
+            self._state += ":allgather"
             dist.all_gather(gathered, sizes_d, group=group)
+            self._state += ":p2p"
             gathered = [g.cpu() for g in gathered]
             all_stopped = all(int(g[-1]) == 1 for g in gathered)
 
             # post sends/recvs in identical (peer, channel) order on every
-            # rank; meta-then-payload order pairs the tensors
+            # rank. On gloo, several same-pair messages are outstanding at
+            # once (nch channels x meta+payload) and matching by posting
+            # order alone is NOT guaranteed under gloo's IO threading —
+            # observed as a rare p2p stall (one rank stuck in p2p, the
+            # rest in the next all-gather). Distinct tags per (channel,
+            # meta|payload) make the matching explicit; phases cannot mix
+            # because every phase is fenced by the size all-gather. NCCL
+            # matches grouped P2P by posting order within the group call
+            # and ignores tags, so tags are gloo-only.
+            use_tags = rt.backend != "nccl"
             p2p = []
             recv_bufs = {}
             for peer in range(world):
                 if peer == rank:
                     continue
                 for ch in range(nch):
+                    t_meta = (2 * ch) if use_tags else 0
+                    t_pay = (2 * ch + 1) if use_tags else 0
                     base = (rank * nch + ch) * 3
                     n_meta = int(gathered[peer][base + 0])
                     n_pay = int(gathered[peer][base + 1])
@@ -327,17 +347,17 @@ class SyncManager:
                         rp = torch.empty(n_pay, dtype=torch.float32, device=comm_dev)
                         recv_bufs[(peer, ch)] = (rm, rp)
                         if n_meta:
-                            p2p.append(dist.P2POp(dist.irecv, rm, peer, group))
+                            p2p.append(dist.P2POp(dist.irecv, rm, peer, group, t_meta))
                         if n_pay:
-                            p2p.append(dist.P2POp(dist.irecv, rp, peer, group))
+                            p2p.append(dist.P2POp(dist.irecv, rp, peer, group, t_pay))
                     if (peer, ch) in msgs:
                         meta, payload = msgs[(peer, ch)]
                         sm = meta.reshape(-1).to(comm_dev)
                         sp = payload.to(comm_dev) if payload.device != comm_dev else payload
                         if sm.numel():
-                            p2p.append(dist.P2POp(dist.isend, sm, peer, group))
+                            p2p.append(dist.P2POp(dist.isend, sm, peer, group, t_meta))
                         if sp.numel():
-                            p2p.append(dist.P2POp(dist.isend, sp, peer, group))
+                            p2p.append(dist.P2POp(dist.isend, sp, peer, group, t_pay))
             if p2p:
                 reqs = dist.batch_isend_irecv(p2p)
                 for r in reqs:
