@@ -145,9 +145,13 @@ def _threshold_dist(rank, world):
     # conflicting intent -> REPLICATION (a sole remote intent would
     # relocate instead, and thresholds only govern replica sync)
     w.intent(key, 1, 20)
-    time.sleep(0.3)
     if rank == 1:
-        assert w.is_local(key[0])
+        deadline = time.monotonic() + 20
+        while not w.is_local(key[0]) and time.monotonic() < deadline:
+            time.sleep(0.05)
+        assert w.is_local(key[0]), "replication never arrived"
+    else:
+        time.sleep(0.3)
         # two pushes: the ACCUMULATED delta (what the threshold tests,
         # like the reference's val - sync_state) reaches L2 norm 0.4 < 0.5
         for _ in range(2):
