@@ -144,14 +144,21 @@ def _threshold_dist(rank, world):
     key = np.array([0], dtype=np.int64)  # owner = rank 0
     # conflicting intent -> REPLICATION (a sole remote intent would
     # relocate instead, and thresholds only govern replica sync)
-    w.intent(key, 1, 20)
+    # conflicting-intent setup must be ORDERED: if rank 1's D_NEW delta
+    # reaches the owner before rank 0's own intent is registered, the
+    # owner legitimately RELOCATES (no conflict visible yet) and owner
+    # updates are then immediately visible — not a threshold violation.
+    # Register the owner's intent first, then rank 1's.
+    if rank == 0:
+        w.intent(key, 1, 20)
+        w.wait_sync()  # intent registered within 2 rounds
+    w.barrier()
     if rank == 1:
+        w.intent(key, 1, 20)
         deadline = time.monotonic() + 20
         while not w.is_local(key[0]) and time.monotonic() < deadline:
             time.sleep(0.05)
         assert w.is_local(key[0]), "replication never arrived"
-    else:
-        time.sleep(0.3)
         # two pushes: the ACCUMULATED delta (what the threshold tests,
         # like the reference's val - sync_state) reaches L2 norm 0.4 < 0.5
         for _ in range(2):
