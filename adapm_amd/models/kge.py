@@ -269,11 +269,18 @@ class ComplEx:
 
     @torch.no_grad()
     def evaluate_full(self, triples: np.ndarray, hits_at=(1, 3, 10),
-                      chunk: int = 65536) -> dict:
+                      chunk: int = 65536, filter_triples: np.ndarray = None) -> dict:
         """Rank the true object among ALL entities, chunked (the reference
         evaluates against every entity, knowledge_graph_embeddings.cc:
         716-774 — there with OpenMP, here with the batched scoring
-        kernel)."""
+        kernel).
+
+        filter_triples: known-true (s, r, o) triples (train+valid+test).
+        When given, the FILTERED rank excludes other true objects of the
+        same (s, r) from the competitors (reference computes filtered and
+        raw ranks, knowledge_graph_embeddings.cc:544-712): `mrr`/`mr`/
+        `hits@k` become the filtered metrics and `*_raw` carry the raw
+        ones."""
         cfg = self.cfg
         w = self.worker
         B = len(triples)
@@ -301,11 +308,47 @@ class ComplEx:
             w.pull(cand, cv)
             _C.kge_complex_score(s_v, r_v, cv, sv, cfg.dim)
             better += (sv > true_scores).sum(1).float()
-        rank = 1 + better
-        out = {"mrr": float((1.0 / rank).mean().item()),
-               "mr": float(rank.mean().item()), "n": float(B)}
-        for h in hits_at:
-            out[f"hits@{h}"] = float((rank <= h).float().mean().item())
+        rank_raw = 1 + better
+
+        def metrics(rank, suffix=""):
+            m = {f"mrr{suffix}": float((1.0 / rank).mean().item()),
+                 f"mr{suffix}": float(rank.mean().item())}
+            for h in hits_at:
+                m[f"hits@{h}{suffix}"] = float((rank <= h).float().mean().item())
+            return m
+
+        if filter_triples is None:
+            out = metrics(rank_raw)
+        else:
+            # filtered rank: other KNOWN-TRUE objects of the same (s, r)
+            # do not count as competitors. Score only those few objects
+            # and subtract the ones that out-ranked the true object.
+            from collections import defaultdict
+
+            true_objs = defaultdict(list)
+            for fs, fr, fo in filter_triples:
+                true_objs[(int(fs), int(fr))].append(int(fo))
+            b_idx, obj_ids = [], []
+            for b, (ts, tr, to) in enumerate(triples):
+                for o2 in true_objs.get((int(ts), int(tr)), ()):
+                    if o2 != int(to):
+                        b_idx.append(b)
+                        obj_ids.append(o2)
+            filtered_better = torch.zeros_like(better)
+            if b_idx:
+                uniq, inv = np.unique(np.asarray(obj_ids, dtype=np.int64),
+                                      return_inverse=True)
+                u_v = torch.empty(len(uniq), cfg.row, **opts)
+                w.pull(uniq, u_v)
+                bi = torch.as_tensor(b_idx, dtype=torch.long, device=self.dev)
+                ov2 = u_v[torch.as_tensor(inv, dtype=torch.long, device=self.dev)]
+                sc = (sr_re[bi] * ov2[:, :dc] + sr_im[bi] * ov2[:, dc:cfg.dim]).sum(1)
+                beats = (sc > true_scores[bi, 0]).float()
+                filtered_better.scatter_add_(0, bi, beats)
+            rank_filt = 1 + better - filtered_better
+            out = metrics(rank_filt)
+            out.update(metrics(rank_raw, "_raw"))
+        out["n"] = float(B)
         if self.world > 1:
             vec = torch.tensor([out["n"]] + [out[k] * out["n"] for k in sorted(out) if k != "n"])
             vec = w.allreduce(vec)
@@ -457,7 +500,7 @@ def main():
         if rank == 0:
             print(f"[kge] epoch {ep}: loss {total / world:.4f} ({time.time()-t0:.1f}s)")
         if a.eval_every and (ep + 1) % a.eval_every == 0:
-            ev = (model.evaluate_full(triples[:a.eval_triples])
+            ev = (model.evaluate_full(triples[:a.eval_triples], filter_triples=triples)
                   if a.eval_full else model.evaluate(triples[:a.eval_triples]))
             if rank == 0:
                 print(f"[kge] eval: {ev}")

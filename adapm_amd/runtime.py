@@ -36,12 +36,15 @@ class Runtime:
     device: torch.device
     backend: str
     num_channels: int
-    # ONE process group for the whole sync engine, driven by ONE comm
-    # thread per rank. With NCCL, concurrent collectives on multiple
-    # communicators sharing a device can deadlock unless every rank
-    # issues them in the same global order; a single group + single
-    # issuing thread makes the order deterministic by construction.
+    # The sync engine is driven by ONE comm thread per rank, issuing on
+    # TWO process groups in a fixed alternation (size all-gather on
+    # sizes_group, batched P2P on sync_group). With NCCL, the single
+    # issuing thread makes the cross-communicator order deterministic on
+    # every rank; with gloo, separating the collective from the P2P
+    # traffic keeps them on different contexts (interleaving both on one
+    # gloo context produced rare matching stalls).
     sync_group: Optional[object] = None
+    sizes_group: Optional[object] = None
     # Worker-side barrier/allreduce ride a SEPARATE gloo group (host TCP):
     # tiny control-plane collectives that must never interleave with the
     # sync engine's NCCL traffic (they are called from worker threads,
@@ -89,14 +92,16 @@ def init_runtime(num_channels: int = 2, device: str | None = None,
                 timeout=datetime.timedelta(seconds=timeout_s),
             )
         sync_group = dist.new_group(backend=backend)
+        sizes_group = dist.new_group(backend=backend)
         worker_group = dist.new_group(backend="gloo")
     else:
         sync_group = None
+        sizes_group = None
         worker_group = None
 
     _RUNTIME = Runtime(rank=rank, world=world, device=dev, backend=backend,
                        num_channels=num_channels, sync_group=sync_group,
-                       worker_group=worker_group)
+                       sizes_group=sizes_group, worker_group=worker_group)
     return _RUNTIME
 
 

@@ -277,10 +277,15 @@ class SyncManager:
         return all_stopped, bool(any_work)
 
     def _exchange(self, nch, comm_dev, outs_per_ch, handler, stop_flag):
-        """One phase for ALL channels: a single size all-gather + a single
-        grouped P2P all-to-all-v, then handlers in deterministic order.
-        Returns (all_stopped, per-channel (meta, payload, local) totals
-        summed over all rank pairs — the superround derives idleness)."""
+        """One phase for ALL channels: a single size all-gather (on its
+        own process group) + one grouped P2P all-to-all-v where all
+        channels' records to a peer are CONCATENATED into one meta and
+        one payload message (so at most 2 same-pair messages are in
+        flight per phase — gloo showed rare matching stalls with more,
+        and interleaving collectives with P2P on one gloo context
+        stalled too, hence the separate groups). Handlers then split by
+        the per-channel sizes. Returns (all_stopped, per-channel (meta,
+        payload, local) totals — the superround derives idleness)."""
         rt = self.rt
         world, rank = rt.world, rt.rank
         group = rt.sync_group
@@ -289,7 +294,7 @@ class SyncManager:
         # size matrix row (per rank): nch x (n_meta, n_payload, local_work)
         # + one stop flag at the end
         sizes = torch.zeros(world * nch * 3 + 1, dtype=torch.int64)
-        msgs = {}
+        per_dest = [([], []) for _ in range(world)]  # (metas, payloads) in ch order
         local = []  # (ch, meta, payload) handled without transport
         for ch, outs in enumerate(outs_per_ch):
             for dest, meta, payload in outs:
@@ -301,7 +306,8 @@ class SyncManager:
                 base = (dest * nch + ch) * 3
                 sizes[base + 0] = meta.numel()
                 sizes[base + 1] = payload.numel()
-                msgs[(dest, ch)] = (meta, payload)
+                per_dest[dest][0].append(meta.reshape(-1))
+                per_dest[dest][1].append(payload)
         sizes[-1] = 1 if stop_flag else 0
 
         if use_streams:
@@ -315,49 +321,48 @@ class SyncManager:
             sizes_d = sizes.to(comm_dev, non_blocking=False)
             gathered = [torch.zeros_like(sizes_d) for _ in range(world)]
             self._state += ":allgather"
-            dist.all_gather(gathered, sizes_d, group=group)
+            dist.all_gather(gathered, sizes_d, group=rt.sizes_group)
             self._state += ":p2p"
             gathered = [g.cpu() for g in gathered]
             all_stopped = all(int(g[-1]) == 1 for g in gathered)
 
-            # post sends/recvs in identical (peer, channel) order on every
-            # rank. On gloo, several same-pair messages are outstanding at
-            # once (nch channels x meta+payload) and matching by posting
-            # order alone is NOT guaranteed under gloo's IO threading —
-            # observed as a rare p2p stall (one rank stuck in p2p, the
-            # rest in the next all-gather). Distinct tags per (channel,
-            # meta|payload) make the matching explicit; phases cannot mix
-            # because every phase is fenced by the size all-gather. NCCL
-            # matches grouped P2P by posting order within the group call
-            # and ignores tags, so tags are gloo-only.
+            # post sends/recvs in identical peer order on every rank;
+            # exactly one meta (tag 0) + one payload (tag 1) per pair
+            # per phase (tags are gloo-only; NCCL matches grouped P2P by
+            # posting order and phases are fenced by the all-gather)
             use_tags = rt.backend != "nccl"
             p2p = []
             recv_bufs = {}
             for peer in range(world):
                 if peer == rank:
                     continue
+                n_meta = n_pay = 0
                 for ch in range(nch):
-                    t_meta = (2 * ch) if use_tags else 0
-                    t_pay = (2 * ch + 1) if use_tags else 0
                     base = (rank * nch + ch) * 3
-                    n_meta = int(gathered[peer][base + 0])
-                    n_pay = int(gathered[peer][base + 1])
-                    if n_meta > 0 or n_pay > 0:
-                        rm = torch.empty(n_meta, dtype=torch.int64, device=comm_dev)
-                        rp = torch.empty(n_pay, dtype=torch.float32, device=comm_dev)
-                        recv_bufs[(peer, ch)] = (rm, rp)
-                        if n_meta:
-                            p2p.append(dist.P2POp(dist.irecv, rm, peer, group, t_meta))
-                        if n_pay:
-                            p2p.append(dist.P2POp(dist.irecv, rp, peer, group, t_pay))
-                    if (peer, ch) in msgs:
-                        meta, payload = msgs[(peer, ch)]
-                        sm = meta.reshape(-1).to(comm_dev)
-                        sp = payload.to(comm_dev) if payload.device != comm_dev else payload
-                        if sm.numel():
-                            p2p.append(dist.P2POp(dist.isend, sm, peer, group, t_meta))
-                        if sp.numel():
-                            p2p.append(dist.P2POp(dist.isend, sp, peer, group, t_pay))
+                    n_meta += int(gathered[peer][base + 0])
+                    n_pay += int(gathered[peer][base + 1])
+                if n_meta > 0 or n_pay > 0:
+                    rm = torch.empty(n_meta, dtype=torch.int64, device=comm_dev)
+                    rp = torch.empty(n_pay, dtype=torch.float32, device=comm_dev)
+                    recv_bufs[peer] = (rm, rp)
+                    if n_meta:
+                        p2p.append(dist.P2POp(dist.irecv, rm, peer, group,
+                                              0 if use_tags else 0))
+                    if n_pay:
+                        p2p.append(dist.P2POp(dist.irecv, rp, peer, group,
+                                              1 if use_tags else 0))
+                metas, pays = per_dest[peer]
+                if metas:
+                    sm = (metas[0] if len(metas) == 1 else torch.cat(metas)).to(comm_dev)
+                    sp = (pays[0] if len(pays) == 1 else torch.cat(pays))
+                    if sp.device != comm_dev:
+                        sp = sp.to(comm_dev)
+                    if sm.numel():
+                        p2p.append(dist.P2POp(dist.isend, sm, peer, group,
+                                              0 if use_tags else 0))
+                    if sp.numel():
+                        p2p.append(dist.P2POp(dist.isend, sp, peer, group,
+                                              1 if use_tags else 0))
             if p2p:
                 reqs = dist.batch_isend_irecv(p2p)
                 for r in reqs:
@@ -375,17 +380,29 @@ class SyncManager:
         stats = [(int(tot[ch, 0]), int(tot[ch, 1]), int(tot[ch, 2])) for ch in range(nch)]
 
         # handle local (self-targeted) messages first, then incoming in
-        # fixed (peer, channel) order for determinism
+        # fixed peer order, split per channel (ascending ch — the
+        # concatenation order on the sender)
         store_dev = rt.device
         th0 = time.perf_counter()
         for ch, meta, payload in local:
             handler(ch, rank, meta, payload)
-        for (peer, ch) in sorted(recv_bufs):
-            rm, rp = recv_bufs[(peer, ch)]
-            meta = rm.cpu()
-            _trace(rank, ch, "in ", peer, meta)
-            payload = rp if rp.device == store_dev else rp.to(store_dev)
-            handler(ch, peer, meta, payload)
+        for peer in sorted(recv_bufs):
+            rm, rp = recv_bufs[peer]
+            meta_all = rm.cpu()
+            payload_all = rp if rp.device == store_dev else rp.to(store_dev)
+            mo = po = 0
+            for ch in range(nch):
+                base = (rank * nch + ch) * 3
+                nm = int(gathered[peer][base + 0])
+                npay = int(gathered[peer][base + 1])
+                if nm == 0 and npay == 0:
+                    continue
+                meta = meta_all.narrow(0, mo, nm)
+                payload = payload_all.narrow(0, po, npay)
+                mo += nm
+                po += npay
+                _trace(rank, ch, "in ", peer, meta)
+                handler(ch, peer, meta, payload)
         self.phase_totals["handlers"] += time.perf_counter() - th0
         return all_stopped, stats
 

@@ -36,6 +36,20 @@ def _run_kge(device):
     ev = model.evaluate(triples[:64], num_candidates=100)
     assert 0.0 < ev["mrr"] <= 1.0 and ev["hits@10"] >= 0.0
 
+    # filtered eval (reference computes filtered + raw ranks,
+    # knowledge_graph_embeddings.cc:544-712): excluding known-true
+    # competitors can only improve (or tie) every rank
+    evf = model.evaluate_full(triples[:48], filter_triples=triples)
+    assert evf["mrr"] >= evf["mrr_raw"] - 1e-9
+    assert evf["mr"] <= evf["mr_raw"] + 1e-9
+    assert 0.0 < evf["mrr"] <= 1.0
+    # a (s, r) with two true objects: the other true object must not
+    # count against the evaluated one. Build a duplicate-(s,r) case:
+    dup = np.array([[1, 0, 2], [1, 0, 3]], dtype=np.int64)
+    e1 = model.evaluate_full(dup, filter_triples=dup)
+    e2 = model.evaluate_full(dup)  # raw only
+    assert e1["mrr"] >= e2["mrr"] - 1e-9
+
     # checkpoint round-trip
     path = "/tmp/kge_ckpt_test.npz"
     model.save_checkpoint(path)
