@@ -260,8 +260,19 @@ def main():
     ap.add_argument("--schedule", choices=["plain_sgd", "dsgd", "columnwise"],
                     default="dsgd")
     ap.add_argument("--lr", type=float, default=0.02)
+    ap.add_argument("--data", type=str, default="",
+                    help="train on a MatrixMarket .mma ratings file (reference "
+                         "apps/mf/io.h); default: synthetic low-rank ratings")
     ap.add_argument("--device", type=str, default=None)
     a = ap.parse_args()
+
+    data = None
+    if a.data:
+        from .data_io import read_matrix_market
+
+        rows, cols, ratings, (m, n) = read_matrix_market(a.data)
+        a.rows, a.cols = m, n
+        data = (rows, cols, ratings)
 
     cfg = MFConfig(num_rows=a.rows, num_cols=a.cols, rank=a.rank, lr=a.lr)
     _a.setup(num_keys=cfg.num_keys, num_threads=1, device=a.device)
@@ -271,7 +282,10 @@ def main():
     model.init_factors()
     rank_id = server.my_rank()
     world = server.rt.world
-    rows, cols, ratings = make_synthetic_ratings(a.nnz, a.rows, a.cols, seed=7)
+    if data is not None:
+        rows, cols, ratings = data
+    else:
+        rows, cols, ratings = make_synthetic_ratings(a.nnz, a.rows, a.cols, seed=7)
     mine = rows % world == rank_id  # row partition (reference data split)
     rows, cols, ratings = rows[mine], cols[mine], ratings[mine]
     ep_fn = getattr(model, f"epoch_{a.schedule}")

@@ -64,6 +64,7 @@ class Word2Vec:
         self.rng = np.random.default_rng(cfg.seed + self.rank)
         self._pending = []
         self._keep_prob = None
+        self.words = None  # real-corpus vocab strings (data_io.build_vocab)
 
     def init_embeddings(self):
         cfg = self.cfg
@@ -184,7 +185,27 @@ class Word2Vec:
                     buf = np.zeros((len(ks), self.cfg.row), dtype=np.float32)
                     self.worker.pull(ks, buf)
                     for j, wv in enumerate(buf[:, :self.cfg.dim]):
-                        f.write(f"w{i+j} " + " ".join(f"{x:.6f}" for x in wv) + "\n")
+                        name = self.words[i + j] if self.words else f"w{i+j}"
+                        f.write(f"{name} " + " ".join(f"{x:.6f}" for x in wv) + "\n")
+        self.worker.barrier()
+
+    def export_binary(self, path: str, max_words: int = None, chunk: int = 65536):
+        """Classic word2vec BINARY format (reference word2vec.cc:367-380)."""
+        from .data_io import export_word2vec_binary
+
+        self.drain()
+        self.worker.wait_sync()
+        self.worker.barrier()
+        if self.rank == 0:
+            n = min(self.cfg.vocab_size, max_words or self.cfg.vocab_size)
+            vecs = np.zeros((n, self.cfg.dim), dtype=np.float32)
+            for i in range(0, n, chunk):
+                ks = syn0(np.arange(i, min(i + chunk, n)))
+                buf = np.zeros((len(ks), self.cfg.row), dtype=np.float32)
+                self.worker.pull(ks, buf)
+                vecs[i:i + len(ks)] = buf[:, :self.cfg.dim]
+            names = self.words[:n] if self.words else [f"w{i}" for i in range(n)]
+            export_word2vec_binary(path, names, vecs)
         self.worker.barrier()
 
 
@@ -216,22 +237,43 @@ def main():
     ap.add_argument("--epochs", type=int, default=2)
     ap.add_argument("--batch-pairs", type=int, default=16384)
     ap.add_argument("--output", type=str, default="")
+    ap.add_argument("--binary-output", type=str, default="",
+                    help="also write the classic word2vec binary format")
+    ap.add_argument("--corpus", type=str, default="",
+                    help="train on a real text corpus (vocabulary built from the "
+                         "file, reference LearnVocabFromTrainFile); default: synthetic")
+    ap.add_argument("--min-count", type=int, default=5)
     ap.add_argument("--device", type=str, default=None)
     a = ap.parse_args()
+
+    words = None
+    if a.corpus:
+        from .data_io import build_vocab, read_sentences
+
+        words, wcounts, word2id = build_vocab(a.corpus, min_count=a.min_count,
+                                              max_vocab=a.vocab)
+        a.vocab = len(words)
+        counts = wcounts.astype(np.float64)
+    else:
+        counts = (1.0 / np.arange(1, a.vocab + 1)) ** 0.75 * 1e9
 
     cfg = W2VConfig(vocab_size=a.vocab, dim=a.dim, window=a.window, negative=a.negative,
                     batch_pairs=a.batch_pairs)
     _a.setup(num_keys=cfg.num_keys, num_threads=1, device=a.device)
     server = _a.Server(cfg.row)
-    counts = (1.0 / np.arange(1, a.vocab + 1)) ** 0.75 * 1e9
     server.enable_sampling_support("local", True, "unigram", 0, a.vocab, counts=counts)
     worker = _a.Worker(0, server)
     model = Word2Vec(cfg, server, worker)
     model.set_vocab_counts(counts)
+    model.words = words
     model.init_embeddings()
     rank = server.my_rank()
     world = server.rt.world
-    sents = make_synthetic_sentences(a.sentences // world, a.vocab, seed=rank)
+    if a.corpus:
+        sents = [s for i, s in enumerate(read_sentences(a.corpus, word2id))
+                 if i % world == rank]  # sentence partition (reference file-offset split)
+    else:
+        sents = make_synthetic_sentences(a.sentences // world, a.vocab, seed=rank)
     for ep in range(a.epochs):
         t0 = time.time()
         ctr, ctx = model.pairs_from_sentences(sents)
@@ -251,6 +293,10 @@ def main():
         model.export_text(a.output, max_words=min(a.vocab, 10000))
         if rank == 0:
             print(f"[w2v] embeddings -> {a.output}")
+    if a.binary_output:
+        model.export_binary(a.binary_output, max_words=min(a.vocab, 10000))
+        if rank == 0:
+            print(f"[w2v] binary embeddings -> {a.binary_output}")
     worker.finalize()
     server.shutdown()
 
