@@ -168,11 +168,18 @@ class Server:
         if os.environ.get("ADAPM_VERBOSE", "0") != "0":
             pl = st["pull_local"] / max(1, st["pull_keys"])
             ph = st["push_local"] / max(1, st["push_keys"])
+            rr, rp = st["replica_records"], st["replica_payloads"]
+            pct = 100.0 * rp / rr if rr else 0.0
+            hh = st["hop_hist"]
+            served = sum(hh)
+            mean_hops = (sum(i * h for i, h in enumerate(hh)) / served) if served else 0.0
             print(f"[adapm rank {self.rt.rank}] pulls: {st['pull_keys']} ({pl:.1%} local, "
                   f"{st['pull_replica']} from replicas); pushes: {st['push_keys']} "
                   f"({ph:.1%} local); relocations {st['relocations_out']}/"
                   f"{st['relocations_in']} out/in; replications {st['replications']}; "
                   f"drops {st['replica_drops']}; sync rounds {st['sync_rounds']}; "
+                  f"{pct:.1f}% replica records carried payload; remote ops served "
+                  f"{served} (mean {mean_hops:.2f} hops, hist {hh}); "
                   f"{st['bytes_sent']/1e6:.1f}/{st['bytes_recv']/1e6:.1f} MB sent/recv",
                   file=sys.stderr, flush=True)
         if self._stats_out:

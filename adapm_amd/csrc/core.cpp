@@ -1419,11 +1419,13 @@ class Server {
       return m;
     };
 
+    stat_replica_records_ += (int64_t)deltas.size();
     for (auto& d : deltas) {
       int dest = directions(d.k);
       if (dest == rank_) continue;  // raced with becoming owner
       Msg& m = add_rec(dest, M_DELTA, d.k, d.f0, d.f1, rank_);
       if (d.f1 & D_HAS_PAYLOAD) {
+        stat_replica_payloads_ += 1;
         m.extracts.add(d.val_off, m.payload_floats, d.len);
         m.extract_sync.push_back(d.sync_off);
         m.payload_floats += d.len;
@@ -1629,6 +1631,7 @@ class Server {
       }
     }
     if (!resp.keys.empty()) {
+      hop_hist_[hops > 8 ? 8 : hops] += (int64_t)resp.keys.size();
       stat_remote_pulls_served_ += (int64_t)resp.keys.size();
       resp.key = (int64_t)resp.keys.size();
       std::lock_guard<std::mutex> g(C.mu);
@@ -1681,6 +1684,7 @@ class Server {
     }
     run_scatter(apply, rows_flat, set_mode);
     if (applied > 0) {
+      hop_hist_[hops > 8 ? 8 : hops] += applied;
       stat_remote_pushes_served_ += applied;
       if (origin == rank_) {
         complete_ticket(req_id, (int)applied);
@@ -1789,6 +1793,8 @@ class Server {
             }
           }
           if (owner) {
+            int hops = (int)(f2 >> 32);
+            hop_hist_[hops > 8 ? 8 : hops] += 1;
             std::lock_guard<std::mutex> g(C.mu);
             C.responses.push_back(RespRec{(int)f0, M_PUSH_ACK, k, f1, 1, 0, -1, 0, false, {}});
             stat_remote_pushes_served_ += 1;
@@ -1810,6 +1816,8 @@ class Server {
             if (owner) voff = mloc(m);
           }
           if (owner) {
+            int hops = (int)(f2 >> 32);
+            hop_hist_[hops > 8 ? 8 : hops] += 1;
             std::lock_guard<std::mutex> g(C.mu);
             C.responses.push_back(
                 RespRec{(int)f0, M_PULL_RESP, k, f1, f2 & 0xffffffff, 0, voff, l, false, {}});
@@ -2988,6 +2996,13 @@ class Server {
     d["forwards"] = stat_forwards_.load();
     d["dropped_records"] = stat_dropped_records_.load();
     d["delta_overhops"] = stat_delta_overhops_.load();
+    {
+      py::list hh;
+      for (int i = 0; i < 9; ++i) hh.append(hop_hist_[i].load());
+      d["hop_hist"] = hh;  // served remote ops by forward-hop count (8 = 8+)
+    }
+    d["replica_records"] = stat_replica_records_.load();
+    d["replica_payloads"] = stat_replica_payloads_.load();
     d["bytes_sent"] = stat_bytes_sent_.load();
     d["bytes_recv"] = stat_bytes_recv_.load();
     d["sampling_checks"] = stat_sampling_checks_.load();
@@ -3119,6 +3134,10 @@ class Server {
   std::vector<TraceEv> trace_;
   std::chrono::steady_clock::time_point t0_ = std::chrono::steady_clock::now();
 
+  // response-hop histogram (reference sync_manager.h:482-519 hop stats):
+  // bucket = min(hops, 8) recorded when a remote request is SERVED
+  std::atomic<int64_t> hop_hist_[9] = {};
+  std::atomic<int64_t> stat_replica_records_{0}, stat_replica_payloads_{0};
   std::atomic<int64_t> stat_pulls_{0}, stat_pushes_{0}, stat_pull_keys_{0}, stat_push_keys_{0},
       stat_pull_local_{0}, stat_push_local_{0}, stat_pull_replica_{0}, stat_push_replica_{0},
       stat_remote_pulls_served_{0}, stat_remote_pushes_served_{0}, stat_relocations_{0},

@@ -115,6 +115,9 @@ class SyncManager:
         self.watchdog_s = float(os.environ.get("ADAPM_WATCHDOG_S", "120"))
         self._progress = (0, time.monotonic())  # (total rounds, when it last moved)
         self._state = "init"  # coarse comm-thread position, for the watchdog dump
+        # ADAPM_DEBUG_ROUNDS=N: every N rounds, dump outgoing record heads
+        # (diagnoses livelocks where a record regenerates every round)
+        self._dbg_every = int(os.environ.get("ADAPM_DEBUG_ROUNDS", "0"))
         if not time_intent_actions:
             server.set_intent_ahead(1 << 40)
 
@@ -215,9 +218,13 @@ class SyncManager:
             t0 = time.monotonic()
             if verbose and t0 - last_report > 10.0:
                 clocks = self.server.worker_clocks()
+                st = self.server.stats()
+                rr, rp = st["replica_records"], st["replica_payloads"]
+                pct = 100.0 * rp / rr if rr else 0.0
                 print(f"[adapm sync r{rt.rank}] "
                       f"{(n_rounds - rounds_at_report) / (t0 - last_report):.0f} rounds/s, "
-                      f"worker clocks {clocks}", flush=True)
+                      f"worker clocks {clocks}, {pct:.1f}% of replica records "
+                      f"carried payload", flush=True)
                 last_report, rounds_at_report = t0, n_rounds
             if self.time_intent_actions:
                 self.server.set_intent_ahead(self.timer.update(self.server.worker_clocks()))
@@ -274,6 +281,13 @@ class SyncManager:
         pt["respond"] += t3 - t2
         pt["exchange_b"] += t4 - t3
         pt["rounds"] += 1
+        if self._dbg_every and int(pt["rounds"]) % self._dbg_every == 0:
+            def heads(outs_all):
+                return [[(dest, meta.reshape(-1)[:15].tolist()) for dest, meta, _ in outs]
+                        for outs in outs_all]
+            print(f"[adapm dbg r{self.rt.rank} round {int(pt['rounds'])}] "
+                  f"stA={st_a} stB={st_b} outA={heads(outs_a)} outB={heads(outs_b)}",
+                  file=sys.stderr, flush=True)
         return all_stopped, bool(any_work)
 
     def _exchange(self, nch, comm_dev, outs_per_ch, handler, stop_flag):

@@ -52,6 +52,17 @@ def _dist_obs(rank, world, out):
     time.sleep(0.3)
     o = np.zeros((2, 4), dtype=np.float32)
     w.pull(np.array([0, 1], dtype=np.int64), o)
+    # a remote pull so hop/remote-served stats have data (key 2+world is
+    # owned by the other rank and has no intent signaled)
+    w.pull(np.array([2 + (1 - rank), 4 + rank], dtype=np.int64),
+           np.zeros((2, 4), dtype=np.float32))
+    w.barrier()
+    w.wait_sync(strong=True)
+    st = s.stats()
+    # hop histogram (reference sync_manager.h hop stats): every served
+    # remote op lands in a bucket; mean hops ~0 in a 2-rank run
+    assert sum(st["hop_hist"]) == st["remote_pulls_served"] + st["remote_pushes_served"]
+    assert st["replica_records"] >= st["replica_payloads"] >= 0
     w.barrier()
     w.finalize()
     s.shutdown()
