@@ -1560,7 +1560,8 @@ class Server {
         }
         if (owner) {
           run_scatter(b, r.payload, false);
-          handle_owner_delta(ch, C, r.key, (int)r.f2, r.f0, r.f1 & 0xffff);
+          handle_owner_delta(ch, C, r.key, (int)r.f2, r.f0, r.f1 & 0xffff,
+                             (int)(r.f1 >> 32));
         } else {
           requeue_bounded(ch, r, /*hops_field=*/1);
         }
@@ -1775,7 +1776,7 @@ class Server {
             requeue_bounded(ch, OutRec{0, M_DELTA, k, f0, f1, f2, pay}, /*hops_field=*/1);
             stat_forwards_ += 1;
           } else {
-            handle_owner_delta(ch, C, k, origin, f0, f1 & 0xffff);
+            handle_owner_delta(ch, C, k, origin, f0, f1 & 0xffff, (int)(f1 >> 32));
           }
           if (has_payload) poff += l;
           break;
@@ -1842,7 +1843,7 @@ class Server {
 
   // owner-side replicate-vs-relocate decision (reference sync_manager.h:612-689)
   void handle_owner_delta(int ch, ChannelState& C, Key k, int origin_rank, int64_t reported_ver,
-                          int64_t dflags) {
+                          int64_t dflags, int hops = 0) {
     (void)ch;
     if (dflags & D_DROPPING) {
       bool none_left = false;
@@ -1920,6 +1921,17 @@ class Server {
       trace_event(k, "RELOC_OUT");
     } else {
       bool is_new = dflags & D_NEW;
+      // a FORWARDED delta means the origin's believed location of this
+      // key is stale: without a correction every per-replica poll
+      // re-chases through the manager each round (pure overhead, and a
+      // permanent multi-hop journey for its eventual data). Send the
+      // origin a residence hint (applied to its location cache).
+      if (hops > 0 && origin_rank != rank_) {
+        auto it = C.reloc_ctr.find(k);
+        int64_t rc = it == C.reloc_ctr.end() ? 0 : (int64_t)it->second;
+        std::lock_guard<std::mutex> g(C.mu);
+        C.responses.push_back(RespRec{origin_rank, M_RESIDENCE, k, rank_, rc, 0, -1, 0, false, {}});
+      }
       // a replica of one of our keys now exists: versions become
       // observable, so the fast paths (identity layout, and the
       // per-key F_HASREP version-bump skip) end. HASREP must be set
@@ -1948,7 +1960,12 @@ class Server {
   }
 
   void apply_residence(Key k, int new_owner, uint32_t ctr) {
-    if (manager_of(k) != rank_) return;  // stale routing; drop
+    if (manager_of(k) != rank_) {
+      // not the manager: the record is a location-cache hint (owner-
+      // side correction of our stale believed location)
+      if (use_loc_cache_ && new_owner != rank_) loc_cache_[k] = new_owner;
+      return;
+    }
     int64_t idx = k / world_;
     std::lock_guard<std::mutex> lk(stripe(k));
     if ((int32_t)(ctr - mgr_reloc_ctr_[idx]) > 0) {  // monotonic, wrap-safe
@@ -2274,10 +2291,21 @@ class Server {
   void sync_finish(int ch, bool globally_idle = false) {
     if (globally_idle) {
       // self-handled records can requeue work invisibly to peers; a
-      // round with anything still queued locally is not idle
+      // round with DATA still queued locally is not idle. Payload-less
+      // forwarded polls (M_DELTA without D_HAS_PAYLOAD, M_PULL_REQ*)
+      // carry no updates and do not block idleness — a stale location
+      // cache can keep a poll chasing for a few rounds and strong
+      // WaitSync must not hang on that.
       std::lock_guard<std::mutex> g(channels_[ch].mu);
-      if (!channels_[ch].out_queue.empty() || !channels_[ch].responses.empty())
-        globally_idle = false;
+      if (!channels_[ch].responses.empty()) globally_idle = false;
+      for (const auto& r : channels_[ch].out_queue) {
+        bool data = (r.payload.defined() && r.payload.numel() > 0) ||
+                    r.code == M_PUSH_REQ || r.code == M_SET_REQ || r.code == M_PUSH_REQ_BULK;
+        if (data) {
+          globally_idle = false;
+          break;
+        }
+      }
     }
     {
       std::lock_guard<std::mutex> g(rounds_mu_);
