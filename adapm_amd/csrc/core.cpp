@@ -3224,11 +3224,85 @@ void kge_complex_score(torch::Tensor s, torch::Tensor r, torch::Tensor cand,
   int B = (int)scores.size(0), E = (int)scores.size(1);
   if (s.is_cuda()) {
     TORCH_CHECK(hip_available(), "kge_complex_score: CUDA tensor but no HIP device");
-    kge_complex_score_gpu(cfp(s), cfp(r), cfp(cand), fp(scores), B, E, (int)D,
-                          current_stream(s.device()));
+    if (D % 4 == 0) {
+      // GEMM-shaped B x E scoring -> MFMA-tiled kernel (matrix cores;
+      // exact f32, bitwise an fmaf chain — reference evaluates the
+      // same psi against every entity, knowledge_graph_embeddings.cc:716-774)
+      auto qbuf = torch::empty({(int64_t)B * D},
+                               torch::TensorOptions().dtype(torch::kFloat32).device(s.device()));
+      kge_complex_score_mfma_gpu(cfp(s), cfp(r), cfp(cand), fp(scores),
+                                 qbuf.data_ptr<float>(), B, E, (int)D,
+                                 current_stream(s.device()));
+    } else {
+      kge_complex_score_gpu(cfp(s), cfp(r), cfp(cand), fp(scores), B, E, (int)D,
+                            current_stream(s.device()));
+    }
   } else {
     kge_complex_score_cpu(cfp(s), cfp(r), cfp(cand), fp(scores), B, E, (int)D);
   }
+}
+
+// Grouped RESCAL step (GPU): triples pre-sorted by relation; group g =
+// triples [starts[g], starts[g+1]) sharing relation matrix rm[g]. The
+// three dim^2 products run as MFMA-tiled grouped GEMMs (U = S R,
+// dS_raw = W R^T, dR_raw = S^T W — the per-triple scalar loops of the
+// reference, knowledge_graph_embeddings.cc:895-922, recast as GEMMs);
+// the per-triple score/object-grad/W phase and the AdaGrad epilogues
+// are elementwise kernels. Numerics note: dR gets AdaGrad applied to
+// the GROUP-SUMMED gradient (minibatch semantics) where the classic
+// per-triple path transforms each triple's outer product separately
+// (both from the same pulled accumulator snapshot).
+torch::Tensor rescal_step_grouped(torch::Tensor s, torch::Tensor rm, torch::Tensor o,
+                                  torch::Tensor neg, torch::Tensor ds, torch::Tensor drl,
+                                  torch::Tensor do_, torch::Tensor dneg, torch::Tensor starts,
+                                  int64_t N, int64_t D, double lr, double eps) {
+  for (auto* t : {&s, &rm, &o, &neg, &ds, &drl, &do_, &dneg})
+    check_f32(*t, "rescal_grouped tensor");
+  TORCH_CHECK(s.is_cuda(), "rescal_step_grouped is the GPU path (CPU uses rescal_step)");
+  TORCH_CHECK(D % 4 == 0, "rescal_step_grouped: D must be a multiple of 4");
+  TORCH_CHECK(starts.scalar_type() == torch::kInt32 && starts.device().is_cpu() &&
+              starts.is_contiguous());
+  int G = (int)starts.numel() - 1;
+  int B = (int)(s.numel() / (2 * D));
+  TORCH_CHECK(rm.numel() == (int64_t)G * 2 * D * D, "rm must be [G][2*D*D]");
+  TORCH_CHECK(starts.data_ptr<int32_t>()[G] == B, "starts[-1] must equal B");
+  auto dev = s.device();
+  auto opts = torch::TensorOptions().dtype(torch::kFloat32).device(dev);
+  auto U = torch::empty({(int64_t)B * D}, opts);
+  auto W = torch::empty({(int64_t)B * D}, opts);
+  auto dRraw = torch::empty({(int64_t)G * D * D}, opts);
+  auto loss = torch::empty({B}, opts);
+  auto starts_d = starts.to(dev, /*non_blocking=*/true);
+  void* st = current_stream(dev);
+  const int32_t* sp = starts.data_ptr<int32_t>();
+  int ntiles = (int)((D + 63) / 64);
+  int tiles_bn = 0, tiles_dd = 0;
+  for (int g = 0; g < G; ++g) {
+    int Bg = sp[g + 1] - sp[g];
+    tiles_bn += ((Bg + 15) / 16) * ntiles;
+    tiles_dd += ((int)(D + 15) / 16) * ntiles;
+  }
+  // U = S @ R  (per group)
+  mfma_grouped_gemm_gpu(cfp(s), W.data_ptr<float>(), cfp(rm), U.data_ptr<float>(),
+                        starts_d.data_ptr<int32_t>(), G, (int)D, (int)(2 * D), (int)D,
+                        2 * D * D, (int)D, 0, 0, tiles_bn, st);
+  // scores / object grads / W
+  rescal_mid_gpu(U.data_ptr<float>(), cfp(o), cfp(neg), fp(do_), fp(dneg),
+                 W.data_ptr<float>(), loss.data_ptr<float>(), B, (int)N, (int)D, (float)lr,
+                 (float)eps, st);
+  // dS_raw = W @ R^T into ds's first-D columns, then AdaGrad in place
+  mfma_grouped_gemm_gpu(cfp(s), W.data_ptr<float>(), cfp(rm), fp(ds),
+                        starts_d.data_ptr<int32_t>(), G, (int)D, (int)(2 * D), (int)D,
+                        2 * D * D, (int)(2 * D), 0, 1, tiles_bn, st);
+  adagrad_rows_gpu(cfp(ds), cfp(s), fp(ds), B, (int)D, (int)(2 * D), (int)(2 * D), (float)lr,
+                   (float)eps, st);
+  // dR_raw = S^T @ W per group, then AdaGrad into drl
+  mfma_grouped_gemm_gpu(cfp(s), W.data_ptr<float>(), cfp(rm), dRraw.data_ptr<float>(),
+                        starts_d.data_ptr<int32_t>(), G, (int)D, (int)(2 * D), (int)D,
+                        2 * D * D, (int)D, (int64_t)D * D, 2, tiles_dd, st);
+  adagrad_rows_gpu(dRraw.data_ptr<float>(), cfp(rm), fp(drl), G, (int)(D * D),
+                   (int)(D * D), (int)(2 * D * D), (float)lr, (float)eps, st);
+  return loss;
 }
 
 void w2v_sgns_step(torch::Tensor ctr, torch::Tensor ctx, torch::Tensor neg, torch::Tensor dctr,
@@ -3284,6 +3358,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kge_complex_step", &kge_complex_step, py::call_guard<py::gil_scoped_release>());
   m.def("alias_draw", &alias_draw, py::call_guard<py::gil_scoped_release>());
   m.def("rescal_step", &rescal_step, py::call_guard<py::gil_scoped_release>());
+  m.def("rescal_step_grouped", &rescal_step_grouped, py::call_guard<py::gil_scoped_release>());
   m.def("kge_complex_score", &kge_complex_score, py::call_guard<py::gil_scoped_release>());
   m.def("w2v_sgns_step", &w2v_sgns_step, py::call_guard<py::gil_scoped_release>());
   m.def("mf_update_step", &mf_update_step, py::call_guard<py::gil_scoped_release>());

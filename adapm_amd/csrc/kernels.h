@@ -115,3 +115,38 @@ void alias_draw_cpu(const float* prob, const int32_t* alias, int64_t n, uint64_t
                     int64_t N, int64_t* out);
 
 }  // namespace adapm
+
+namespace adapm {
+
+// ---- MFMA (matrix-core) kernels, mfma_hip.hip (gfx950
+// v_mfma_f32_16x16x4_f32: exact f32 at the vector rate, ~2.4-2.8x a
+// VALU f32 GEMM). GPU-only: the CPU tier uses the scalar references.
+
+// Full-entity ComplEx eval scoring as a GEMM: scores[B][E] = Q @ Cand^T
+// where Q (built by an elementwise prologue into qbuf[B*D]) is the
+// (s o r) query and Cand rows are [emb(D)|accum(D)].
+void kge_complex_score_mfma_gpu(const float* s, const float* r, const float* cand,
+                                float* scores, float* qbuf, int B, int E, int D, void* stream);
+
+// Batched varying-M grouped GEMM (triples sorted by relation; group g =
+// rows [starts[g], starts[g+1]), relation matrix g at Rm + g*rstride):
+//   mode 0: Out[row] = S_g @ R_g        (U = R^T e_s per row)
+//   mode 1: Out[row] = W_g @ R_g^T      (de_s raw)
+//   mode 2: Out[g]   = S_g^T @ W_g      (dR raw, [G][D*D])
+void mfma_grouped_gemm_gpu(const float* S, const float* W, const float* R, float* out,
+                           const int* starts, int G, int D, int sstride, int wstride,
+                           int64_t rstride, int out_rstride, int64_t ostride, int mode,
+                           int total_tiles, void* stream);
+
+// per-triple middle phase of the grouped RESCAL step (scores, object
+// grads with fused AdaGrad, W accumulation) given precomputed U
+void rescal_mid_gpu(const float* U, const float* o, const float* neg, float* do_, float* dneg,
+                    float* Wm, float* loss, int B, int N, int D, float lr, float eps,
+                    void* stream);
+
+// AdaGrad epilogue: out rows = [ -lr*g/sqrt(G+g^2+eps) | g^2 ] from a
+// raw-gradient buffer and the accumulator half of the pulled rows
+void adagrad_rows_gpu(const float* grad, const float* rows, float* out, int64_t n_rows,
+                      int dvals, int gstride, int rstride, float lr, float eps, void* stream);
+
+}  // namespace adapm

@@ -32,6 +32,7 @@ SOURCES = [
     "ops_hip.hip",
     "core.cpp",
     "kernels_hip.hip",
+    "mfma_hip.hip",
     "kernels_cpu.cpp",
 ]
 
