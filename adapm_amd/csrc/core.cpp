@@ -363,6 +363,15 @@ class Server {
     return (int64_t)((uint64_t)m >> 6) | ((m & MSPILL) ? SPILL_BIT : 0);
   }
 
+  inline bool on_default_stream() const {
+#ifdef ADAPM_WITH_HIP
+    if (dev_.is_cuda())
+      return c10::hip::getCurrentHIPStream(dev_.index()) ==
+             c10::hip::getDefaultHIPStream(dev_.index());
+#endif
+    return true;
+  }
+
   inline int32_t len_of(Key k) const { return uniform_len_ >= 0 ? uniform_len_ : lens_[k]; }
   inline int manager_of(Key k) const { return (int)(k % world_); }
   inline int channel_of(Key k) const {
@@ -934,7 +943,13 @@ class Server {
                 int64_t off = mloc(m);
                 // spilled merges go through the non-atomic RMW kernel
                 // (PCIe atomics are ~144x slower; dups re-routed below)
-                bool spill_merge = heat_ && (off & SPILL_BIT) && !set_mode && dev_.is_cuda();
+                // The non-atomic RMW merge relies on STREAM ORDER to
+                // serialize concurrent merges to one slot: safe only
+                // when every worker launches on the device's default
+                // stream (our convention). A worker on a custom stream
+                // falls back to the atomic path.
+                bool spill_merge = heat_ && (off & SPILL_BIT) && !set_mode && dev_.is_cuda() &&
+                                   on_default_stream();
                 if (spill_merge)
                   P.merge_spill.add(off, offs[i], l);
                 else
