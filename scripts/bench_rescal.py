@@ -7,10 +7,14 @@ relation-sorted groups). VERDICT r01 item 8 asks for >=2x at D=128.
 Run on a GPU box:  python scripts/bench_rescal.py [--dim 128] [--batch 2048]
 """
 import argparse
+import os
+import sys
 import time
 
 import numpy as np
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
 def main():
