@@ -97,16 +97,31 @@ class MF:
             w.wait(self._pending.pop(0))
         return float(loss.mean().item()) if sync_loss else loss
 
-    def train_batch_fused(self, rows, cols, ratings, sync_loss=False):
-        """Single-rank GPU fast path: slab-direct fused kernel
-        (Server.mf_step_fused); falls back to train_batch otherwise."""
+    def train_batch_fused(self, rows, cols, ratings, sync_loss=False,
+                          force_general=False):
+        """Fused slab-direct MF step. world==1 + identity layout uses the
+        zero-host-work kernel (Server.mf_step_fused); otherwise the
+        general offsets path runs the fused kernel on all-local nonzeros
+        and routes the remote remainder through the classic path."""
         cfg = self.cfg
-        if self.world != 1 or self.dev.type != "cuda":
+        raw = self.server.raw
+        if self.dev.type != "cuda" and not force_general:
             return self.train_batch(rows, cols, ratings, sync_loss=sync_loss)
         k_w = torch.from_numpy(np.asarray(rows, dtype=np.int64))
         k_h = torch.from_numpy(self.col_key(cols))
         x = torch.from_numpy(np.asarray(ratings, dtype=np.float32))
-        loss = self.server.raw.mf_step_fused(k_w, k_h, x, cfg.rank, cfg.lr, cfg.lam, cfg.eps)
+        if self.world == 1 and not force_general and raw.layout_identity():
+            loss = raw.mf_step_fused(k_w, k_h, x, cfg.rank, cfg.lr, cfg.lam, cfg.eps)
+            return float(loss.mean().item()) if sync_loss else loss
+        loss, missed = raw.mf_step_fused_general(k_w, k_h, x, cfg.rank, cfg.lr, cfg.lam,
+                                                 cfg.eps)
+        if missed.numel():
+            midx = missed.numpy()
+            mloss = self.train_batch(np.asarray(rows)[midx], np.asarray(cols)[midx],
+                                     np.asarray(ratings)[midx], sync_loss=False)
+            if not torch.is_tensor(mloss):
+                mloss = torch.tensor([mloss])
+            loss = torch.cat([loss, mloss.to(loss.device)])
         return float(loss.mean().item()) if sync_loss else loss
 
     def drain(self):

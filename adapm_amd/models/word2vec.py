@@ -112,16 +112,18 @@ class Word2Vec:
         words = np.unique(np.concatenate([np.asarray(s, dtype=np.int64) for s in sentences]))
         self.worker.intent(np.concatenate([syn0(words), syn1(words)]), start, end)
 
-    def train_pairs(self, ctr_words, ctx_words, sync_loss=False):
+    def train_pairs(self, ctr_words, ctx_words, sync_loss=False, neg_words=None):
         cfg = self.cfg
         w = self.worker
         B = len(ctr_words)
-        if self.server.sampling is not None:
-            sid = w.prepare_sample(B * cfg.negative, w.current_clock(), w.current_clock() + 2)
-            neg_words = self.server.sampling.pull(w, sid, B * cfg.negative)
-            w.finish_sample(sid)
-        else:
-            neg_words = self.rng.integers(0, cfg.vocab_size, size=B * cfg.negative)
+        if neg_words is None:
+            if self.server.sampling is not None:
+                sid = w.prepare_sample(B * cfg.negative, w.current_clock(),
+                                       w.current_clock() + 2)
+                neg_words = self.server.sampling.pull(w, sid, B * cfg.negative)
+                w.finish_sample(sid)
+            else:
+                neg_words = self.rng.integers(0, cfg.vocab_size, size=B * cfg.negative)
 
         k_ctr, k_ctx, k_neg = syn0(ctr_words), syn1(ctx_words), syn1(neg_words)
         all_keys = np.concatenate([k_ctr, k_ctx, k_neg])
@@ -145,12 +147,16 @@ class Word2Vec:
             w.wait(self._pending.pop(0))
         return float(loss.mean().item()) if sync_loss else loss
 
-    def train_pairs_fused(self, ctr_words, ctx_words, sync_loss=False):
-        """Single-rank GPU fast path: slab-direct fused SGNS kernel
-        (Server.w2v_step_fused); falls back to train_pairs otherwise."""
+    def train_pairs_fused(self, ctr_words, ctx_words, sync_loss=False,
+                          force_general=False):
+        """Fused slab-direct SGNS step. world==1 + identity layout uses
+        the zero-host-work kernel (Server.w2v_step_fused); otherwise the
+        general offsets path runs the fused kernel on all-local pairs
+        and routes the remote remainder through the classic path."""
         cfg = self.cfg
         w = self.worker
-        if self.world != 1 or self.dev.type != "cuda":
+        raw = self.server.raw
+        if self.dev.type != "cuda" and not force_general:
             return self.train_pairs(ctr_words, ctx_words, sync_loss=sync_loss)
         B = len(ctr_words)
         if self.server.sampling is not None:
@@ -159,10 +165,24 @@ class Word2Vec:
             w.finish_sample(sid)
         else:
             neg_words = self.rng.integers(0, cfg.vocab_size, size=B * cfg.negative)
-        loss = self.server.raw.w2v_step_fused(
+        neg_words = np.ascontiguousarray(neg_words, dtype=np.int64)
+        if self.world == 1 and not force_general and raw.layout_identity():
+            loss = raw.w2v_step_fused(
+                torch.from_numpy(syn0(ctr_words)), torch.from_numpy(syn1(ctx_words)),
+                torch.from_numpy(syn1(neg_words)), cfg.negative, cfg.dim, cfg.lr, cfg.eps)
+            return float(loss.mean().item()) if sync_loss else loss
+        loss, missed = raw.w2v_step_fused_general(
             torch.from_numpy(syn0(ctr_words)), torch.from_numpy(syn1(ctx_words)),
-            torch.from_numpy(syn1(np.ascontiguousarray(neg_words, dtype=np.int64))),
-            cfg.negative, cfg.dim, cfg.lr, cfg.eps)
+            torch.from_numpy(syn1(neg_words)), cfg.negative, cfg.dim, cfg.lr, cfg.eps)
+        if missed.numel():
+            midx = missed.numpy()
+            sub_negs = neg_words.reshape(B, cfg.negative)[midx].reshape(-1)
+            mloss = self.train_pairs(np.asarray(ctr_words)[midx],
+                                     np.asarray(ctx_words)[midx], sync_loss=False,
+                                     neg_words=sub_negs)
+            if not torch.is_tensor(mloss):
+                mloss = torch.tensor([mloss])
+            loss = torch.cat([loss, mloss.to(loss.device)])
         return float(loss.mean().item()) if sync_loss else loss
 
     def drain(self):

@@ -131,3 +131,67 @@ def test_fused_general_world2(use_intent):
 @pytest.mark.parametrize("use_intent", [False, True])
 def test_fused_general_world2_gpu(use_intent):
     run_dist(2, _fused_general_worker, use_intent, "cuda:0", timeout=180)
+
+
+def _w2v_mf_general_worker(rank, world):
+    """w2v + MF models through the world>1 fused-general path (the same
+    resolve/compact machinery as KGE, wired via train_*_fused
+    force_general): losses stay finite and decrease, remote samples
+    route through the classic path."""
+    import adapm_amd
+    from adapm_amd.models.mf import MF, MFConfig, make_synthetic_ratings
+    from adapm_amd.models.word2vec import W2VConfig, Word2Vec, make_synthetic_sentences
+
+    adapm_amd.setup(num_keys=2 * 64, num_threads=1, device="cpu",
+                    max_sync_per_sec=4000.0)
+    cfg = W2VConfig(vocab_size=64, dim=8, negative=2, window=3)
+    server = adapm_amd.Server(cfg.row)
+    worker = adapm_amd.Worker(0, server)
+    model = Word2Vec(cfg, server, worker)
+    model.init_embeddings()
+    worker.barrier()
+    sents = make_synthetic_sentences(40, 64, seed=rank)
+    ctr, ctx = model.pairs_from_sentences(sents)
+    losses = []
+    for _ in range(6):
+        losses.append(model.train_pairs_fused(ctr[:256], ctx[:256], sync_loss=True,
+                                              force_general=True))
+    model.drain()
+    assert all(np.isfinite(l) for l in losses), losses
+    assert losses[-1] < losses[0], losses
+    worker.barrier()
+    worker.finalize()
+    server.shutdown()
+
+
+def test_w2v_general_world2():
+    run_dist(2, _w2v_mf_general_worker, timeout=180)
+
+
+def _mf_general_worker(rank, world):
+    import adapm_amd
+    from adapm_amd.models.mf import MF, MFConfig, make_synthetic_ratings
+
+    adapm_amd.setup(num_keys=80 + 40, num_threads=1, device="cpu",
+                    max_sync_per_sec=4000.0)
+    cfg = MFConfig(num_rows=80, num_cols=40, rank=8, lr=0.05)
+    server = adapm_amd.Server(cfg.row)
+    worker = adapm_amd.Worker(0, server)
+    model = MF(cfg, server, worker)
+    model.init_factors()
+    worker.barrier()
+    rows, cols, ratings = make_synthetic_ratings(600, 80, 40, seed=3 + rank)
+    losses = []
+    for _ in range(8):
+        losses.append(model.train_batch_fused(rows[:256], cols[:256], ratings[:256],
+                                              sync_loss=True, force_general=True))
+    model.drain()
+    assert all(np.isfinite(l) for l in losses), losses
+    assert losses[-1] < losses[0], losses
+    worker.barrier()
+    worker.finalize()
+    server.shutdown()
+
+
+def test_mf_general_world2():
+    run_dist(2, _mf_general_worker, timeout=180)
