@@ -64,6 +64,7 @@ class ComplEx:
         self.world = server.rt.world
         self.rng = np.random.default_rng(cfg.seed + self.rank)
         self._pending = []
+        self._deferred = None  # fused-general: missed samples retried next step
         self.phase_times = defaultdict(float)
 
     def _ph(self, name, t0):
@@ -196,20 +197,51 @@ class ComplEx:
                 torch.from_numpy(neg_keys), cfg.neg_samples, cfg.dim, cfg.lr, cfg.eps)
             return float(loss.mean().item()) if sync_loss else loss
 
+        # Missed (any-key-remote) samples would block THIS step on a
+        # sync round-trip through the classic path. Intent was signaled
+        # `lookahead` steps ago, so a miss is usually a straggler whose
+        # replica lands within a round — DEFER it one step and retry
+        # fused (second miss goes classic immediately, so nothing starves).
+        negs2 = neg_keys.reshape(B, cfg.neg_samples)
+        if self._deferred is not None:
+            dtr, dng = self._deferred
+            self._deferred = None
+            n_def = len(dtr)
+            triples = np.concatenate([dtr, triples])
+            negs2 = np.concatenate([dng, negs2])
+            s_keys, r_keys, o_keys = self.keys_of(triples)
+            neg_keys = negs2.reshape(-1)
+        else:
+            n_def = 0
+
         loss, missed = raw.kge_step_fused_general(
             torch.from_numpy(s_keys), torch.from_numpy(r_keys), torch.from_numpy(o_keys),
-            torch.from_numpy(neg_keys), cfg.neg_samples, cfg.dim, cfg.lr, cfg.eps)
+            torch.from_numpy(np.ascontiguousarray(neg_keys)), cfg.neg_samples, cfg.dim,
+            cfg.lr, cfg.eps)
         if missed.numel() == 0:
             return float(loss.mean().item()) if sync_loss else loss
         midx = missed.numpy()
-        sub_negs = neg_keys.reshape(B, cfg.neg_samples)[midx].reshape(-1)
-        mloss = self.train_batch(triples[midx], sync_loss=False, neg_keys=sub_negs)
-        if not torch.is_tensor(mloss):
-            mloss = torch.tensor([mloss])
-        full = torch.cat([loss, mloss.to(loss.device)])
-        return float(full.mean().item()) if sync_loss else full
+        old = midx[midx < n_def]      # second miss: classic now
+        fresh = midx[midx >= n_def]   # first miss: retry fused next step
+        if len(fresh):
+            self._deferred = (triples[fresh], negs2[fresh])
+        if len(old):
+            mloss = self.train_batch(triples[old], sync_loss=False,
+                                     neg_keys=negs2[old].reshape(-1))
+            if not torch.is_tensor(mloss):
+                mloss = torch.tensor([mloss])
+            loss = torch.cat([loss, mloss.to(loss.device)])
+        if loss.numel() == 0:
+            return 0.0 if sync_loss else loss
+        return float(loss.mean().item()) if sync_loss else loss
 
     def drain(self):
+        if self._deferred is not None:
+            # flush straggler samples through the classic path so an
+            # epoch boundary never drops work
+            dtr, dng = self._deferred
+            self._deferred = None
+            self.train_batch(dtr, sync_loss=False, neg_keys=dng.reshape(-1))
         for t in self._pending:
             self.worker.wait(t)
         self._pending.clear()
