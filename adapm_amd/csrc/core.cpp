@@ -244,7 +244,9 @@ struct ChannelState {
   std::mutex mu;
   std::deque<IntentReq> intent_queue;                                   // worker -> sync
   std::vector<IntentReq> future_intents;                                // not yet due
-  std::unordered_map<Key, std::vector<std::pair<int, Clock>>> intents;  // active local intents
+  // active local intents live in Server::intent_cnt_ (flat per-key
+  // atomic counters — a key belongs to exactly one channel); the heap
+  // below expires them against worker clocks
   // expiry min-heap (end, wid, key): avoids sweeping the whole intents
   // map every round — only due entries are popped
   std::priority_queue<std::tuple<Clock, int, Key>, std::vector<std::tuple<Clock, int, Key>>,
@@ -338,6 +340,11 @@ class Server {
     }
     // slab is zero-initialized by torch::zeros; bump-fresh slots stay zero
 
+    // flat per-key intent counters: has-intent checks are lock-free
+    // array reads instead of probes into a multi-million-entry hash map
+    // (the round's hottest lookup at high relocation churn)
+    intent_cnt_.reset(new std::atomic<uint16_t>[num_keys_]);
+    for (Key k = 0; k < num_keys_; ++k) intent_cnt_[k].store(0, std::memory_order_relaxed);
     channels_ = std::vector<ChannelState>(nch_);
     clocks_ = std::vector<std::atomic<Clock>>(std::max(1, num_workers));
     for (auto& c : clocks_) c = 0;
@@ -1229,6 +1236,8 @@ class Server {
   // MUST be called by the channel's single sync thread.
   std::vector<std::tuple<int, torch::Tensor, torch::Tensor>> sync_collect(int ch) {
     ChannelState& C = channels_[ch];
+    auto tick = [&]() { return cpp_timing_ ? std::chrono::steady_clock::now().time_since_epoch().count() : 0; };
+    int64_t tc0 = tick();
     std::deque<IntentReq> intents_in;
     std::deque<OutRec> ops_out;
     {
@@ -1288,13 +1297,14 @@ class Server {
       std::lock_guard<std::mutex> g(C.mu);
       for (auto& req : due) {
         for (Key k : req.keys) {
-          C.intents[k].push_back({req.wid, req.end});
+          intent_cnt_[k].fetch_add(1, std::memory_order_relaxed);
           C.intent_expiry.push({req.end, req.wid, k});
         }
       }
       for (auto& ns : new_stubs) C.replicas.insert(ns.k);
     }
 
+    if (cpp_timing_) { int64_t t1 = tick(); t_sy_intents_ += t1 - tc0; tc0 = t1; }
     // 2. expire due intents (heap pop — no full-map sweep) and snapshot
     // the replica set
     std::vector<Key> replica_snapshot;
@@ -1304,15 +1314,7 @@ class Server {
         auto [end, wid, k] = C.intent_expiry.top();
         if (end > clocks_[wid].load()) break;
         C.intent_expiry.pop();
-        auto it = C.intents.find(k);
-        if (it == C.intents.end()) continue;
-        auto& vec = it->second;
-        vec.erase(std::remove_if(vec.begin(), vec.end(),
-                                 [&](const std::pair<int, Clock>& p) {
-                                   return p.first == wid && p.second == end;
-                                 }),
-                  vec.end());
-        if (vec.empty()) C.intents.erase(it);
+        intent_cnt_[k].fetch_sub(1, std::memory_order_relaxed);
       }
       replica_snapshot.assign(C.replicas.begin(), C.replicas.end());
     }
@@ -1331,12 +1333,8 @@ class Server {
       std::vector<int64_t> nsync;
       std::vector<Key> cand;
       for (Key k : replica_snapshot) {
-        bool has_intent;
-        {
-          std::lock_guard<std::mutex> g(C.mu);
-          has_intent = C.intents.count(k) > 0;
-        }
-        if (!has_intent) continue;  // would drop: always ships
+        if (intent_cnt_[k].load(std::memory_order_relaxed) == 0)
+          continue;  // would drop: always ships
         std::lock_guard<std::mutex> lk(stripe(k));
         int64_t m = meta_[k].load();
         uint8_t f = mflags(m);
@@ -1369,11 +1367,7 @@ class Server {
     std::vector<DeltaRec> deltas;
     std::vector<std::pair<int64_t, int32_t>> frees;
     for (Key k : replica_snapshot) {
-      bool has_intent;
-      {
-        std::lock_guard<std::mutex> g(C.mu);
-        has_intent = C.intents.count(k) > 0;
-      }
+      bool has_intent = intent_cnt_[k].load(std::memory_order_relaxed) > 0;
       bool erase_from_replicas = false;
       {
         std::lock_guard<std::mutex> lk(stripe(k));
@@ -1412,6 +1406,8 @@ class Server {
         C.replicas.erase(k);
       }
     }
+    if (cpp_timing_) { int64_t t1 = tick(); t_sy_replicas_ += t1 - tc0; tc0 = t1;
+                       n_sy_deltas_ += (int64_t)deltas.size(); }
     // no worker op may still be between its metadata read and kernel
     // launch: (a) dropped slots must not be reused under it, (b) the
     // delta-extract kernels must be enqueued AFTER any concurrent push
@@ -1498,6 +1494,7 @@ class Server {
     // already cleared under stripe locks + quiesce)
     for (auto& [off, len] : frees) slab_.free_(off, len);
 
+    if (cpp_timing_) t_sy_build_ += tick() - tc0;
     return out;
   }
 
@@ -1738,6 +1735,7 @@ class Server {
 
   void sync_process(int ch, int src, torch::Tensor meta, torch::Tensor payload) {
     ChannelState& C = channels_[ch];
+    int64_t tp0 = cpp_timing_ ? std::chrono::steady_clock::now().time_since_epoch().count() : 0;
     meta = meta.contiguous();
     int64_t n_words = meta.numel();
     const int64_t* mp = meta.data_ptr<int64_t>();
@@ -1854,6 +1852,10 @@ class Server {
     }
     run_scatter(merges, payload, false);
     run_scatter(assigns, payload, true);
+    if (cpp_timing_) {
+      t_sy_proc_ += std::chrono::steady_clock::now().time_since_epoch().count() - tp0;
+      n_sy_proc_recs_ += n_words / REC_I64;
+    }
   }
 
   // owner-side replicate-vs-relocate decision (reference sync_manager.h:612-689)
@@ -1877,11 +1879,10 @@ class Server {
       if (none_left) meta_[k].fetch_and(~(int64_t)F_HASREP);
       return;
     }
-    bool local_intent;
+    bool local_intent = intent_cnt_[k].load(std::memory_order_relaxed) > 0;
     uint64_t other_holders;
     {
       std::lock_guard<std::mutex> g(C.mu);
-      local_intent = C.intents.count(k) > 0;
       auto it = C.holders.find(k);
       other_holders = (it == C.holders.end() ? 0 : it->second) & ~(1ULL << origin_rank);
     }
@@ -2048,6 +2049,7 @@ class Server {
 
   void sync_apply(int ch, int src, torch::Tensor meta, torch::Tensor payload) {
     ChannelState& C = channels_[ch];
+    int64_t tp0 = cpp_timing_ ? std::chrono::steady_clock::now().time_since_epoch().count() : 0;
     meta = meta.contiguous();
     int64_t n_words = meta.numel();
     const int64_t* mp = meta.data_ptr<int64_t>();
@@ -2176,6 +2178,10 @@ class Server {
       }
     }
     for (auto& [off, len] : local_frees) slab_.free_(off, len);
+    if (cpp_timing_) {
+      t_sy_apply_ += std::chrono::steady_clock::now().time_since_epoch().count() - tp0;
+      n_sy_apply_recs_ += n_words / REC_I64;
+    }
   }
 
   // bulk pull response: ONE batched copy kernel from the payload rows into
@@ -3059,6 +3065,14 @@ class Server {
     for (auto& c : channels_) rounds += c.rounds.load();
     d["sync_rounds"] = rounds;
     d["t_pass_ms"] = t_pass_.load() / 1e6;
+    d["t_sy_intents_ms"] = t_sy_intents_.load() / 1e6;
+    d["t_sy_replicas_ms"] = t_sy_replicas_.load() / 1e6;
+    d["t_sy_build_ms"] = t_sy_build_.load() / 1e6;
+    d["t_sy_proc_ms"] = t_sy_proc_.load() / 1e6;
+    d["t_sy_apply_ms"] = t_sy_apply_.load() / 1e6;
+    d["n_sy_deltas"] = n_sy_deltas_.load();
+    d["n_sy_proc_recs"] = n_sy_proc_recs_.load();
+    d["n_sy_apply_recs"] = n_sy_apply_recs_.load();
     d["t_todev_ms"] = t_todev_.load() / 1e6;
     d["t_launch_ms"] = t_launch_.load() / 1e6;
     d["t_misc_ms"] = t_misc_.load() / 1e6;
@@ -3127,6 +3141,7 @@ class Server {
   std::mutex cpu_val_mu_;
 
   std::vector<ChannelState> channels_;
+  std::unique_ptr<std::atomic<uint16_t>[]> intent_cnt_;  // active-intent count per key
   std::vector<std::atomic<Clock>> clocks_;
   Clock intent_ahead_ = 1LL << 40;  // default: act on intents immediately
 
@@ -3155,6 +3170,10 @@ class Server {
  public:
   // env ADAPM_CPP_TIMING=1: nanosecond accounting of the worker-op host path
   std::atomic<int64_t> t_pass_{0}, t_todev_{0}, t_launch_{0}, t_calls_{0}, t_misc_{0};
+  // sync-path phase accounting (ADAPM_CPP_TIMING=1): where round host
+  // time goes at high relocation churn
+  std::atomic<int64_t> t_sy_intents_{0}, t_sy_replicas_{0}, t_sy_build_{0}, t_sy_proc_{0},
+      t_sy_apply_{0}, n_sy_deltas_{0}, n_sy_proc_recs_{0}, n_sy_apply_recs_{0};
   bool cpp_timing_ = getenv("ADAPM_CPP_TIMING") != nullptr;
  private:
   std::mutex tickets_mu_;
