@@ -1913,26 +1913,28 @@ class Server {
         meta_[k].store(0);  // absent: new local ops route remotely
         new_ver = version_[k].fetch_add(1) + 1;
       }
-      quiesce();  // no worker op may still hold the old offset
+      // NO per-key quiesce here (it ran once per relocation — hundreds
+      // of inflight-drain spins per round at churn): the slot at voff
+      // is only gathered AND freed in sync_respond, which quiesces once
+      // before its gathers — any worker op whose pass saw OWNER has its
+      // kernel enqueued by then, so the relocation payload includes
+      // every racing merge and the slot-reuse stays stream-ordered.
       if (use_loc_cache_) loc_cache_[k] = origin_rank;
       uint32_t ctr = ++C.reloc_ctr[k];  // travels with ownership (sync thread only)
       C.reloc_ctr.erase(k);
       C.reloc_round.erase(k);
+      int mgr = manager_of(k);
       {
         std::lock_guard<std::mutex> g(C.mu);
         C.holders.erase(k);
         C.responses.push_back(RespRec{origin_rank, M_REFRESH, k, new_ver,
                                       R_RELOCATE | ((int64_t)ctr << 8), 0, voff, l,
                                       /*free_after=*/true, {}});
+        if (mgr != rank_)
+          C.responses.push_back(
+              RespRec{mgr, M_RESIDENCE, k, origin_rank, (int64_t)ctr, 0, -1, 0, false, {}});
       }
-      int mgr = manager_of(k);
-      if (mgr == rank_) {
-        apply_residence(k, origin_rank, ctr);
-      } else {
-        std::lock_guard<std::mutex> g(C.mu);
-        C.responses.push_back(
-            RespRec{mgr, M_RESIDENCE, k, origin_rank, (int64_t)ctr, 0, -1, 0, false, {}});
-      }
+      if (mgr == rank_) apply_residence(k, origin_rank, ctr);
       stat_relocations_ += 1;
       trace_event(k, "RELOC_OUT");
     } else {
