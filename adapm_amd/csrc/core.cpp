@@ -247,10 +247,17 @@ struct ChannelState {
   // active local intents live in Server::intent_cnt_ (flat per-key
   // atomic counters — a key belongs to exactly one channel); the heap
   // below expires them against worker clocks
-  // expiry min-heap (end, wid, key): avoids sweeping the whole intents
-  // map every round — only due entries are popped
-  std::priority_queue<std::tuple<Clock, int, Key>, std::vector<std::tuple<Clock, int, Key>>,
-                      std::greater<>> intent_expiry;
+  // expiry min-heap: ONE entry per intent REQUEST (not per key — a
+  // request covers thousands of keys; per-key heap pushes were a
+  // measurable share of the round at churn). Popping an entry
+  // decrements every key's counter.
+  struct IntentExpiry {
+    Clock end;
+    int wid;
+    std::shared_ptr<std::vector<Key>> keys;
+    bool operator>(const IntentExpiry& o) const { return end > o.end; }
+  };
+  std::priority_queue<IntentExpiry, std::vector<IntentExpiry>, std::greater<>> intent_expiry;
   std::unordered_set<Key> replicas;                                     // local replica/stub keys
   std::unordered_map<Key, uint64_t> holders;    // owner side: ranks holding replicas
   std::deque<OutRec> out_queue;                 // pending remote ops + forwards
@@ -1296,10 +1303,9 @@ class Server {
     {
       std::lock_guard<std::mutex> g(C.mu);
       for (auto& req : due) {
-        for (Key k : req.keys) {
-          intent_cnt_[k].fetch_add(1, std::memory_order_relaxed);
-          C.intent_expiry.push({req.end, req.wid, k});
-        }
+        for (Key k : req.keys) intent_cnt_[k].fetch_add(1, std::memory_order_relaxed);
+        C.intent_expiry.push(ChannelState::IntentExpiry{
+            req.end, req.wid, std::make_shared<std::vector<Key>>(std::move(req.keys))});
       }
       for (auto& ns : new_stubs) C.replicas.insert(ns.k);
     }
@@ -1311,10 +1317,11 @@ class Server {
     {
       std::lock_guard<std::mutex> g(C.mu);
       while (!C.intent_expiry.empty()) {
-        auto [end, wid, k] = C.intent_expiry.top();
-        if (end > clocks_[wid].load()) break;
+        const auto& top = C.intent_expiry.top();
+        if (top.end > clocks_[top.wid].load()) break;
+        auto keys = top.keys;
         C.intent_expiry.pop();
-        intent_cnt_[k].fetch_sub(1, std::memory_order_relaxed);
+        for (Key k : *keys) intent_cnt_[k].fetch_sub(1, std::memory_order_relaxed);
       }
       replica_snapshot.assign(C.replicas.begin(), C.replicas.end());
     }
@@ -2073,6 +2080,7 @@ class Server {
       uint32_t ctr;
     };
     std::vector<Post> posts;
+    std::vector<Key> reloc_in;
     int64_t poff = 0;
     int64_t pos = 0;
 
@@ -2170,14 +2178,15 @@ class Server {
       if (p.relocate) {
         C.reloc_ctr[p.k] = p.ctr;
         C.reloc_round[p.k] = C.rounds.load();
-        {
-          std::lock_guard<std::mutex> g(C.mu);
-          C.replicas.erase(p.k);
-        }
+        reloc_in.push_back(p.k);
         if (use_loc_cache_) loc_cache_[p.k] = -1;
         stat_relocated_in_ += 1;
         trace_event(p.k, "RELOC_IN");
       }
+    }
+    if (!reloc_in.empty()) {  // one lock for all replica-set erases
+      std::lock_guard<std::mutex> g(C.mu);
+      for (Key k : reloc_in) C.replicas.erase(k);
     }
     for (auto& [off, len] : local_frees) slab_.free_(off, len);
     if (cpp_timing_) {
