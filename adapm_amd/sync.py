@@ -74,7 +74,14 @@ def _trace(rank, ch, direction, peer, meta):
 class ActionTimer:
     """Estimate how many worker clock ticks pass per sync round, to decide
     how far ahead of a worker's clock to act on its intents (reference
-    sync_manager.h:62-105 estimate_sync_windows_and_tune)."""
+    sync_manager.h:62-105 estimate_sync_windows_and_tune): exponential
+    smoothing (alpha=0.1) of the per-round clock advance, acted-ahead
+    window = the 0.9999 quantile of Poisson(2*lambda) — exact inverse
+    CDF for small lambda, normal approximation above 1e6 like the
+    reference."""
+
+    Q = 0.9999
+    Z = 3.7190165  # Phi^-1(0.9999)
 
     def __init__(self, alpha: float = 0.1, rounds_ahead: float = 2.0):
         self.alpha = alpha
@@ -91,10 +98,27 @@ class ActionTimer:
         self.rate = (1 - self.alpha) * self.rate + self.alpha * max(0, d)
         return self.ahead()
 
+    @classmethod
+    def poisson_quantile(cls, lam: float) -> int:
+        if lam <= 0:
+            return 0
+        if lam > 500:  # normal approximation (exp(-lam) underflows past ~745;
+            # at lam=500 the approximation is already within ~0.1%)
+            return int(math.ceil(lam + cls.Z * math.sqrt(lam)))
+        # exact inverse CDF walk
+        p = math.exp(-lam)
+        cdf = p
+        k = 0
+        while cdf < cls.Q and k < 10_000_000:
+            k += 1
+            p *= lam / k
+            cdf += p
+        return k
+
     def ahead(self) -> int:
-        # rate*rounds_ahead plus a Poisson-style safety buffer
-        r = self.rate * self.rounds_ahead
-        return int(math.ceil(r + 3.0 * math.sqrt(r + 1.0) + 8.0))
+        lam = self.rate * self.rounds_ahead
+        # +small floor so cold-start intents are still acted on promptly
+        return max(8, self.poisson_quantile(2.0 * lam))
 
 
 class SyncManager:

@@ -39,9 +39,16 @@ SOURCES = [
 MODULE = "_C"
 
 
-def build(verbose: bool = True) -> Path:
+def build(verbose: bool = True, tsan: bool = False) -> Path:
+    """tsan=True (or ADAPM_TSAN=1 / `python build.py --tsan`) compiles the
+    HOST side of the extension with ThreadSanitizer — the race-hunt build
+    the SURVEY prescribes next to the NaN-poison debug mode (the striped
+    locks / lock-free metadata / PassPool are all host code). Expect
+    noise from torch's own un-instrumented runtime; use a suppressions
+    file for torch/* frames."""
     inc, libdir = torch_paths()
-    OBJ.mkdir(exist_ok=True)
+    obj_dir = OBJ if not tsan else (CSRC / ".obj_tsan")
+    obj_dir.mkdir(exist_ok=True)
     py_inc = sysconfig.get_paths()["include"]
     soname = MODULE + sysconfig.get_config_var("EXT_SUFFIX")
     out = REPO / "adapm_amd" / soname
@@ -52,6 +59,8 @@ def build(verbose: bool = True) -> Path:
         "-DUSE_ROCM", "-D__HIP_PLATFORM_AMD__", "-DADAPM_WITH_HIP",
         "-Wno-unused-result",
     ] + [f"-I{p}" for p in inc] + [f"-I{py_inc}", f"-I{CSRC}"]
+    if tsan:
+        cflags += ["-fsanitize=thread", "-g", "-fno-omit-frame-pointer"]
 
     headers = list(CSRC.glob("*.h"))
     hdr_mtime = max((h.stat().st_mtime for h in headers), default=0)
@@ -62,7 +71,7 @@ def build(verbose: bool = True) -> Path:
         sp = CSRC / src
         if not sp.exists():
             continue
-        op = OBJ / (src.replace(".", "_") + ".o")
+        op = obj_dir / (src.replace(".", "_") + ".o")
         objs.append(op)
         if op.exists() and op.stat().st_mtime > max(sp.stat().st_mtime, hdr_mtime):
             continue
@@ -84,7 +93,8 @@ def build(verbose: bool = True) -> Path:
 
     if jobs or not out.exists():
         link = (
-            ["hipcc", "-shared", "-fPIC"] + [str(o) for o in objs] + ["-o", str(out)]
+            ["hipcc", "-shared", "-fPIC"] + (["-fsanitize=thread"] if tsan else [])
+            + [str(o) for o in objs] + ["-o", str(out)]
             + [f"-L{libdir}", "-ltorch", "-ltorch_cpu", "-lc10", "-ltorch_python",
                "-ltorch_hip", "-lc10_hip", f"-Wl,-rpath,{libdir}"]
         )
@@ -97,5 +107,9 @@ def build(verbose: bool = True) -> Path:
 
 
 if __name__ == "__main__":
-    p = build()
+    import os as _os
+    import sys as _sys
+
+    _tsan = "--tsan" in _sys.argv or _os.environ.get("ADAPM_TSAN", "0") == "1"
+    p = build(tsan=_tsan)
     print(f"built {p}")
