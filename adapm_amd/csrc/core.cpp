@@ -3380,6 +3380,21 @@ void mf_update_step(torch::Tensor w, torch::Tensor h, torch::Tensor x, torch::Te
   }
 }
 
+// MF NZSL+L2 loss reduction (reference apps/mf/loss.h); returns (se, reg)
+torch::Tensor mf_loss(torch::Tensor w, torch::Tensor h, torch::Tensor x, int64_t R,
+                      double lambda) {
+  for (auto* t : {&w, &h, &x}) check_f32(*t, "mf loss tensor");
+  int B = (int)x.numel();
+  auto out = torch::zeros({2}, torch::TensorOptions().dtype(torch::kFloat32).device(w.device()));
+  if (w.is_cuda()) {
+    mf_loss_gpu(cfp(w), cfp(h), cfp(x), fp(out), B, (int)R, (float)lambda,
+                current_stream(w.device()));
+  } else {
+    mf_loss_cpu(cfp(w), cfp(h), cfp(x), fp(out), B, (int)R, (float)lambda);
+  }
+  return out;
+}
+
 // alias-table draw: prob/alias on the op device; returns int64 keys there
 torch::Tensor alias_draw(torch::Tensor prob, torch::Tensor alias, int64_t seed, int64_t N) {
   TORCH_CHECK(prob.is_contiguous() && prob.scalar_type() == torch::kFloat32);
@@ -3407,6 +3422,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kge_complex_score", &kge_complex_score, py::call_guard<py::gil_scoped_release>());
   m.def("w2v_sgns_step", &w2v_sgns_step, py::call_guard<py::gil_scoped_release>());
   m.def("mf_update_step", &mf_update_step, py::call_guard<py::gil_scoped_release>());
+  m.def("mf_loss", &mf_loss, py::call_guard<py::gil_scoped_release>());
   py::class_<Server>(m, "Server")
       .def(py::init<int64_t, torch::Tensor, int, int, int, int, std::string, double, int, bool,
                     int64_t, int64_t, double>(),

@@ -105,6 +105,14 @@ void mf_update_step_gpu(const float* w, const float* h, const float* x, float* d
 void mf_update_step_cpu(const float* w, const float* h, const float* x, float* dw, float* dh,
                         float* loss, int B, int R, float lr, float lambda, float eps);
 
+// MF NZSL+L2 loss reduction over B nonzeros (reference apps/mf/loss.h:
+// 49-120: sum (x - w.h)^2 + lambda*(|w|^2 + |h|^2) per nonzero):
+// out[0] += squared-error sum, out[1] += regularizer sum.
+void mf_loss_gpu(const float* w, const float* h, const float* x, float* out2, int B, int R,
+                 float lambda, void* stream);
+void mf_loss_cpu(const float* w, const float* h, const float* x, float* out2, int B, int R,
+                 float lambda);
+
 // Alias-table sampling draw (negative sampling; reference unigram table,
 // word2vec.cc:125-146 — rebuilt as an O(1)-per-draw alias table):
 //  prob[n] f32, alias[n] i32 built host-side; out[N] int64 drawn keys.

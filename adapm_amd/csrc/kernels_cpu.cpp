@@ -257,6 +257,26 @@ void mf_update_step_fused_offs_cpu(float* slab, const int64_t* offs_w, const int
   }
 }
 
+void mf_loss_cpu(const float* w, const float* h, const float* x, float* out2, int B, int R,
+                 float lambda) {
+  const int row = R << 1;
+  double se = 0, reg = 0;
+  for (int b = 0; b < B; ++b) {
+    const float* wb = w + (int64_t)b * row;
+    const float* hb = h + (int64_t)b * row;
+    double pred = 0, nrm = 0;
+    for (int k = 0; k < R; ++k) {
+      pred += wb[k] * hb[k];
+      nrm += wb[k] * wb[k] + hb[k] * hb[k];
+    }
+    double e = x[b] - pred;
+    se += e * e;
+    reg += lambda * nrm;
+  }
+  out2[0] += (float)se;
+  out2[1] += (float)reg;
+}
+
 static inline uint64_t pcg_hash64_c(uint64_t x) {
   x ^= x >> 33; x *= 0xff51afd7ed558ccdULL;
   x ^= x >> 33; x *= 0xc4ceb9fe1a85ec53ULL;

@@ -254,6 +254,45 @@ __global__ void k_mf_step(const float* __restrict__ w, const float* __restrict__
   }
 }
 
+// --------------------------------------------------------------- MF loss
+
+__global__ void k_mf_loss(const float* __restrict__ w, const float* __restrict__ h,
+                          const float* __restrict__ x, float* __restrict__ out2, int B, int R,
+                          float lambda) {
+  __shared__ float lds[2][KT / 64];
+  const int row = R << 1;
+  float se = 0.f, reg = 0.f;
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    const float* wb = w + (int64_t)b * row;
+    const float* hb = h + (int64_t)b * row;
+    float part = 0.f, nrm = 0.f;
+    for (int k = threadIdx.x; k < R; k += KT) {
+      float wv = wb[k], hv = hb[k];
+      part += wv * hv;
+      nrm += wv * wv + hv * hv;
+    }
+    float pred = block_reduce_sum(part, lds[0]);
+    float n2 = block_reduce_sum(nrm, lds[1]);
+    if (threadIdx.x == 0) {
+      float e = x[b] - pred;
+      se += e * e;
+      reg += lambda * n2;
+    }
+  }
+  if (threadIdx.x == 0) {
+    atomicAdd(&out2[0], se);
+    atomicAdd(&out2[1], reg);
+  }
+}
+
+void mf_loss_gpu(const float* w, const float* h, const float* x, float* out2, int B, int R,
+                 float lambda, void* stream) {
+  if (B < 1) return;
+  int g = B > 16384 ? 16384 : B;
+  hipLaunchKernelGGL(k_mf_loss, dim3(g), dim3(KT), 0, (hipStream_t)stream, w, h, x,
+                     out2, B, R, lambda);
+}
+
 // --------------------------------------------------------------- alias draw
 
 __device__ __host__ inline uint64_t pcg_hash64(uint64_t x) {

@@ -196,8 +196,10 @@ class MF:
         self.drain()
         return float(np.mean(losses)) if losses else 0.0
 
-    def test_loss(self, rows, cols, ratings):
-        """NZSL + L2 loss over a sample, aggregated across ranks."""
+    def test_loss(self, rows, cols, ratings, include_reg: bool = False):
+        """NZSL (+ optional L2 regularizer) loss over a sample via the
+        dedicated reduction kernel (reference apps/mf/loss.h:49-120),
+        aggregated across ranks."""
         B = len(rows)
         cfg = self.cfg
         k_w = np.asarray(rows, dtype=np.int64)
@@ -206,9 +208,9 @@ class MF:
         hv = torch.zeros(B, cfg.row, dtype=torch.float32, device=self.dev)
         self.worker.pull(k_w, wv)
         self.worker.pull(k_h, hv)
-        pred = (wv[:, :cfg.rank] * hv[:, :cfg.rank]).sum(1)
         x = torch.as_tensor(np.asarray(ratings, dtype=np.float32), device=self.dev)
-        se = float(((x - pred) ** 2).sum().item())
+        out2 = _C.mf_loss(wv, hv, x, cfg.rank, cfg.lam)
+        se = float(out2[0].item()) + (float(out2[1].item()) if include_reg else 0.0)
         return self.worker.allreduce(se) / self.worker.allreduce(float(B))
 
 

@@ -434,3 +434,43 @@ def test_mf_fused_matches_classic():
     err = (after - expected).abs().max().item()
     assert err < 1e-4, f"fused mf store update mismatch: {err}"
     server.shutdown()
+
+
+def test_mf_loss_kernel_cpu():
+    """Dedicated MF loss reduction (reference apps/mf/loss.h): NZSL + L2
+    vs a torch fp32 reference."""
+    import adapm_amd
+    from adapm_amd import _C
+
+    g = torch.Generator().manual_seed(5)
+    B, R, lam = 257, 16, 0.05
+    w = torch.randn(B, 2 * R, generator=g)
+    h = torch.randn(B, 2 * R, generator=g)
+    x = torch.randn(B, generator=g)
+    out2 = _C.mf_loss(w, h, x, R, lam)
+    pred = (w[:, :R] * h[:, :R]).sum(1)
+    se_ref = ((x - pred) ** 2).sum()
+    reg_ref = lam * ((w[:, :R] ** 2).sum() + (h[:, :R] ** 2).sum())
+    torch.testing.assert_close(out2[0], se_ref, rtol=1e-4, atol=1e-3)
+    torch.testing.assert_close(out2[1], reg_ref, rtol=1e-4, atol=1e-3)
+
+
+@pytest.mark.gpu
+def test_mf_loss_kernel_gpu():
+    import adapm_amd
+    from adapm_amd import _C
+
+    if not torch.cuda.is_available():
+        pytest.skip("needs GPU")
+    g = torch.Generator().manual_seed(6)
+    B, R, lam = 4097, 128, 0.02
+    w = torch.randn(B, 2 * R, generator=g).cuda()
+    h = torch.randn(B, 2 * R, generator=g).cuda()
+    x = torch.randn(B, generator=g).cuda()
+    out2 = _C.mf_loss(w, h, x, R, lam)
+    pred = (w[:, :R] * h[:, :R]).sum(1)
+    se_ref = ((x - pred) ** 2).sum()
+    reg_ref = lam * ((w[:, :R] ** 2).sum() + (h[:, :R] ** 2).sum())
+    torch.cuda.synchronize()
+    torch.testing.assert_close(out2[0], se_ref, rtol=1e-3, atol=1e-1)
+    torch.testing.assert_close(out2[1], reg_ref, rtol=1e-3, atol=1e-1)
