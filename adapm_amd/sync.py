@@ -136,6 +136,11 @@ class SyncManager:
         self.phase_totals = defaultdict(float)
         self.threads = []
         self.timer = ActionTimer()
+        from concurrent.futures import ThreadPoolExecutor
+
+        self._handler_pool = ThreadPoolExecutor(
+            max_workers=max(1, runtime.num_channels - 1),
+            thread_name_prefix="adapm-handler")
         self.watchdog_s = float(os.environ.get("ADAPM_WATCHDOG_S", "120"))
         self._progress = (0, time.monotonic())  # (total rounds, when it last moved)
         self._state = "init"  # coarse comm-thread position, for the watchdog dump
@@ -417,13 +422,13 @@ class SyncManager:
         tot = gm.sum(dim=(0, 1))  # [nch, 3] = (meta, payload, local)
         stats = [(int(tot[ch, 0]), int(tot[ch, 1]), int(tot[ch, 2])) for ch in range(nch)]
 
-        # handle local (self-targeted) messages first, then incoming in
-        # fixed peer order, split per channel (ascending ch — the
-        # concatenation order on the sender)
+        # split incoming messages per channel (local/self-targeted first,
+        # then fixed peer order — the concatenation order on the sender)
         store_dev = rt.device
         th0 = time.perf_counter()
+        per_ch = [[] for _ in range(nch)]
         for ch, meta, payload in local:
-            handler(ch, rank, meta, payload)
+            per_ch[ch].append((rank, meta, payload))
         for peer in sorted(recv_bufs):
             rm, rp = recv_bufs[peer]
             meta_all = rm.cpu()
@@ -440,7 +445,26 @@ class SyncManager:
                 mo += nm
                 po += npay
                 _trace(rank, ch, "in ", peer, meta)
+                per_ch[ch].append((peer, meta, payload))
+
+        # run each channel's handlers as one ordered unit, channels in
+        # parallel (channels partition the key space, and the C++
+        # handlers release the GIL — this restores the per-channel
+        # parallelism the round-1 thread-per-channel engine had, without
+        # re-introducing its NCCL issuance hazards: all COMM still
+        # happens on the single comm thread)
+        def run_ch(ch):
+            for peer, meta, payload in per_ch[ch]:
                 handler(ch, peer, meta, payload)
+
+        busy_chs = [ch for ch in range(nch) if per_ch[ch]]
+        if len(busy_chs) > 1:
+            futs = [self._handler_pool.submit(run_ch, ch) for ch in busy_chs[1:]]
+            run_ch(busy_chs[0])
+            for f in futs:
+                f.result()
+        elif busy_chs:
+            run_ch(busy_chs[0])
         self.phase_totals["handlers"] += time.perf_counter() - th0
         return all_stopped, stats
 
