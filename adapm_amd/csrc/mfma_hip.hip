@@ -26,54 +26,61 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 // scores[b][e] = sum_k Q[b][k] * C[e][k]
 // Q: [B][qstride] dense query rows (first K used)
 // C: [E][cstride] candidate rows (first K used)
-// One workgroup: M-tile 16 queries x N-tile 64 candidates.
+// Workgroup tile: 16 queries x 256 candidates — each of the 4 waves
+// holds FOUR 16x16 accumulators over its 64-candidate strip, so every
+// staged A element feeds 4 MFMAs (the single-accumulator version
+// measured 29 TF/s; the issue-rate cap for f32-in MFMA is 155).
 __global__ void k_mfma_scores(const float* __restrict__ Q, const float* __restrict__ C,
                               float* __restrict__ out, int B, int E, int K, int qstride,
                               int cstride) {
   constexpr int KC = 32;
   __shared__ float lq[16][KC + 1];
-  __shared__ float lc[64][KC + 1];
+  __shared__ float lc[256][KC + 1];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
 
   int mtiles = (B + 15) / 16;
-  int ntiles = (E + 63) / 64;
+  int ntiles = (E + 255) / 256;
   for (int tile = blockIdx.x; tile < mtiles * ntiles; tile += gridDim.x) {
     int m0 = (tile % mtiles) * 16;
-    int n0 = (tile / mtiles) * 64;
-    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    int n0 = (tile / mtiles) * 256;
+    f32x4 acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
+                    {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
     for (int k0 = 0; k0 < K; k0 += KC) {
       int kc = min(KC, K - k0);
-      // stage A (16 x kc): thread t loads row t/16, cols (t%16)*2 + ...
-      // simple: 256 threads cover 16*KC = 512 floats -> 2 each
+      // stage A (16 x kc): 256 threads cover 512 floats -> 2 each
       for (int idx = threadIdx.x; idx < 16 * KC; idx += MT) {
         int r = idx / KC, c = idx % KC;
         lq[r][c] = (m0 + r < B && c < kc) ? Q[(int64_t)(m0 + r) * qstride + k0 + c] : 0.f;
       }
-      // stage B (64 x kc): 64*KC = 2048 floats -> 8 each
-      for (int idx = threadIdx.x; idx < 64 * KC; idx += MT) {
+      // stage B (256 x kc): 8192 floats -> 32 each
+      for (int idx = threadIdx.x; idx < 256 * KC; idx += MT) {
         int r = idx / KC, c = idx % KC;
         lc[r][c] = (n0 + r < E && c < kc) ? C[(int64_t)(n0 + r) * cstride + k0 + c] : 0.f;
       }
       __syncthreads();
       const int arow = lane & 15;          // A row (query within tile)
-      const int brow = (wave << 4) | (lane & 15);  // B row (candidate)
       const int kk = lane >> 4;            // k within the 4-slice
 #pragma unroll
       for (int ks = 0; ks < KC; ks += 4) {
         float a = lq[arow][ks + kk];
-        float b = lc[brow][ks + kk];
-        acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          int brow = (wave << 6) | (j << 4) | (lane & 15);
+          acc[j] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, lc[brow][ks + kk], acc[j], 0, 0, 0);
+        }
       }
       __syncthreads();
     }
-    // write: col = lane&15 (candidate within wave tile), row = (lane>>4)*4+reg
-    int col = n0 + (wave << 4) + (lane & 15);
-    if (col < E) {
+    // write: col = lane&15 within sub-tile j, row = (lane>>4)*4+reg
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int col = n0 + (wave << 6) + (j << 4) + (lane & 15);
+      if (col >= E) continue;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int row = m0 + ((lane >> 4) << 2) + r;
-        if (row < B) out[(int64_t)row * E + col] = acc[r];
+        if (row < B) out[(int64_t)row * E + col] = acc[j][r];
       }
     }
   }
@@ -107,7 +114,7 @@ void kge_complex_score_mfma_gpu(const float* s, const float* r, const float* can
     int blocks = (int)std::min<int64_t>((total + 255) / 256, 4096);
     hipLaunchKernelGGL(k_build_query, dim3(blocks), dim3(256), 0, st, s, r, qbuf, B, D);
   }
-  int mtiles = (B + 15) / 16, ntiles = (E + 63) / 64;
+  int mtiles = (B + 15) / 16, ntiles = (E + 255) / 256;
   int blocks = (int)std::min<int64_t>((int64_t)mtiles * ntiles, 8192);
   // candidate rows are [emb(D) | accum(D)]: stride 2D, first D used; the
   // query buffer is dense D
