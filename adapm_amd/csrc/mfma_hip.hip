@@ -33,7 +33,7 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 __global__ void k_mfma_scores(const float* __restrict__ Q, const float* __restrict__ C,
                               float* __restrict__ out, int B, int E, int K, int qstride,
                               int cstride) {
-  constexpr int KC = 32;
+  constexpr int KC = 16;  // keeps LDS ~17 KB/WG so ~8 workgroups co-reside per CU
   __shared__ float lq[16][KC + 1];
   __shared__ float lc[256][KC + 1];
   const int lane = threadIdx.x & 63;
