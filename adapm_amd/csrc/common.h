@@ -58,6 +58,14 @@ enum MsgCode : int64_t {
   M_PULL_RESP_BULK = 14, // hdr(code, nkeys, req_id, 0, 0) + keys + out_idx. payload: nkeys rows
   M_NACK = 15,           // hop-limit give-up: f0=req_id, f1=count. The origin's
                          // ticket fails loudly instead of hanging/acking a drop.
+  // Bulk forms of the replica-sync flow (uniform-length stores): the
+  // per-key M_DELTA/M_REFRESH/M_RESIDENCE records dominate round host
+  // cost at relocation churn (record parse + response-object per key).
+  M_DELTA_BULK = 16,     // hdr(code, nkeys, origin, has_payload, 0) + keys
+                         //  + info[nk] ((version<<8)|dflags). payload: nk rows iff has_payload
+  M_REFRESH_BULK = 17,   // hdr(code, nkeys, 0, 0, 0) + keys + vers[nk]
+                         //  + fc[nk] ((reloc_ctr<<8)|rflags). payload: nk rows
+  M_RESIDENCE_BULK = 18, // hdr(code, nkeys, 0, 0, 0) + keys + oc[nk] ((ctr<<8)|owner)
 };
 
 enum DeltaFlags : int64_t {

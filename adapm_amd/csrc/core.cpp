@@ -230,7 +230,8 @@ struct RespRec {
   bool free_after = false;  // relocation: free the slot after gathering
   torch::Tensor payload;    // alternative payload source
   std::vector<Key> keys;    // bulk responses
-  std::vector<int64_t> aux;       // bulk pulls: out indices
+  std::vector<int64_t> aux;       // bulk pulls: out indices / bulk refresh: versions
+  std::vector<int64_t> aux2;      // bulk refresh: (reloc_ctr<<8)|rflags per key
   std::vector<int64_t> slab_offs; // bulk responses: per-key gather sources
 };
 
@@ -1438,15 +1439,57 @@ class Server {
     };
 
     stat_replica_records_ += (int64_t)deltas.size();
-    for (auto& d : deltas) {
-      int dest = directions(d.k);
-      if (dest == rank_) continue;  // raced with becoming owner
-      Msg& m = add_rec(dest, M_DELTA, d.k, d.f0, d.f1, rank_);
-      if (d.f1 & D_HAS_PAYLOAD) {
-        stat_replica_payloads_ += 1;
-        m.extracts.add(d.val_off, m.payload_floats, d.len);
-        m.extract_sync.push_back(d.sync_off);
-        m.payload_floats += d.len;
+    if (uniform_len_ >= 0) {
+      // BULK delta emission: one M_DELTA_BULK record per (dest,
+      // has-payload) instead of a 5-word record per replica — the
+      // per-key record parse + response-object cost dominated round
+      // host time at relocation churn.
+      struct BD {
+        std::vector<Key> keys;
+        std::vector<int64_t> info, offs, soffs;
+      };
+      std::unordered_map<int, std::array<BD, 2>> bulk;
+      const int32_t l = uniform_len_;
+      for (auto& d : deltas) {
+        int dest = directions(d.k);
+        if (dest == rank_) continue;  // raced with becoming owner
+        int hp = (d.f1 & D_HAS_PAYLOAD) ? 1 : 0;
+        BD& bd = bulk[dest][hp];
+        bd.keys.push_back(d.k);
+        bd.info.push_back((d.f0 << 8) | (d.f1 & 0xff));
+        if (hp) {
+          stat_replica_payloads_ += 1;
+          bd.offs.push_back(d.val_off);
+          bd.soffs.push_back(d.sync_off);
+        }
+      }
+      for (auto& [dest, arr] : bulk) {
+        Msg& m = msgs[dest];
+        for (int hp = 0; hp < 2; ++hp) {
+          BD& bd = arr[hp];
+          if (bd.keys.empty()) continue;
+          int64_t nk = (int64_t)bd.keys.size();
+          m.meta.insert(m.meta.end(), {M_DELTA_BULK, nk, rank_, hp, 0});
+          m.meta.insert(m.meta.end(), bd.keys.begin(), bd.keys.end());
+          m.meta.insert(m.meta.end(), bd.info.begin(), bd.info.end());
+          for (int64_t i = 0; i < (int64_t)bd.offs.size(); ++i) {
+            m.extracts.add(bd.offs[i], m.payload_floats, l);
+            m.extract_sync.push_back(bd.soffs[i]);
+            m.payload_floats += l;
+          }
+        }
+      }
+    } else {
+      for (auto& d : deltas) {
+        int dest = directions(d.k);
+        if (dest == rank_) continue;  // raced with becoming owner
+        Msg& m = add_rec(dest, M_DELTA, d.k, d.f0, d.f1, rank_);
+        if (d.f1 & D_HAS_PAYLOAD) {
+          stat_replica_payloads_ += 1;
+          m.extracts.add(d.val_off, m.payload_floats, d.len);
+          m.extract_sync.push_back(d.sync_off);
+          m.payload_floats += d.len;
+        }
       }
     }
     for (auto& r : ops_out) {
@@ -1774,6 +1817,15 @@ class Server {
         poff += rows_floats;
         continue;
       }
+      if (code == M_DELTA_BULK) {
+        int64_t nk = k;
+        bool hp = f1 != 0;
+        handle_delta_bulk(ch, C, (int)f0, hp, mp + pos, mp + pos + nk, nk, payload, poff,
+                          merges);
+        pos += 2 * nk;
+        if (hp) poff += nk * (int64_t)uniform_len_;
+        continue;
+      }
       int32_t l = len_of(k);
 
       switch (code) {
@@ -1862,6 +1914,169 @@ class Server {
     if (cpp_timing_) {
       t_sy_proc_ += std::chrono::steady_clock::now().time_since_epoch().count() - tp0;
       n_sy_proc_recs_ += n_words / REC_I64;
+    }
+  }
+
+  // BULK owner-side delta handling (uniform stores): one record covers
+  // every replica delta from `origin` for this channel; the decisions
+  // match handle_owner_delta key-for-key but run with batched channel-
+  // mutex acquisitions and ONE bulk refresh response + per-manager bulk
+  // residence records instead of a response object per key.
+  void handle_delta_bulk(int ch, ChannelState& C, int origin, bool has_payload,
+                         const Key* keys, const int64_t* info, int64_t nk,
+                         torch::Tensor payload, int64_t poff, HostBatch& merges) {
+    const int32_t l = uniform_len_;
+    struct Item {
+      Key k;
+      int64_t ver;
+      uint8_t dflags;
+      int cls;  // 0=drop, 1=decide, 2=not-owner (requeue)
+      int64_t voff = -1;
+      uint64_t other_holders = 0;
+      bool relocate = false;
+      bool granted = false;  // replica granted: record the holder in pass 3
+    };
+    std::vector<Item> items;
+    items.reserve(nk);
+    for (int64_t i = 0; i < nk; ++i) {
+      Key k = keys[i];
+      Item it{k, info[i] >> 8, (uint8_t)(info[i] & 0xff), 1};
+      bool owner;
+      {
+        std::lock_guard<std::mutex> lk(stripe(k));
+        int64_t m = meta_[k].load();
+        owner = m & F_OWNER;
+        if (owner && has_payload) {
+          merges.add(mloc(m), poff + i * (int64_t)l, l);
+          if (m & F_HASREP) version_[k]++;
+        }
+      }
+      if (!owner) {
+        it.cls = 2;
+      } else if (it.dflags & D_DROPPING) {
+        it.cls = 0;
+      }
+      items.push_back(it);
+    }
+    // requeue non-owned records toward the believed owner (rare)
+    for (int64_t i = 0; i < nk; ++i) {
+      if (items[i].cls != 2) continue;
+      torch::Tensor pay;
+      if (has_payload) pay = payload.narrow(0, poff + i * (int64_t)l, l).clone();
+      requeue_bounded(ch, OutRec{0, M_DELTA, items[i].k, items[i].ver,
+                                 (int64_t)items[i].dflags, origin, pay},
+                      /*hops_field=*/1);
+      stat_forwards_ += 1;
+    }
+    // pass 1 (one channel lock): drop-holder removals + holder reads
+    std::vector<Key> hasrep_clear;
+    {
+      std::lock_guard<std::mutex> g(C.mu);
+      for (auto& it : items) {
+        if (it.cls == 0) {
+          auto h = C.holders.find(it.k);
+          if (h != C.holders.end()) {
+            h->second &= ~(1ULL << origin);
+            if (h->second == 0) {
+              C.holders.erase(h);
+              hasrep_clear.push_back(it.k);
+            }
+          }
+        } else if (it.cls == 1) {
+          auto h = C.holders.find(it.k);
+          it.other_holders =
+              (h == C.holders.end() ? 0 : h->second) & ~(1ULL << origin);
+        }
+      }
+    }
+    for (Key k : hasrep_clear) meta_[k].fetch_and(~(int64_t)F_HASREP);
+
+    // pass 2 (no channel lock): decisions + metadata transitions
+    RespRec rb;  // bulk refresh to `origin`
+    rb.dest = origin;
+    rb.code = M_REFRESH_BULK;
+    std::map<int, std::pair<std::vector<Key>, std::vector<int64_t>>> resid;  // mgr -> keys, oc
+    bool any_decided = false;
+    for (auto& it : items) {
+      if (it.cls != 1) continue;
+      any_decided = true;
+      bool local_intent = intent_cnt_[it.k].load(std::memory_order_relaxed) > 0;
+      bool relocate =
+          techniques_ != TECH_REPLICATION_ONLY && !local_intent && it.other_holders == 0;
+      if (relocate) {
+        auto rit = C.reloc_round.find(it.k);
+        if (rit != C.reloc_round.end() && C.rounds.load() - rit->second < 3) relocate = false;
+      }
+      if (techniques_ == TECH_RELOCATION_ONLY && !relocate) continue;
+      if (relocate) {
+        layout_identity_.store(false, std::memory_order_release);
+        int64_t new_ver;
+        {
+          std::lock_guard<std::mutex> lk(stripe(it.k));
+          int64_t m = meta_[it.k].load();
+          if (!(m & F_OWNER)) continue;  // raced
+          it.voff = mloc(m);
+          meta_[it.k].store(0);
+          new_ver = version_[it.k].fetch_add(1) + 1;
+        }
+        if (use_loc_cache_) loc_cache_[it.k] = origin;
+        uint32_t ctr = ++C.reloc_ctr[it.k];
+        C.reloc_ctr.erase(it.k);
+        C.reloc_round.erase(it.k);
+        it.relocate = true;
+        rb.keys.push_back(it.k);
+        rb.aux.push_back(new_ver);
+        rb.aux2.push_back(((int64_t)ctr << 8) | R_RELOCATE);
+        rb.slab_offs.push_back(it.voff);
+        int mgr = manager_of(it.k);
+        if (mgr == rank_) {
+          apply_residence(it.k, origin, ctr);
+        } else {
+          auto& rv = resid[mgr];
+          rv.first.push_back(it.k);
+          rv.second.push_back(((int64_t)ctr << 8) | origin);
+        }
+        stat_relocations_ += 1;
+        trace_event(it.k, "RELOC_OUT");
+      } else {
+        bool is_new = it.dflags & D_NEW;
+        it.granted = true;
+        layout_identity_.store(false, std::memory_order_release);
+        meta_[it.k].fetch_or((int64_t)F_HASREP);
+        int64_t cur_ver, voff;
+        {
+          std::lock_guard<std::mutex> lk(stripe(it.k));
+          int64_t m = meta_[it.k].load();
+          cur_ver = version_[it.k];
+          voff = mloc(m);
+        }
+        if (is_new || cur_ver != it.ver) {
+          rb.keys.push_back(it.k);
+          rb.aux.push_back(cur_ver);
+          rb.aux2.push_back(0);
+          rb.slab_offs.push_back(voff);
+          if (is_new) stat_replications_ += 1;
+        }
+      }
+    }
+    // pass 3 (one channel lock): holder grants + queue the responses
+    if (any_decided || !rb.keys.empty() || !resid.empty()) {
+      std::lock_guard<std::mutex> g(C.mu);
+      for (auto& it : items)
+        if (it.granted) C.holders[it.k] |= 1ULL << origin;
+      if (!rb.keys.empty()) {
+        rb.key = (int64_t)rb.keys.size();
+        C.responses.push_back(std::move(rb));
+      }
+      for (auto& [mgr, rv] : resid) {
+        RespRec rr;
+        rr.dest = mgr;
+        rr.code = M_RESIDENCE_BULK;
+        rr.key = (int64_t)rv.first.size();
+        rr.keys = std::move(rv.first);
+        rr.aux = std::move(rv.second);
+        C.responses.push_back(std::move(rr));
+      }
     }
   }
 
@@ -2031,6 +2246,23 @@ class Server {
         }
         continue;
       }
+      if (r.code == M_REFRESH_BULK) {
+        m.meta.insert(m.meta.end(), r.keys.begin(), r.keys.end());
+        m.meta.insert(m.meta.end(), r.aux.begin(), r.aux.end());
+        m.meta.insert(m.meta.end(), r.aux2.begin(), r.aux2.end());
+        const int32_t l = uniform_len_;
+        for (size_t i = 0; i < r.slab_offs.size(); ++i) {
+          m.gathers.add(r.slab_offs[i], m.payload_floats, l);
+          m.payload_floats += l;
+          if (r.aux2[i] & R_RELOCATE) frees.push_back({r.slab_offs[i], l});
+        }
+        continue;
+      }
+      if (r.code == M_RESIDENCE_BULK) {
+        m.meta.insert(m.meta.end(), r.keys.begin(), r.keys.end());
+        m.meta.insert(m.meta.end(), r.aux.begin(), r.aux.end());
+        continue;
+      }
       if (r.slab_off >= 0) {
         m.gathers.add(r.slab_off, m.payload_floats, r.len);
         m.payload_floats += r.len;
@@ -2095,6 +2327,48 @@ class Server {
         apply_pull_resp_bulk(f0, mp + pos, mp + pos + nk, nk, payload, poff, src);
         pos += 2 * nk;
         poff += nk * (int64_t)uniform_len_;
+        continue;
+      }
+      if (code == M_REFRESH_BULK) {
+        int64_t nk = k;
+        const Key* ks = mp + pos;
+        const int64_t* vers = mp + pos + nk;
+        const int64_t* fc = mp + pos + 2 * nk;
+        const int32_t l = uniform_len_;
+        for (int64_t i = 0; i < nk; ++i) {
+          Key kk = ks[i];
+          bool relocate = fc[i] & R_RELOCATE;
+          uint32_t ctr = (uint32_t)(fc[i] >> 8);
+          bool handled = false;
+          {
+            std::lock_guard<std::mutex> lk(stripe(kk));
+            int64_t m = meta_[kk].load();
+            uint8_t f = mflags(m);
+            if ((f & F_PRESENT) && !(f & F_OWNER)) {
+              refreshes.add(mloc(m), poff, l);
+              refresh_sync.push_back(sync_loc_[kk]);
+              posts.push_back({kk, vers[i], relocate, -1, ctr});
+              handled = true;
+            }
+          }
+          if (!handled && relocate) {
+            layout_identity_.store(false, std::memory_order_release);
+            int64_t voff = slab_.alloc(l);
+            acquires.add(voff, poff, l);
+            posts.push_back({kk, vers[i], true, voff, ctr});
+          }
+          poff += l;
+        }
+        pos += 3 * nk;
+        continue;
+      }
+      if (code == M_RESIDENCE_BULK) {
+        int64_t nk = k;
+        const Key* ks = mp + pos;
+        const int64_t* oc = mp + pos + nk;
+        for (int64_t i = 0; i < nk; ++i)
+          apply_residence(ks[i], (int)(oc[i] & 0xff), (uint32_t)(oc[i] >> 8));
+        pos += 2 * nk;
         continue;
       }
       int32_t l = len_of(k);

@@ -48,7 +48,7 @@ import torch.distributed as dist
 
 _TRACE_KEY = int(os.environ.get("ADAPM_TRACE_KEY", "-1"))
 
-_BULK_CODES = {4: 2, 5: 1, 14: 2}  # code -> extra words per key
+_BULK_CODES = {4: 2, 5: 1, 14: 2, 16: 2, 17: 3, 18: 2}  # code -> extra words per key
 
 
 def _trace(rank, ch, direction, peer, meta):
