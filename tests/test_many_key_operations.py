@@ -99,6 +99,15 @@ def _worker_loop(rank, world, techniques):
     assert np.allclose(out, exp, atol=1e-2), \
         f"rank {rank} final mismatch at keys {np.where(np.abs(out - exp) > 1e-2)[0]}"
 
+    # conservation: after a strong WaitSync every granted relocation has
+    # been applied somewhere — global out == global in, and exactly one
+    # owner exists per key (the pull above already proves ownership is
+    # reachable for every key)
+    st = s.stats()
+    reloc_out = w.allreduce(float(st["relocations_out"]))
+    reloc_in = w.allreduce(float(st["relocations_in"]))
+    assert reloc_out == reloc_in, f"relocation imbalance: {reloc_out} != {reloc_in}"
+
     w.barrier()
     w.finalize()
     s.shutdown()
