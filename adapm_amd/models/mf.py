@@ -55,6 +55,7 @@ class MF:
         self.world = server.rt.world
         self.rng = np.random.default_rng(cfg.seed + self.rank_id)
         self._pending = []
+        self._deferred = None  # fused-general: missed nonzeros retried next step
 
     def col_key(self, j):
         return self.cfg.num_rows + np.asarray(j, dtype=np.int64)
@@ -113,18 +114,45 @@ class MF:
         if self.world == 1 and not force_general and raw.layout_identity():
             loss = raw.mf_step_fused(k_w, k_h, x, cfg.rank, cfg.lr, cfg.lam, cfg.eps)
             return float(loss.mean().item()) if sync_loss else loss
+        # defer first-time misses one step (intent usually localizes
+        # them within a round); a second miss goes classic immediately
+        rows = np.asarray(rows, dtype=np.int64)
+        cols = np.asarray(cols, dtype=np.int64)
+        ratings = np.asarray(ratings, dtype=np.float32)
+        n_def = 0
+        if self._deferred is not None:
+            dr, dcs, drt = self._deferred
+            self._deferred = None
+            n_def = len(dr)
+            rows = np.concatenate([dr, rows])
+            cols = np.concatenate([dcs, cols])
+            ratings = np.concatenate([drt, ratings])
+            k_w = torch.from_numpy(rows)
+            k_h = torch.from_numpy(self.col_key(cols))
+            x = torch.from_numpy(ratings)
         loss, missed = raw.mf_step_fused_general(k_w, k_h, x, cfg.rank, cfg.lr, cfg.lam,
                                                  cfg.eps)
         if missed.numel():
             midx = missed.numpy()
-            mloss = self.train_batch(np.asarray(rows)[midx], np.asarray(cols)[midx],
-                                     np.asarray(ratings)[midx], sync_loss=False)
-            if not torch.is_tensor(mloss):
-                mloss = torch.tensor([mloss])
-            loss = torch.cat([loss, mloss.to(loss.device)])
+            old_i = midx[midx < n_def]
+            fresh = midx[midx >= n_def]
+            if len(fresh):
+                self._deferred = (rows[fresh], cols[fresh], ratings[fresh])
+            if len(old_i):
+                mloss = self.train_batch(rows[old_i], cols[old_i], ratings[old_i],
+                                         sync_loss=False)
+                if not torch.is_tensor(mloss):
+                    mloss = torch.tensor([mloss])
+                loss = torch.cat([loss, mloss.to(loss.device)])
+        if loss.numel() == 0:
+            return 0.0 if sync_loss else loss
         return float(loss.mean().item()) if sync_loss else loss
 
     def drain(self):
+        if self._deferred is not None:
+            dr, dcs, drt = self._deferred
+            self._deferred = None
+            self.train_batch(dr, dcs, drt, sync_loss=False)
         for t in self._pending:
             self.worker.wait(t)
         self._pending.clear()

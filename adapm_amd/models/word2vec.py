@@ -63,6 +63,7 @@ class Word2Vec:
         self.world = server.rt.world
         self.rng = np.random.default_rng(cfg.seed + self.rank)
         self._pending = []
+        self._deferred = None  # fused-general: missed pairs retried next step
         self._keep_prob = None
         self.words = None  # real-corpus vocab strings (data_io.build_vocab)
 
@@ -171,21 +172,44 @@ class Word2Vec:
                 torch.from_numpy(syn0(ctr_words)), torch.from_numpy(syn1(ctx_words)),
                 torch.from_numpy(syn1(neg_words)), cfg.negative, cfg.dim, cfg.lr, cfg.eps)
             return float(loss.mean().item()) if sync_loss else loss
+        # defer first-time misses one step (intent usually localizes them
+        # within a round); a second miss goes classic immediately
+        ctr_words = np.asarray(ctr_words, dtype=np.int64)
+        ctx_words = np.asarray(ctx_words, dtype=np.int64)
+        negs2 = neg_words.reshape(B, cfg.negative)
+        n_def = 0
+        if self._deferred is not None:
+            dc, dx, dn = self._deferred
+            self._deferred = None
+            n_def = len(dc)
+            ctr_words = np.concatenate([dc, ctr_words])
+            ctx_words = np.concatenate([dx, ctx_words])
+            negs2 = np.concatenate([dn, negs2])
         loss, missed = raw.w2v_step_fused_general(
             torch.from_numpy(syn0(ctr_words)), torch.from_numpy(syn1(ctx_words)),
-            torch.from_numpy(syn1(neg_words)), cfg.negative, cfg.dim, cfg.lr, cfg.eps)
+            torch.from_numpy(syn1(np.ascontiguousarray(negs2.reshape(-1)))),
+            cfg.negative, cfg.dim, cfg.lr, cfg.eps)
         if missed.numel():
             midx = missed.numpy()
-            sub_negs = neg_words.reshape(B, cfg.negative)[midx].reshape(-1)
-            mloss = self.train_pairs(np.asarray(ctr_words)[midx],
-                                     np.asarray(ctx_words)[midx], sync_loss=False,
-                                     neg_words=sub_negs)
-            if not torch.is_tensor(mloss):
-                mloss = torch.tensor([mloss])
-            loss = torch.cat([loss, mloss.to(loss.device)])
+            old = midx[midx < n_def]
+            fresh = midx[midx >= n_def]
+            if len(fresh):
+                self._deferred = (ctr_words[fresh], ctx_words[fresh], negs2[fresh])
+            if len(old):
+                mloss = self.train_pairs(ctr_words[old], ctx_words[old], sync_loss=False,
+                                         neg_words=negs2[old].reshape(-1))
+                if not torch.is_tensor(mloss):
+                    mloss = torch.tensor([mloss])
+                loss = torch.cat([loss, mloss.to(loss.device)])
+        if loss.numel() == 0:
+            return 0.0 if sync_loss else loss
         return float(loss.mean().item()) if sync_loss else loss
 
     def drain(self):
+        if self._deferred is not None:
+            dc, dx, dn = self._deferred
+            self._deferred = None
+            self.train_pairs(dc, dx, sync_loss=False, neg_words=dn.reshape(-1))
         for t in self._pending:
             self.worker.wait(t)
         self._pending.clear()
