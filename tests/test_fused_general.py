@@ -158,7 +158,13 @@ def _w2v_mf_general_worker(rank, world):
                                               force_general=True))
     model.drain()
     assert all(np.isfinite(l) for l in losses), losses
-    assert losses[-1] < losses[0], losses
+    # a step that deferred every sample reports 0.0 (nothing processed):
+    # at world=2 the syn0/syn1 parity keying makes pairs never all-local
+    # until replication kicks in, so skip empty steps
+    nz = [l for l in losses if l > 0]
+    # skip the first processed step: it covers only the biased local
+    # subset (the deferred remainder lands in step 2)
+    assert len(nz) >= 3 and nz[-1] < nz[1], losses
     worker.barrier()
     worker.finalize()
     server.shutdown()
@@ -187,7 +193,8 @@ def _mf_general_worker(rank, world):
                                               sync_loss=True, force_general=True))
     model.drain()
     assert all(np.isfinite(l) for l in losses), losses
-    assert losses[-1] < losses[0], losses
+    nz = [l for l in losses if l > 0]
+    assert len(nz) >= 3 and nz[-1] < nz[1], losses
     worker.barrier()
     worker.finalize()
     server.shutdown()
