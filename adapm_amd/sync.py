@@ -13,13 +13,16 @@ ONE comm thread per rank drives ALL channels in lockstep "superrounds":
            server.sync_apply(ch, src, ...).
   server.sync_finish(ch, globally_idle) per channel.
 
-Why one thread and one process group (round-1 ran a thread per channel on
-its own group): with NCCL, concurrent collectives on multiple
-communicators sharing one device deadlock unless every rank issues them
-in the same global order. A single comm thread iterating channels in a
-fixed order on a single communicator makes the issuance order identical
-on every rank by construction — and merging the channels' size
-all-gathers into one cuts per-round collective count. Worker-side
+Why one thread (round-1 ran a thread per channel on its own group):
+with NCCL, concurrent collectives on multiple communicators sharing one
+device deadlock unless every rank issues them in the same global order.
+A single comm thread alternating TWO fixed process groups (size
+all-gathers on one, batched P2P on the other — separate gloo contexts
+avoid collective/P2P interleaving stalls) makes the issuance order
+identical on every rank by construction — and merging the channels'
+size all-gathers into one cuts per-round collective count. Incoming
+messages are handled as one ordered unit per channel, channels in
+parallel (the C++ handlers release the GIL). Worker-side
 barrier/allreduce ride a separate *gloo* group (host TCP), so they can
 never interleave with the engine's NCCL traffic. Channels remain the
 unit of key partitioning and per-channel protocol state (reference
