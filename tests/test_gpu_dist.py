@@ -14,11 +14,12 @@ from dist_helper import run_dist
 pytestmark = pytest.mark.gpu
 
 
-def _gpu_hammer(rank, world):
+def _gpu_hammer(rank, world, techniques="all"):
     os.environ["ADAPM_FORCE_GLOO"] = "1"
     import adapm_amd
 
-    adapm_amd.setup(num_keys=64, num_threads=1, device="cuda:0", max_sync_per_sec=4000.0)
+    adapm_amd.setup(num_keys=64, num_threads=1, device="cuda:0", max_sync_per_sec=4000.0,
+                    use_techniques=techniques)
     s = adapm_amd.Server(8)
     w = adapm_amd.Worker(0, s)
     w.barrier()
@@ -37,8 +38,7 @@ def _gpu_hammer(rank, world):
         w.advance_clock()
     w.waitall()
     w.barrier()
-    w.wait_sync()
-    w.wait_sync()
+    w.wait_sync(strong=True)  # exact visibility for the sum check
     w.barrier()
     total = w.allreduce(torch.tensor(pushes, dtype=torch.float32)).numpy()
     out = torch.zeros(64, 8, device="cuda")
@@ -52,8 +52,9 @@ def _gpu_hammer(rank, world):
     s.shutdown()
 
 
-def test_gpu_store_exact_sum_under_churn_ws2():
-    run_dist(2, _gpu_hammer, timeout=300)
+@pytest.mark.parametrize("techniques", ["all", "replication_only", "relocation_only"])
+def test_gpu_store_exact_sum_under_churn_ws2(techniques):
+    run_dist(2, _gpu_hammer, techniques, timeout=300)
 
 
 def _gpu_kge_dist(rank, world):
