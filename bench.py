@@ -37,6 +37,10 @@ def main():
     ap.add_argument("--triples", type=int, default=90_000_000,
                     help="nominal epoch size (for epoch-time reporting)")
     ap.add_argument("--lookahead", type=int, default=4)
+    ap.add_argument("--zipf", type=float, default=0.0,
+                    help="entity skew exponent (0 = uniform, the default; e.g. "
+                         "1.1 gives a Zipf-like hot set as in real KGs — hot "
+                         "entities localize once instead of relocating per use)")
     ap.add_argument("--warmup-budget-s", type=float, default=4.0,
                     help="minimum untimed warmup wall-clock: after the W warmup "
                          "steps, keep running untimed steps until this budget is "
@@ -83,11 +87,16 @@ def main():
     is_cuda = dev.type == "cuda"
     rng = np.random.default_rng(1000 + rank)
 
+    def draw_entities(n):
+        if args.zipf > 0:
+            return (rng.zipf(args.zipf, size=n) - 1) % args.entities
+        return rng.integers(0, args.entities, size=n)
+
     def make_batch(i):
         return np.stack([
-            rng.integers(0, args.entities, size=args.batch),
+            draw_entities(args.batch),
             rng.integers(0, args.relations, size=args.batch),
-            rng.integers(0, args.entities, size=args.batch),
+            draw_entities(args.batch),
         ], axis=1).astype(np.int64)
 
     total_steps = args.warmup + args.steps
@@ -258,6 +267,7 @@ def main():
                 "global_batch": args.batch * world,
                 "triples": args.triples,
                 "parallelism": f"ps-async-dp{world}",
+                "zipf": args.zipf,
                 "fused_step": use_fused,
                 "epoch_time_est_s": epoch_time_est_s,
                 "triples_per_s": triples_per_s,
