@@ -127,6 +127,12 @@ def test_fused_general_world2(use_intent):
     run_dist(2, _fused_general_worker, use_intent, timeout=180)
 
 
+def test_fused_general_world3():
+    """world=3: manager != owner routing is common (key % 3), so misses,
+    forwards and the classic remainder all exercise multi-hop paths."""
+    run_dist(3, _fused_general_worker, True, timeout=180)
+
+
 @pytest.mark.gpu
 @pytest.mark.parametrize("use_intent", [False, True])
 def test_fused_general_world2_gpu(use_intent):
