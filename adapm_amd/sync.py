@@ -179,6 +179,7 @@ class SyncManager:
         for t in self.threads:
             t.join()
         self.threads = []
+        self._handler_pool.shutdown(wait=True)
         if os.environ.get("ADAPM_VERBOSE", "0") != "0" and self.phase_totals.get("rounds"):
             pt = dict(self.phase_totals)
             n = pt.pop("rounds")
