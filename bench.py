@@ -58,7 +58,10 @@ def main():
     ap.add_argument("--no-fused", action="store_true",
                     help="disable the fused slab-direct kernel (single-rank GPU fast path)")
     ap.add_argument("--device", type=str, default=None)
-    ap.add_argument("--channels", type=int, default=2)
+    ap.add_argument("--channels", type=int, default=4,
+                    help="sync channels (key-space partitions; handlers run one "
+                         "ordered unit per channel in parallel — 4 measured +5%% "
+                         "over 2 on the 2-rank rehearsal)")
     ap.add_argument("--capacity-factor", type=float, default=3.0)
     args = ap.parse_args()
 
