@@ -204,3 +204,58 @@ def test_host_spill_dist_ws2():
     from dist_helper import run_dist
 
     run_dist(2, _spill_dist_worker, timeout=240)
+
+
+@pytest.mark.gpu
+def test_host_spill_dist_ws2_gpu():
+    """Same protocol churn on a spilled GPU store (pinned host arena,
+    device-visible zero-copy; 2 ranks share cuda:0 over gloo)."""
+    import os
+
+    from dist_helper import run_dist
+
+    os.environ["ADAPM_FORCE_GLOO"] = "1"
+    run_dist(2, _spill_dist_worker_gpu, timeout=240)
+
+
+def _spill_dist_worker_gpu(rank, world):
+    import os
+
+    os.environ["ADAPM_FORCE_GLOO"] = "1"
+    _spill_dist_worker_impl(rank, world, "cuda:0")
+
+
+def _spill_dist_worker_impl(rank, world, device):
+    import adapm_amd
+
+    adapm_amd.setup(num_keys=64, num_threads=1, device=device,
+                    device_cap_gb=16 * 64 * 4 / 2**30, host_spill_gb=0.01,
+                    max_sync_per_sec=4000.0)
+    s = adapm_amd.Server(64)
+    assert s.stats()["host_spill_in_use"] > 0
+    w = adapm_amd.Worker(0, s)
+    w.barrier()
+    rng = np.random.default_rng(rank)
+    pushes = np.zeros(64)
+    for i in range(120):
+        keys = rng.choice(64, size=3, replace=False).astype(np.int64)
+        if rng.random() < 0.4:
+            w.intent(keys, w.current_clock() + 1,
+                     w.current_clock() + int(rng.integers(2, 10)))
+        w.push(keys, np.ones((3, 64), dtype=np.float32), async_=True)
+        pushes[keys] += 1
+        if rng.random() < 0.2:
+            w.pull(keys, np.zeros((3, 64), dtype=np.float32))
+        w.advance_clock()
+    w.waitall()
+    w.barrier()
+    w.wait_sync(strong=True)
+    w.barrier()
+    total = w.allreduce(torch.tensor(pushes, dtype=torch.float32)).numpy()
+    out = np.zeros((64, 64), dtype=np.float32)
+    w.pull(np.arange(64, dtype=np.int64), out)
+    assert np.allclose(out[:, 0], total, atol=1e-2), \
+        f"rank {rank} mismatch at {np.where(np.abs(out[:, 0] - total) > 1e-2)[0]}"
+    w.barrier()
+    w.finalize()
+    s.shutdown()
