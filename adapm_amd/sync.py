@@ -233,7 +233,11 @@ class SyncManager:
         # comm tensors live on the backend's device: GPU for NCCL (xGMI
         # P2P), CPU for gloo
         comm_dev = rt.device if rt.backend == "nccl" else torch.device("cpu")
-        use_streams = rt.backend == "nccl" and rt.is_cuda
+        # ADAPM_FORCE_COMM_STREAM=1: exercise the comm-stream event
+        # choreography on the gloo+CUDA test tier (otherwise it only
+        # runs under NCCL, which needs one GPU per rank)
+        use_streams = rt.is_cuda and (rt.backend == "nccl" or
+                                      os.environ.get("ADAPM_FORCE_COMM_STREAM", "0") == "1")
         if rt.is_cuda:
             torch.cuda.set_device(rt.device)
         if use_streams:
